@@ -1,0 +1,154 @@
+"""Flagship training-step benchmark (driver contract).
+
+Measures the BASELINE.json headline metric: MPGCN train-step samples/sec on a
+256-region OD config — full training semantics per step: on-device dynamic
+support construction (the reference rebuilds supports every step,
+Model_Trainer.py:106), forward, MSE loss, backward, gradient all-reduce (N>1,
+RCCL over xGMI), Adam step. bf16 compute with fp32 master weights; synthetic
+OD data (no dataset is bundled with the reference repo) and random-init
+weights. Weak scaling: per-GPU batch fixed as N grows.
+
+  python bench.py --gpus N --steps K --warmup W
+  (N>1 is launched by the driver via torch.distributed.run, one rank per GPU)
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--nodes", type=int, default=256, help="region count N")
+    ap.add_argument("--batch", type=int, default=32, help="per-GPU batch size")
+    ap.add_argument("--hidden", type=int, default=32)
+    ap.add_argument("--obs-len", type=int, default=7)
+    ap.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "float32"])
+    ap.add_argument("--device", type=str, default=None,
+                    help="override device (cpu for debug)")
+    args = ap.parse_args()
+
+    import torch.distributed as dist
+
+    from mpgcn_amd.graph import build_supports
+    from mpgcn_amd.models import MPGCN
+    from mpgcn_amd.parallel import GradAllReducer, init_distributed
+
+    rank = int(os.environ.get("RANK", 0))
+    world = int(os.environ.get("WORLD_SIZE", 1))
+
+    if args.device:
+        device = args.device
+    elif torch.cuda.is_available():
+        device = f"cuda:{int(os.environ.get('LOCAL_RANK', 0))}"
+    else:
+        device = "cpu"
+    ctx = init_distributed(device)
+    if device.startswith("cuda"):
+        torch.cuda.set_device(device)
+    is_cuda = device.startswith("cuda")
+
+    torch.manual_seed(1234 + rank)
+    N, B, H, T = args.nodes, args.batch, args.hidden, args.obs_len
+    K_order = 2
+    kernel = "random_walk_diffusion"
+    S = K_order + 1
+    cdtype = torch.bfloat16 if (args.dtype == "bf16" and is_cuda) else torch.float32
+
+    model = MPGCN(M=2, K=S, input_dim=1, lstm_hidden_dim=H, lstm_num_layers=1,
+                  gcn_hidden_dim=H, gcn_num_layers=3, num_nodes=N,
+                  compute_dtype=cdtype).to(device)
+    opt = torch.optim.Adam(model.parameters(), lr=1e-4)
+    reducer = GradAllReducer(model, ctx)
+    criterion = torch.nn.MSELoss()
+
+    # device-resident synthetic OD pool (log1p-scale magnitudes), windowed by index
+    T_pool = 64
+    pool = torch.log1p(20.0 * torch.rand(T_pool, N, N, 1, device=device))
+    adj = (torch.rand(N, N, device=device) < 0.1).float()
+    G_static = build_supports(adj.unsqueeze(0), kernel, K_order).squeeze(0)
+    # raw day-of-week correlation graphs (support build runs per-step, timed)
+    O_dyn_raw = torch.rand(7, N, N, device=device)
+    D_dyn_raw = torch.rand(7, N, N, device=device)
+
+    def step(i: int):
+        g = (torch.arange(B, device=device) * 7 + i) % (T_pool - T - 1)
+        x = pool[g.unsqueeze(1) + torch.arange(T, device=device)]  # (B,T,N,N,1)
+        y = pool[(g + T).unsqueeze(1) + torch.arange(1, device=device)]
+        key = (g + T) % 7
+        G_o = build_supports(O_dyn_raw[key], kernel, K_order)
+        G_d = build_supports(D_dyn_raw[key], kernel, K_order)
+        y_pred = model(x, [G_static, (G_o, G_d)])
+        loss = criterion(y_pred, y)
+        opt.zero_grad(set_to_none=True)
+        loss.backward()
+        reducer.finalize()
+        opt.step()
+        return loss
+
+    def barrier_sync():
+        if ctx.enabled:
+            dist.barrier()
+        if is_cuda:
+            torch.cuda.synchronize()
+
+    for i in range(args.warmup):
+        step(i)
+    barrier_sync()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        loss = step(args.warmup + i)
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks (slowest rank defines job time)
+    if ctx.enabled:
+        t = torch.tensor([elapsed], device=device if is_cuda else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = t.item()
+
+    n_gpus = world if world > 1 else (1 if is_cuda else args.gpus)
+    global_batch = B * max(world, 1)
+    samples_per_sec = global_batch * args.steps / elapsed
+    if rank == 0:
+        print(json.dumps({
+            "metric": "train_samples_per_sec",
+            "value": round(samples_per_sec, 2),
+            "unit": "samples/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if cdtype == torch.bfloat16 else "float32",
+            "data": "synthetic",
+            "config": {
+                "model": "MPGCN",
+                "regions": N,
+                "global_batch": global_batch,
+                "seq_len": T,
+                "hidden": H,
+                "kernel": kernel,
+                "supports_K": S,
+                "gcn_layers": 3,
+                "branches": 2,
+                "parallelism": f"dp{max(world, 1)}",
+                "final_loss": round(float(loss.item()), 5),
+            },
+        }))
+    if ctx.enabled:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

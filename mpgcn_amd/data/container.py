@@ -1,0 +1,184 @@
+"""OD data containers: ingest, normalization, windowing, batching.
+
+Capability parity with the reference data layer (Data_Container_OD.py), with the
+MI355X-native deltas called out in SURVEY.md §7:
+  * windows are INDEXED, never materialized as copies (the reference's
+    get_feats at Data_Container_OD.py:158-163 copies obs_len slices per sample);
+  * the whole OD tensor lives device-resident once (288 GB HBM3E budget) and
+    batch gathers happen on device;
+  * the dynamic day-of-week graphs are built with batched GEMMs
+    (mpgcn_amd/graph/dynamic.py), not the O(7 N^3) scipy loop.
+
+The npz path keeps the reference's exact file contract: sparse
+`od_day20180101_20210228.npz` densified to (-1, 47, 47), last 425 days,
+log1p transform, `adjacency_matrix.npy` (Data_Container_OD.py:15-19,34).
+"""
+
+from __future__ import annotations
+
+import numpy as np
+import torch
+
+from mpgcn_amd.data.synthetic import synthetic_adjacency, synthetic_od
+from mpgcn_amd.graph.dynamic import construct_dynamic_graphs
+
+
+class DataInput:
+    """Loads (or synthesizes) the OD dataset dict:
+    {'OD': (T,N,N,1) log1p-transformed, 'adj': (N,N),
+     'O_dyn_G': (N,N,7), 'D_dyn_G': (N,N,7)} — torch tensors."""
+
+    def __init__(self, params: dict):
+        self.params = params
+        self._max = self._min = self._mean = self._std = None
+
+    def load_data(self) -> dict:
+        p = self.params
+        if p.get("synthetic_nodes"):
+            N = int(p["synthetic_nodes"])
+            T = int(p.get("synthetic_days", 425))
+            raw = synthetic_od(T, N, seed=int(p.get("seed", 0)))
+            adj = synthetic_adjacency(N, seed=int(p.get("seed", 0)))
+        else:
+            import scipy.sparse as ss
+
+            sp = ss.load_npz(p["input_dir"] + "/od_day20180101_20210228.npz")
+            dense = np.asarray(sp.todense()).reshape((-1, 47, 47))
+            dense = dense[-425:, :, :, np.newaxis]  # last 425 days
+            raw = torch.from_numpy(dense).float()
+            adj = torch.from_numpy(
+                np.load(p["input_dir"] + "/adjacency_matrix.npy")
+            ).float()
+
+        OD = torch.log(raw + 1.0)  # log1p transform (Data_Container_OD.py:19)
+        norm = p.get("norm", "none")
+        if norm == "minmax":
+            OD = self.minmax_normalize(OD)
+        elif norm == "std":
+            OD = self.std_normalize(OD)
+        elif norm != "none":
+            raise ValueError(f"unknown norm {norm!r}")
+
+        # dynamic graphs from UN-normalized data over whole weeks of the train
+        # split (Data_Container_OD.py:35,39-42)
+        split = p["split_ratio"]
+        train_len = int(raw.shape[0] * split[0] / sum(split))
+        period = 7
+        whole = (train_len // period) * period
+        O_dyn, D_dyn = construct_dynamic_graphs(raw[:whole], period=period)
+
+        return {"OD": OD, "adj": adj, "O_dyn_G": O_dyn, "D_dyn_G": D_dyn}
+
+    # -- normalization (stored-stat transforms, Data_Container_OD.py:61-79) --
+    def minmax_normalize(self, x):
+        self._max, self._min = x.max().item(), x.min().item()
+        return (x - self._min) / (self._max - self._min)
+
+    def minmax_denormalize(self, x):
+        return (self._max - self._min) * x + self._min
+
+    def std_normalize(self, x):
+        self._mean, self._std = x.mean().item(), x.std().item()
+        return (x - self._mean) / self._std
+
+    def std_denormalize(self, x):
+        return x * self._std + self._mean
+
+
+class ODBatchIterator:
+    """Device-resident windowed batch iterator for one mode.
+
+    Yields (x_seq (B,obs,N,N,1), y (B,pred,N,N,1), O_dyn (B,N,N), D_dyn (B,N,N))
+    — the reference DataLoader's per-batch contract (Data_Container_OD.py:93-95).
+    Windows are index gathers into the single resident OD tensor.
+    """
+
+    def __init__(self, OD, O_dyn, D_dyn, start: int, length: int, obs_len: int,
+                 pred_len: int, batch_size: int, shuffle: bool = False,
+                 seed: int = 0, rank: int = 0, world_size: int = 1):
+        self.OD = OD
+        # (N,N,7) -> (7,N,N) for day-of-week row gathers
+        self.O_dyn = O_dyn.permute(2, 0, 1).contiguous()
+        self.D_dyn = D_dyn.permute(2, 0, 1).contiguous()
+        self.start = start
+        self.length = length
+        self.obs_len = obs_len
+        self.pred_len = pred_len
+        self.batch_size = batch_size
+        self.shuffle = shuffle
+        self.seed = seed
+        self.epoch = 0
+        self.rank = rank
+        self.world_size = world_size
+        self._offsets = torch.arange(obs_len, device=OD.device)
+        self._poffsets = torch.arange(obs_len, obs_len + pred_len, device=OD.device)
+
+    def set_epoch(self, epoch: int):
+        self.epoch = epoch
+
+    def __len__(self):
+        per_rank = self.length // self.world_size if self.world_size > 1 else self.length
+        return (per_rank + self.batch_size - 1) // self.batch_size
+
+    def __iter__(self):
+        idx = torch.arange(self.length, device=self.OD.device)
+        if self.shuffle:
+            g = torch.Generator(device="cpu").manual_seed(self.seed + self.epoch)
+            idx = idx[torch.randperm(self.length, generator=g).to(idx.device)]
+        if self.world_size > 1:
+            # contiguous per-rank shard, truncated to equal length across ranks
+            per_rank = self.length // self.world_size
+            idx = idx[self.rank * per_rank:(self.rank + 1) * per_rank]
+        for b0 in range(0, idx.numel(), self.batch_size):
+            b = idx[b0:b0 + self.batch_size]
+            g = b + self.start  # global sample index
+            x = self.OD[g.unsqueeze(1) + self._offsets]   # (B, obs, N, N, 1)
+            y = self.OD[g.unsqueeze(1) + self._poffsets]  # (B, pred, N, N, 1)
+            key = (g + self.obs_len) % 7                  # day-of-week of first target
+            yield x, y, self.O_dyn[key], self.D_dyn[key]
+
+
+class DataGenerator:
+    """Split bookkeeping + loader construction (Data_Container_OD.py:129-156)."""
+
+    def __init__(self, obs_len: int, pred_len: int, data_split_ratio):
+        self.obs_len = obs_len
+        self.pred_len = pred_len
+        self.data_split_ratio = data_split_ratio
+
+    def split2len(self, data_len: int) -> dict:
+        r = self.data_split_ratio
+        mode_len = {
+            "validate": int(r[1] / sum(r) * data_len),
+            "test": int(r[2] / sum(r) * data_len),
+        }
+        mode_len["train"] = data_len - mode_len["validate"] - mode_len["test"]
+        return mode_len
+
+    def get_data_loader(self, data: dict, params: dict, device="cpu",
+                        rank: int = 0, world_size: int = 1) -> dict:
+        OD = data["OD"].to(device)
+        O_dyn = data["O_dyn_G"].to(device)
+        D_dyn = data["D_dyn_G"].to(device)
+        T = OD.shape[0]
+        n_samples = T - self.obs_len - self.pred_len  # get_feats window count
+        mode_len = self.split2len(n_samples)
+
+        starts = {
+            "train": 0,
+            "validate": mode_len["train"],
+            "test": mode_len["train"] + mode_len["validate"],
+        }
+        loaders = {}
+        for mode in ("train", "validate", "test"):
+            loaders[mode] = ODBatchIterator(
+                OD, O_dyn, D_dyn,
+                start=starts[mode], length=mode_len[mode],
+                obs_len=self.obs_len, pred_len=self.pred_len,
+                batch_size=params["batch_size"],
+                shuffle=bool(params.get("shuffle", False)) and mode == "train",
+                seed=int(params.get("seed", 0)),
+                rank=rank if mode == "train" else 0,
+                world_size=world_size if mode == "train" else 1,
+            )
+        return loaders

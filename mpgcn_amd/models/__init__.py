@@ -1,0 +1,4 @@
+from mpgcn_amd.models.bdgcn import BDGCN
+from mpgcn_amd.models.mpgcn import MPGCN, TemporalEncoder
+
+__all__ = ["BDGCN", "MPGCN", "TemporalEncoder"]

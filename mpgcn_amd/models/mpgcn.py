@@ -1,0 +1,135 @@
+"""MPGCN: the multi-perspective ensemble model.
+
+Structure mirrors the reference (MPGCN.py:54-112): M parallel branches, each
+{temporal LSTM encoder -> gcn_num_layers stacked BDGCNs -> FC head}, fused by
+arithmetic mean, single-step output (batch, 1, N, N, 1).
+
+Checkpoint compatibility: the module tree reproduces the reference's state_dict
+key space exactly —
+    branch_models.{m}.temporal.{weight_ih_l0, weight_hh_l0, bias_ih_l0, bias_hh_l0}
+    branch_models.{m}.spatial.{n}.{W, b}
+    branch_models.{m}.fc.0.{weight, bias}
+so `{'epoch', 'state_dict'}` checkpoints round-trip between the two frameworks.
+
+MI355X execution: parameters are fp32 masters; `compute_dtype=torch.bfloat16`
+runs the forward/backward through the HIP kernels in bf16 with f32 MFMA
+accumulation (grads flow back into the fp32 masters through the cast nodes).
+"""
+
+from __future__ import annotations
+
+import math
+
+import torch
+from torch import nn
+
+from mpgcn_amd.models.bdgcn import BDGCN
+from mpgcn_amd.ops import GraphOperator, fused_lstm_last, linear_act
+
+
+class TemporalEncoder(nn.Module):
+    """1-layer batch-first LSTM over R = batch*N^2 scalar sequences, returning
+    the last hidden state only (the only timestep MPGCN consumes,
+    MPGCN.py:104). Parameter names/shapes/init match nn.LSTM(input, hidden,
+    num_layers=1, batch_first=True) for checkpoint compatibility."""
+
+    def __init__(self, input_size: int, hidden_size: int):
+        super().__init__()
+        if input_size != 1:
+            raise ValueError("TemporalEncoder supports input_size=1 (OD-flow scalar)")
+        self.input_size = input_size
+        self.hidden_size = hidden_size
+        self.weight_ih_l0 = nn.Parameter(torch.empty(4 * hidden_size, input_size))
+        self.weight_hh_l0 = nn.Parameter(torch.empty(4 * hidden_size, hidden_size))
+        self.bias_ih_l0 = nn.Parameter(torch.empty(4 * hidden_size))
+        self.bias_hh_l0 = nn.Parameter(torch.empty(4 * hidden_size))
+        self.reset_parameters()
+
+    def reset_parameters(self):
+        # nn.LSTM default: U(-1/sqrt(H), 1/sqrt(H)) on every parameter
+        stdv = 1.0 / math.sqrt(self.hidden_size)
+        for p in self.parameters():
+            nn.init.uniform_(p, -stdv, stdv)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        """x: (R, T) scalar sequences -> (R, hidden) last hidden state."""
+        w_ih = self.weight_ih_l0.to(x.dtype)
+        w_hh = self.weight_hh_l0.to(x.dtype)
+        return fused_lstm_last(x, w_ih, w_hh, self.bias_ih_l0, self.bias_hh_l0)
+
+
+class MPGCN(nn.Module):
+    def __init__(self, M: int, K: int, input_dim: int, lstm_hidden_dim: int,
+                 lstm_num_layers: int, gcn_hidden_dim: int, gcn_num_layers: int,
+                 num_nodes: int, user_bias: bool = True, activation: str = "relu",
+                 compute_dtype: torch.dtype = torch.float32):
+        super().__init__()
+        if lstm_num_layers != 1:
+            raise ValueError("MPGCN uses a 1-layer LSTM (Model_Trainer.py:50)")
+        self.M = M
+        self.K = K
+        self.num_nodes = num_nodes
+        self.lstm_hidden_dim = lstm_hidden_dim
+        self.gcn_num_layers = gcn_num_layers
+        self.compute_dtype = compute_dtype
+
+        self.branch_models = nn.ModuleList()
+        for _ in range(M):
+            branch = nn.ModuleDict()
+            branch["temporal"] = TemporalEncoder(input_dim, lstm_hidden_dim)
+            branch["spatial"] = nn.ModuleList()
+            for n in range(gcn_num_layers):
+                cur_in = lstm_hidden_dim if n == 0 else gcn_hidden_dim
+                branch["spatial"].append(
+                    BDGCN(K=K, input_dim=cur_in, hidden_dim=gcn_hidden_dim,
+                          use_bias=user_bias, activation=activation)
+                )
+            branch["fc"] = nn.Sequential(
+                nn.Linear(gcn_hidden_dim, input_dim, bias=True), nn.ReLU()
+            )
+            self.branch_models.append(branch)
+
+    def _graph_operators(self, G_list) -> list[GraphOperator]:
+        """G_list entries: a static (K, N, N) tensor (origin == destination
+        graph), or a (O_dyn, D_dyn) tuple of (B, K, N, N) dynamic supports —
+        the reference's contract (MPGCN.py:89-96)."""
+        if len(G_list) != self.M:
+            raise ValueError(f"expected {self.M} graph inputs, got {len(G_list)}")
+        gops = []
+        for G in G_list:
+            if isinstance(G, torch.Tensor):
+                Gc = G.to(self.compute_dtype)
+                gops.append(GraphOperator(Gc, Gc))
+            else:
+                Go, Gd = G
+                gops.append(GraphOperator(Go.to(self.compute_dtype),
+                                          Gd.to(self.compute_dtype)))
+        return gops
+
+    def forward(self, x_seq: torch.Tensor, G_list: list) -> torch.Tensor:
+        """x_seq: (B, seq, N, N, 1) -> (B, 1, N, N, 1)."""
+        assert x_seq.dim() == 5 and x_seq.shape[2] == x_seq.shape[3] == self.num_nodes
+        B, T, N = x_seq.shape[0], x_seq.shape[1], self.num_nodes
+        gops = self._graph_operators(G_list)
+
+        # (B, T, N, N, 1) -> (B*N*N, T) scalar sequences
+        lstm_in = (
+            x_seq.to(self.compute_dtype)
+            .permute(0, 2, 3, 1, 4)
+            .reshape(B * N * N, T)
+            .contiguous()
+        )
+        branch_out = []
+        for m in range(self.M):
+            branch = self.branch_models[m]
+            h_last = branch["temporal"](lstm_in)  # (B*N*N, H)
+            X = h_last.reshape(B, N, N, self.lstm_hidden_dim)
+            for layer in branch["spatial"]:
+                X = layer(X, gops[m])
+            fc = branch["fc"][0]
+            out = linear_act(
+                X.reshape(B * N * N, -1), fc.weight.to(X.dtype), fc.bias, relu=True
+            )
+            branch_out.append(out.view(B, N, N, 1))
+        ensemble = torch.mean(torch.stack(branch_out, dim=-1), dim=-1)
+        return ensemble.float().unsqueeze(1)

@@ -1,0 +1,262 @@
+"""Autograd-integrated ops: HIP kernels on GPU, eager PyTorch on CPU.
+
+The BDGCN layer uses the factored algorithm (mpgcn_amd/ops/eager.py docstring):
+    U = mode1(X, Go)            K origin-axis products        [axis_gemm kernel]
+    V = U_flat @ Wre            one flat projection GEMM      [row_gemm kernel]
+    H = act(mode2(V, Gd) + b)   K dest-axis products, fused   [axis_gemm kernel]
+
+Graph supports carry no gradient (they are built from input data each step,
+reference Model_Trainer.py:82-84,106), so backward only produces dX, dW, dbias:
+    dY  = dH * 1[H > 0]                       (ReLU mask, elementwise)
+    dV  = mode2_bwd(dY, A2)                   [axis_gemm]
+    dW  = Uflat^T @ dVflat  (reordered)       [rocBLAS reduction GEMM]
+    dU  = dVflat @ Wre^T                      [row_gemm]
+    dX  = mode1_bwd(dU, A3T)                  [axis_gemm]
+"""
+
+from __future__ import annotations
+
+import torch
+
+from mpgcn_amd import ops as _ops
+from mpgcn_amd.ops import eager
+
+_ROW_GEMM_MAX_N = 128
+
+
+class GraphOperator:
+    """Per-step container for one perspective's support stack in every layout
+    the kernels need. Layout permutes are tiny (graph tensors are O(S*N^2))
+    and computed lazily, once per training step, shared by all gcn layers and
+    by forward+backward.
+
+    Go/Gd: (S, N, N) static or (B, S, N, N) dynamic, in compute dtype.
+    """
+
+    def __init__(self, Go: torch.Tensor, Gd: torch.Tensor):
+        self.Go = Go
+        self.Gd = Gd
+        self.S = Go.shape[-3]
+        self.N = Go.shape[-1]
+        self._GoT = None
+        self._A2T = None
+        self._A2 = None
+        self._A3T = None
+
+    @property
+    def GoT(self) -> torch.Tensor:
+        """GT[..., m, n] = Go[..., n, m] — mode-1 A operand."""
+        if self._GoT is None:
+            self._GoT = self.Go.transpose(-2, -1).contiguous()
+        return self._GoT
+
+    @property
+    def A2T(self) -> torch.Tensor:
+        """A2T[d, c*S+s] = Gd[s, c, d] — mode-2 A operand, (N, N*S)."""
+        if self._A2T is None:
+            if self.Gd.dim() == 3:
+                self._A2T = self.Gd.permute(2, 1, 0).reshape(self.N, self.N * self.S).contiguous()
+            else:
+                B = self.Gd.shape[0]
+                self._A2T = self.Gd.permute(0, 3, 2, 1).reshape(B, self.N, self.N * self.S).contiguous()
+        return self._A2T
+
+    @property
+    def A2(self) -> torch.Tensor:
+        """A2[c*S+s, d] = Gd[s, c, d] — mode-2 backward A operand, (N*S, N)."""
+        if self._A2 is None:
+            if self.Gd.dim() == 3:
+                self._A2 = self.Gd.permute(1, 0, 2).reshape(self.N * self.S, self.N).contiguous()
+            else:
+                B = self.Gd.shape[0]
+                self._A2 = self.Gd.permute(0, 2, 1, 3).reshape(B, self.N * self.S, self.N).contiguous()
+        return self._A2
+
+    @property
+    def A3T(self) -> torch.Tensor:
+        """A3T[n, o*N+m] = Go[o, n, m] — mode-1 backward A operand, (N, S*N)."""
+        if self._A3T is None:
+            if self.Go.dim() == 3:
+                self._A3T = self.Go.permute(1, 0, 2).reshape(self.N, self.S * self.N).contiguous()
+            else:
+                B = self.Go.shape[0]
+                self._A3T = self.Go.permute(0, 2, 1, 3).reshape(B, self.N, self.S * self.N).contiguous()
+        return self._A3T
+
+
+def _row_gemm_chunked(ext, X2d, W, bias, relu):
+    """row_gemm with column chunking for N > 128 (e.g. dual-RWD S=5: S*H=160)."""
+    R, _ = X2d.shape
+    N = W.shape[1]
+    if N <= _ROW_GEMM_MAX_N:
+        return ext.row_gemm(X2d, W, bias, relu)
+    out = torch.empty(R, N, device=X2d.device, dtype=X2d.dtype)
+    for n0 in range(0, N, _ROW_GEMM_MAX_N):
+        n1 = min(n0 + _ROW_GEMM_MAX_N, N)
+        bchunk = bias[n0:n1].contiguous() if bias is not None else None
+        ext.row_gemm_out(X2d, W[:, n0:n1].contiguous(), bchunk, relu, out, N, n0)
+    return out
+
+
+class _BDGCNLayerFn(torch.autograd.Function):
+    """GPU path of one BDGCN layer via the HIP kernels."""
+
+    @staticmethod
+    def forward(ctx, X, W, bias, gop: GraphOperator, relu: bool):
+        ext = _ops.get_ext()
+        B, N = X.shape[0], X.shape[1]
+        C = X.shape[-1]
+        S = gop.S
+        Hdim = W.shape[1]
+
+        U = ext.bdgcn_mode1(X, gop.GoT)  # (B,N,N,S,C)
+        Wre = eager.reorder_projection_weight(W, S, C).contiguous()
+        Vflat = _row_gemm_chunked(ext, U.reshape(B * N * N, S * C), Wre, None, False)
+        bias_f32 = bias.float().contiguous() if bias is not None else None
+        Y = ext.bdgcn_mode2(Vflat.view(B, N, N * S, Hdim), gop.A2T, bias_f32, relu, N, S)
+
+        ctx.save_for_backward(U, Wre, Y)
+        ctx.gop = gop
+        ctx.relu = relu
+        ctx.has_bias = bias is not None
+        ctx.dims = (B, N, S, C, Hdim)
+        return Y
+
+    @staticmethod
+    def backward(ctx, dH):
+        ext = _ops.get_ext()
+        U, Wre, Y = ctx.saved_tensors
+        gop: GraphOperator = ctx.gop
+        B, N, S, C, Hdim = ctx.dims
+
+        dH = dH.contiguous()
+        dY = dH * (Y > 0).to(dH.dtype) if ctx.relu else dH
+        dbias = dY.sum(dim=(0, 1, 2)).to(torch.float32) if ctx.has_bias else None
+
+        dV = ext.bdgcn_mode2_bwd(dY, gop.A2, S)  # (B,N,N,S,H)
+        R = B * N * N
+        dVflat = dV.reshape(R, S * Hdim)
+        # dWre: plain library reduction GEMM over R rows (rocBLAS), f32 accum
+        dWre = U.reshape(R, S * C).t() @ dVflat
+        dW = dWre.view(S, C, S, Hdim).permute(0, 2, 1, 3).reshape(S * S * C, Hdim)
+        dU = _row_gemm_chunked(ext, dVflat, Wre.t().contiguous(), None, False)
+        dX = ext.bdgcn_mode1_bwd(dU.view(B, N, N, S, C), gop.A3T)
+        db = dbias if ctx.has_bias else None
+        return dX, dW, db, None, None
+
+
+def bdgcn_layer(X, W, bias, gop: GraphOperator, relu: bool = True):
+    """One 2-D GCN layer. X: (B,N,N,C); W: (C*S*S, H); bias: (H,) or None."""
+    if X.is_cuda:
+        return _BDGCNLayerFn.apply(X, W, bias, gop, relu)
+    return eager.bdgcn_layer_eager(X, gop.Go, gop.Gd, W, bias, "relu" if relu else "none")
+
+
+class _FusedLSTMLastFn(torch.autograd.Function):
+    """Single-layer LSTM over R sequences, returning only the LAST hidden state
+    (the only timestep MPGCN consumes, reference MPGCN.py:104). Input dim 1.
+
+    x: (R, T) compute-dtype; weights in torch nn.LSTM layout:
+    w_ih (4H, 1), w_hh (4H, H), b_ih (4H,), b_hh (4H,).
+    """
+
+    @staticmethod
+    def forward(ctx, x, w_ih, w_hh, b_ih, b_hh):
+        ext = _ops.get_ext()
+        R, T = x.shape
+        Hd = w_hh.shape[1]
+        dev = x.device
+        whh = w_hh.contiguous()
+        wih_f = w_ih.reshape(-1).float().contiguous()
+        bias_f = (b_ih.float() + b_hh.float()).contiguous()
+
+        h = torch.zeros(R, Hd, device=dev, dtype=x.dtype)
+        c = torch.zeros(R, Hd, device=dev, dtype=torch.float32)
+        hs, cs, gs = [h], [c], []
+        xc = x.contiguous()
+        for t in range(T):
+            h, c, gates = ext.lstm_step_fwd(xc, T, t, h, c, whh, wih_f, bias_f)
+            hs.append(h)
+            cs.append(c)
+            gs.append(gates)
+        ctx.saved_lists = (hs, cs, gs)
+        ctx.save_for_backward(xc, whh, wih_f)
+        ctx.T = T
+        return h
+
+    @staticmethod
+    def backward(ctx, dh_last):
+        ext = _ops.get_ext()
+        xc, whh, wih_f = ctx.saved_tensors
+        hs, cs, gs = ctx.saved_lists
+        T = ctx.T
+        R = xc.shape[0]
+        Hd = whh.shape[1]
+
+        dh = dh_last.contiguous()
+        dc = None
+        dwhh = torch.zeros(4 * Hd, Hd, device=xc.device, dtype=torch.float32)
+        dbias = torch.zeros(4 * Hd, device=xc.device, dtype=torch.float32)
+        dwih = torch.zeros(4 * Hd, device=xc.device, dtype=torch.float32)
+        dx = torch.empty_like(xc)
+        wih_c = wih_f.to(xc.dtype).view(-1, 1).contiguous()
+        for t in range(T - 1, -1, -1):
+            dgates, dc = ext.lstm_step_bwd(dh, dc, gs[t], cs[t], cs[t + 1])
+            # weight grads: plain library reduction GEMMs (rocBLAS), f32 accum
+            dwhh += (dgates.t() @ hs[t]).float()
+            dbias += dgates.sum(dim=0).float()
+            dwih += (dgates.t() @ xc[:, t].unsqueeze(-1)).squeeze(-1).float()
+            dh = ext.row_gemm(dgates, whh, None, False)  # dgates @ W_hh -> dh_prev
+            dx[:, t] = ext.row_gemm(dgates, wih_c, None, False).view(-1)
+        ctx.saved_lists = None
+        wdt = whh.dtype
+        return (
+            dx,
+            dwih.view(-1, 1).to(wdt),
+            dwhh.to(wdt),
+            dbias,
+            dbias.clone(),
+        )
+
+
+def fused_lstm_last(x, w_ih, w_hh, b_ih, b_hh):
+    """Last hidden state of a 1-layer batch-first LSTM over (R, T) scalar inputs."""
+    Hd = w_hh.shape[1]
+    kernel_ok = Hd == 32 or (Hd == 16 and x.dtype == torch.float32)
+    if x.is_cuda and kernel_ok:
+        return _FusedLSTMLastFn.apply(x, w_ih, w_hh, b_ih, b_hh)
+    out, _, _ = eager.lstm_forward_eager(x.unsqueeze(-1), w_ih, w_hh, b_ih, b_hh)
+    return out[:, -1, :]
+
+
+class _LinearActFn(torch.autograd.Function):
+    """Fused Linear(+bias)+ReLU head via row_gemm; backward in plain torch
+    (the FC head is (R,32)->(R,1), a trivial fraction of step time)."""
+
+    @staticmethod
+    def forward(ctx, X2d, weight, bias, relu):
+        ext = _ops.get_ext()
+        Wt = weight.t().contiguous()  # (in, out)
+        bias_f = bias.float().contiguous() if bias is not None else None
+        out = _row_gemm_chunked(ext, X2d, Wt, bias_f, relu)
+        ctx.save_for_backward(X2d, weight, out)
+        ctx.relu = relu
+        ctx.has_bias = bias is not None
+        return out
+
+    @staticmethod
+    def backward(ctx, dOut):
+        X2d, weight, out = ctx.saved_tensors
+        dY = dOut * (out > 0).to(dOut.dtype) if ctx.relu else dOut
+        dX = dY @ weight.to(dY.dtype)
+        dW = dY.t() @ X2d
+        db = dY.sum(0).to(torch.float32) if ctx.has_bias else None
+        return dX, dW, db, None
+
+
+def linear_act(X2d, weight, bias, relu: bool = True):
+    """act(X @ weight^T + bias) with torch nn.Linear weight layout (out, in)."""
+    if X2d.is_cuda:
+        return _LinearActFn.apply(X2d, weight, bias, relu)
+    out = torch.nn.functional.linear(X2d, weight, bias)
+    return torch.relu(out) if relu else out

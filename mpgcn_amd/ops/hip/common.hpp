@@ -1,0 +1,68 @@
+// Shared helpers for the MPGCN-MI355X HIP kernels (gfx950 / CDNA4 only).
+//
+// Wavefront is 64-wide; MFMA tiles are 16x16 with K=32 (bf16, f32 accumulate)
+// or K=4 (exact f32). No CUDA-compat paths, no other archs.
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#define WAVE 64
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+// ---------------------------------------------------------------------------
+// MFMA traits: one 16x16 output fragment per wave.
+//   bf16: v_mfma_f32_16x16x32_bf16 — lane l holds A[row=l&15][k=(l>>4)*8 + j],
+//         B[k=(l>>4)*8 + j][col=l&15], j = 0..7 (one bf16x8 = 4 VGPRs each).
+//   f32:  v_mfma_f32_16x16x4_f32  — lane l holds A[l&15][l>>4], B[l>>4][l&15].
+//   C/D (both): col = l&15, row = (l>>4)*4 + reg, reg = 0..3 (f32x4).
+// ---------------------------------------------------------------------------
+template <typename T>
+struct MfmaTraits;
+
+template <>
+struct MfmaTraits<__bf16> {
+    static constexpr int MFMA_K = 32;     // k-depth of one MFMA instruction
+    static constexpr int FRAG_ELEMS = 8;  // per-lane A/B elements (contiguous in k)
+    static constexpr int LDS_PAD = 8;     // pad elements per LDS row (16 B)
+    using frag_t = bf16x8;
+    __device__ static inline f32x4 mfma(frag_t a, frag_t b, f32x4 c) {
+        return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+    }
+};
+
+template <>
+struct MfmaTraits<float> {
+    static constexpr int MFMA_K = 4;
+    static constexpr int FRAG_ELEMS = 1;
+    static constexpr int LDS_PAD = 4;  // 16 B
+    using frag_t = float;
+    __device__ static inline f32x4 mfma(frag_t a, frag_t b, f32x4 c) {
+        return __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, c, 0, 0, 0);
+    }
+};
+
+template <typename T>
+__device__ __forceinline__ T from_f32(float v);
+template <>
+__device__ __forceinline__ __bf16 from_f32<__bf16>(float v) { return (__bf16)v; }
+template <>
+__device__ __forceinline__ float from_f32<float>(float v) { return v; }
+
+__device__ __forceinline__ float to_f32(__bf16 v) { return (float)v; }
+__device__ __forceinline__ float to_f32(float v) { return v; }
+
+// 16-byte raw copy chunk (8 bf16 / 4 f32).
+struct alignas(16) Chunk16 { int v[4]; };
+
+#define HIP_CHECK(expr)                                              \
+    do {                                                             \
+        hipError_t _e = (expr);                                      \
+        if (_e != hipSuccess) {                                      \
+            fprintf(stderr, "HIP error %s at %s:%d\n",               \
+                    hipGetErrorString(_e), __FILE__, __LINE__);      \
+            abort();                                                 \
+        }                                                            \
+    } while (0)

@@ -1,0 +1,275 @@
+// Torch bindings for the MPGCN-MI355X HIP kernels (gfx950 only).
+//
+// Shape/layout contracts are documented per function; the Python wrappers in
+// mpgcn_amd/ops/functional.py pre-permute the (tiny) graph operands into the
+// row-major (M, K) "AT" layouts the axis_gemm kernel consumes.
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+
+#include "params.hpp"
+
+namespace {
+
+void check_in(const torch::Tensor& t, const char* name) {
+    TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+    TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+    TORCH_CHECK(t.scalar_type() == torch::kBFloat16 || t.scalar_type() == torch::kFloat,
+                name, " must be bf16 or f32");
+}
+
+int is_f32(const torch::Tensor& t) { return t.scalar_type() == torch::kFloat ? 1 : 0; }
+
+int chunk_elems(const torch::Tensor& t) { return is_f32(t) ? 4 : 8; }
+
+hipStream_t stream() { return c10::hip::getCurrentHIPStream().stream(); }
+
+const float* bias_ptr(const c10::optional<torch::Tensor>& b) {
+    if (!b.has_value() || !b->defined()) return nullptr;
+    TORCH_CHECK(b->scalar_type() == torch::kFloat, "bias must be f32");
+    TORCH_CHECK(b->is_contiguous(), "bias must be contiguous");
+    return b->data_ptr<float>();
+}
+
+}  // namespace
+
+// U[b,m,d,o,l] = sum_n GT[(b,)o,m,n] X[b,n,d,l].
+// X: (B, N, N, C); GT: (S, N, N) static or (B, S, N, N) dynamic, ALREADY
+// transposed per support (GT[..., m, n] = G[..., n, m]). Out: (B, N, N, S, C).
+torch::Tensor bdgcn_mode1(torch::Tensor X, torch::Tensor GT) {
+    check_in(X, "X");
+    check_in(GT, "GT");
+    const bool dyn = GT.dim() == 4;
+    const long B = X.size(0), N = X.size(1), C = X.size(3);
+    const long S = dyn ? GT.size(1) : GT.size(0);
+    TORCH_CHECK(X.size(2) == N && GT.size(-1) == N && GT.size(-2) == N, "shape mismatch");
+    TORCH_CHECK(!dyn || GT.size(0) == B, "dynamic GT batch mismatch");
+    TORCH_CHECK(B * S <= 65535, "too many instances");
+    auto U = torch::empty({B, N, N, S, C}, X.options());
+
+    AxisGemmParams p{};
+    p.AT = GT.data_ptr();
+    p.X = X.data_ptr();
+    p.OUT = U.data_ptr();
+    p.M = (int)N; p.K = (int)N; p.L = (int)(N * C);
+    p.a_div = (int)S; p.a_bs1 = dyn ? S * N * N : 0; p.a_bs2 = N * N;
+    p.x_div = (int)S; p.x_bs1 = N * N * C; p.x_bs2 = 0;
+    p.o_div = (int)S; p.o_bs1 = N * N * S * C; p.o_bs2 = C;
+    p.kdiv = 1; p.k_lo = N * C;
+    p.qdiv = 0;
+    p.o_row = N * S * C;
+    p.ogdiv = (int)C; p.og_hi = S * C;
+    const int ch = chunk_elems(X);
+    p.a_vec = (N % ch == 0);
+    p.x_vec = ((N * C) % ch == 0) && (C % ch == 0);
+    axis_gemm_launch(p, (int)(B * S), is_f32(X), stream());
+    return U;
+}
+
+// Y[b,m,d,h] = sum_{cs} A2T[(b,)d,cs] V[b,m,cs,h]  (+ bias + optional ReLU).
+// V: (B, N, N*S, H) flat (i.e. (B,N,N,S,H) contiguous); A2T: (N, N*S) static
+// or (B, N, N*S) dynamic, A2T[d, c*S+s] = Gd[s, c, d]. Out: (B, N, N, H).
+torch::Tensor bdgcn_mode2(torch::Tensor V, torch::Tensor A2T,
+                          c10::optional<torch::Tensor> bias, bool relu,
+                          long N, long S) {
+    check_in(V, "V");
+    check_in(A2T, "A2T");
+    const bool dyn = A2T.dim() == 3;
+    const long B = V.size(0);
+    const long H = V.size(-1);
+    TORCH_CHECK(A2T.size(-1) == N * S && A2T.size(-2) == N, "A2T shape mismatch");
+    TORCH_CHECK(B * N <= 65535, "too many instances");
+    auto Y = torch::empty({B, N, N, H}, V.options());
+
+    AxisGemmParams p{};
+    p.AT = A2T.data_ptr();
+    p.X = V.data_ptr();
+    p.OUT = Y.data_ptr();
+    p.bias = bias_ptr(bias);
+    p.M = (int)N; p.K = (int)(N * S); p.L = (int)H;
+    p.a_div = dyn ? (int)N : 1; p.a_bs1 = dyn ? N * N * S : 0; p.a_bs2 = 0;
+    p.x_div = 1; p.x_bs1 = N * S * H; p.x_bs2 = 0;
+    p.o_div = 1; p.o_bs1 = N * H; p.o_bs2 = 0;
+    p.kdiv = 1; p.k_lo = H;
+    p.qdiv = 0;
+    p.o_row = H;
+    p.ogdiv = 0;
+    p.relu = relu ? 1 : 0;
+    const int ch = chunk_elems(V);
+    p.a_vec = ((N * S) % ch == 0);
+    p.x_vec = (H % ch == 0);
+    axis_gemm_launch(p, (int)(B * N), is_f32(V), stream());
+    return Y;
+}
+
+// dV[b,m,cs,h] = sum_d A2[(b,)cs,d] dY[b,m,d,h].
+// dY: (B, N, N, H); A2: (N*S, N) or (B, N*S, N), A2[c*S+s, d] = Gd[s, c, d].
+torch::Tensor bdgcn_mode2_bwd(torch::Tensor dY, torch::Tensor A2, long S) {
+    check_in(dY, "dY");
+    check_in(A2, "A2");
+    const bool dyn = A2.dim() == 3;
+    const long B = dY.size(0), N = dY.size(1), H = dY.size(3);
+    TORCH_CHECK(A2.size(-2) == N * S && A2.size(-1) == N, "A2 shape mismatch");
+    TORCH_CHECK(B * N <= 65535, "too many instances");
+    auto dV = torch::empty({B, N, N, S, H}, dY.options());
+
+    AxisGemmParams p{};
+    p.AT = A2.data_ptr();
+    p.X = dY.data_ptr();
+    p.OUT = dV.data_ptr();
+    p.M = (int)(N * S); p.K = (int)N; p.L = (int)H;
+    p.a_div = dyn ? (int)N : 1; p.a_bs1 = dyn ? N * S * N : 0; p.a_bs2 = 0;
+    p.x_div = 1; p.x_bs1 = N * H; p.x_bs2 = 0;
+    p.o_div = 1; p.o_bs1 = N * S * H; p.o_bs2 = 0;
+    p.kdiv = 1; p.k_lo = H;
+    p.qdiv = 0;
+    p.o_row = H;
+    p.ogdiv = 0;
+    const int ch = chunk_elems(dY);
+    p.a_vec = (N % ch == 0);
+    p.x_vec = (H % ch == 0);
+    axis_gemm_launch(p, (int)(B * N), is_f32(dY), stream());
+    return dV;
+}
+
+// dX[b,n,d,l] = sum_{om} A3T[(b,)n,o*N+m] dU[b,m,d,o,l].
+// dU: (B, N, N, S, C); A3T: (N, S*N) or (B, N, S*N), A3T[n, o*N+m] = G[o,n,m].
+torch::Tensor bdgcn_mode1_bwd(torch::Tensor dU, torch::Tensor A3T) {
+    check_in(dU, "dU");
+    check_in(A3T, "A3T");
+    const bool dyn = A3T.dim() == 3;
+    const long B = dU.size(0), N = dU.size(1), S = dU.size(3), C = dU.size(4);
+    TORCH_CHECK(A3T.size(-2) == N && A3T.size(-1) == S * N, "A3T shape mismatch");
+    TORCH_CHECK(B <= 65535, "too many instances");
+    auto dX = torch::empty({B, N, N, C}, dU.options());
+
+    AxisGemmParams p{};
+    p.AT = A3T.data_ptr();
+    p.X = dU.data_ptr();
+    p.OUT = dX.data_ptr();
+    p.M = (int)N; p.K = (int)(S * N); p.L = (int)(N * C);
+    p.a_div = 1; p.a_bs1 = dyn ? N * S * N : 0; p.a_bs2 = 0;
+    p.x_div = 1; p.x_bs1 = N * N * S * C; p.x_bs2 = 0;
+    p.o_div = 1; p.o_bs1 = N * N * C; p.o_bs2 = 0;
+    p.kdiv = (int)N; p.k_hi = C; p.k_lo = N * S * C;  // k = o*N + m
+    p.qdiv = (int)C; p.q_hi = S * C;                  // q = d*C + l
+    p.o_row = N * C;
+    p.ogdiv = 0;
+    const int ch = chunk_elems(dU);
+    p.a_vec = ((S * N) % ch == 0);
+    p.x_vec = (C % ch == 0);
+    axis_gemm_launch(p, (int)B, is_f32(dU), stream());
+    return dX;
+}
+
+// OUT[r, n] = act(X[r, :] @ W + bias). X: (R, K), W: (K, N), N <= 128.
+torch::Tensor row_gemm(torch::Tensor X, torch::Tensor W,
+                       c10::optional<torch::Tensor> bias, bool relu) {
+    check_in(X, "X");
+    check_in(W, "W");
+    const long R = X.size(0), K = X.size(1), N = W.size(1);
+    TORCH_CHECK(W.size(0) == K, "W shape mismatch");
+    TORCH_CHECK(N <= 128, "row_gemm: N must be <= 128");
+    TORCH_CHECK(K <= 2048, "row_gemm: K too large for LDS staging");
+    auto OUT = torch::empty({R, N}, X.options());
+    RowGemmParams p{};
+    p.X = X.data_ptr();
+    p.W = W.data_ptr();
+    p.OUT = OUT.data_ptr();
+    p.bias = bias_ptr(bias);
+    p.R = R; p.K = (int)K; p.N = (int)N;
+    p.o_row = N; p.o_off = 0;
+    p.relu = relu ? 1 : 0;
+    p.x_vec = (K % chunk_elems(X) == 0);
+    row_gemm_launch(p, is_f32(X), stream());
+    return OUT;
+}
+
+// Same as row_gemm but writes into a caller-provided flat 2-D buffer at
+// OUT[r * o_row + o_off + n] — lets Python chunk wide projections (N > 128)
+// into column slices of one contiguous output with no extra copy.
+void row_gemm_out(torch::Tensor X, torch::Tensor W,
+                  c10::optional<torch::Tensor> bias, bool relu,
+                  torch::Tensor OUT, long o_row, long o_off) {
+    check_in(X, "X");
+    check_in(W, "W");
+    check_in(OUT, "OUT");
+    const long R = X.size(0), K = X.size(1), N = W.size(1);
+    TORCH_CHECK(W.size(0) == K, "W shape mismatch");
+    TORCH_CHECK(N <= 128, "row_gemm: N must be <= 128");
+    TORCH_CHECK(K <= 2048, "row_gemm: K too large for LDS staging");
+    RowGemmParams p{};
+    p.X = X.data_ptr();
+    p.W = W.data_ptr();
+    p.OUT = OUT.data_ptr();
+    p.bias = bias_ptr(bias);
+    p.R = R; p.K = (int)K; p.N = (int)N;
+    p.o_row = o_row; p.o_off = o_off;
+    p.relu = relu ? 1 : 0;
+    p.x_vec = (K % chunk_elems(X) == 0);
+    row_gemm_launch(p, is_f32(X), stream());
+}
+
+// One fused LSTM step. x: any tensor whose flat layout gives the step input at
+// x[r * x_stride + x_off]; h_prev: (R, H); c_prev: (R, H) f32; whh: (4H, H);
+// wih, bias: (4H) f32. Returns (h, c, gates_postact).
+std::vector<torch::Tensor> lstm_step_fwd(torch::Tensor x, long x_stride, long x_off,
+                                         torch::Tensor h_prev, torch::Tensor c_prev,
+                                         torch::Tensor whh, torch::Tensor wih,
+                                         torch::Tensor bias) {
+    check_in(x, "x");
+    check_in(h_prev, "h_prev");
+    check_in(whh, "whh");
+    TORCH_CHECK(c_prev.scalar_type() == torch::kFloat && c_prev.is_contiguous());
+    TORCH_CHECK(wih.scalar_type() == torch::kFloat && bias.scalar_type() == torch::kFloat);
+    const long R = h_prev.size(0), H = h_prev.size(1);
+    auto h = torch::empty_like(h_prev);
+    auto c = torch::empty_like(c_prev);
+    auto gates = torch::empty({R, 4 * H}, h_prev.options());
+    LstmStepParams p{};
+    p.x = x.data_ptr(); p.x_stride = x_stride; p.x_off = x_off;
+    p.h_prev = h_prev.data_ptr();
+    p.c_prev = c_prev.data_ptr<float>();
+    p.whh = whh.data_ptr();
+    p.wih = wih.data_ptr<float>();
+    p.bias = bias.data_ptr<float>();
+    p.h_out = h.data_ptr();
+    p.c_out = c.data_ptr<float>();
+    p.gates_out = gates.data_ptr();
+    p.R = R; p.H = (int)H;
+    lstm_step_fwd_launch(p, is_f32(h_prev), stream());
+    return {h, c, gates};
+}
+
+// One LSTM backward step (pointwise part). Returns (dgates_preact, dc_prev).
+std::vector<torch::Tensor> lstm_step_bwd(torch::Tensor dh,
+                                         c10::optional<torch::Tensor> dc_in,
+                                         torch::Tensor gates, torch::Tensor c_prev,
+                                         torch::Tensor c) {
+    check_in(dh, "dh");
+    check_in(gates, "gates");
+    const long R = dh.size(0), H = dh.size(1);
+    auto dgates = torch::empty({R, 4 * H}, dh.options());
+    auto dc_prev = torch::empty({R, H}, c.options());
+    LstmBwdParams p{};
+    p.dh = dh.data_ptr();
+    p.dc_in = (dc_in.has_value() && dc_in->defined()) ? dc_in->data_ptr<float>() : nullptr;
+    p.gates = gates.data_ptr();
+    p.c_prev = c_prev.data_ptr<float>();
+    p.c = c.data_ptr<float>();
+    p.dgates = dgates.data_ptr();
+    p.dc_prev = dc_prev.data_ptr<float>();
+    p.R = R; p.H = (int)H;
+    lstm_step_bwd_launch(p, is_f32(dh), stream());
+    return {dgates, dc_prev};
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("bdgcn_mode1", &bdgcn_mode1, "origin-axis graph product (K1)");
+    m.def("bdgcn_mode2", &bdgcn_mode2, "dest-axis graph product + bias + act (K2)");
+    m.def("bdgcn_mode2_bwd", &bdgcn_mode2_bwd, "backward dV of mode2");
+    m.def("bdgcn_mode1_bwd", &bdgcn_mode1_bwd, "backward dX of mode1");
+    m.def("row_gemm", &row_gemm, "fused row GEMM + bias + act (K3/K6)");
+    m.def("row_gemm_out", &row_gemm_out, "row GEMM into strided output slice");
+    m.def("lstm_step_fwd", &lstm_step_fwd, "fused LSTM cell forward step (K4)");
+    m.def("lstm_step_bwd", &lstm_step_bwd, "LSTM cell backward pointwise step");
+}

@@ -1,0 +1,69 @@
+// Launch-parameter structs + C launcher declarations shared between the HIP
+// kernel translation units and the torch binding (ext.hip).
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+struct AxisGemmParams {
+    const void* AT;
+    const void* X;
+    void* OUT;
+    const float* bias;
+    int M, K, L;
+    int a_div; long a_bs1, a_bs2;
+    int x_div; long x_bs1, x_bs2;
+    int o_div; long o_bs1, o_bs2;
+    int kdiv; long k_hi, k_lo;
+    int qdiv; long q_hi;
+    long o_row;
+    int ogdiv; long og_hi;
+    int relu;
+    int a_vec, x_vec;
+    int tiles_l;
+};
+
+struct RowGemmParams {
+    const void* X;
+    const void* W;
+    void* OUT;
+    const float* bias;
+    long R;
+    int K, N;
+    long o_row, o_off;  // OUT element index = m * o_row + o_off + n
+    int relu;
+    int x_vec;
+};
+
+struct LstmStepParams {
+    const void* x;
+    long x_stride, x_off;
+    const void* h_prev;
+    const float* c_prev;
+    const void* whh;
+    const float* wih;
+    const float* bias;
+    void* h_out;
+    float* c_out;
+    void* gates_out;
+    long R;
+    int H;
+};
+
+struct LstmBwdParams {
+    const void* dh;
+    const float* dc_in;
+    const void* gates;
+    const float* c_prev;
+    const float* c;
+    void* dgates;
+    float* dc_prev;
+    long R;
+    int H;
+};
+
+extern "C" {
+void axis_gemm_launch(AxisGemmParams p, int instances, int is_f32, hipStream_t s);
+void row_gemm_launch(RowGemmParams p, int is_f32, hipStream_t s);
+void lstm_step_fwd_launch(LstmStepParams p, int is_f32, hipStream_t s);
+void lstm_step_bwd_launch(LstmBwdParams p, int is_f32, hipStream_t s);
+}

@@ -1,0 +1,98 @@
+// row_gemm — fused row-streaming GEMM: OUT[R, N] = act(X[R, K] @ W[K, N] + bias).
+//
+// For the memory-bound "many tiny rows" GEMMs of the MPGCN stack: the BDGCN
+// projection V = U_flat @ Wre (R = b*N^2 rows, K = S*C, N = S*H; reference op
+// MPGCN.py:44-49), its backward dU = dV @ Wre^T, the FC head Linear+ReLU
+// (MPGCN.py:74-76,107), and the LSTM backward dh_prev = dgates @ W_hh.
+//
+// W is staged once per block into LDS transposed ([n][k]-major) so B-fragments
+// are contiguous reads; X rows feed A-fragments straight from global (each row
+// is a short contiguous K-vector). Grid-stride loop over 64-row tiles.
+#include "common.hpp"
+#include "params.hpp"
+
+// 4 waves * 16 rows = 64 rows per block; NF16 = ceil(N/16) column fragments.
+template <typename T>
+__launch_bounds__(256) __global__ void row_gemm_kernel(RowGemmParams p, int nf16) {
+    using MT = MfmaTraits<T>;
+    constexpr int CH = 16 / sizeof(T);
+    const int KP = p.K + MT::LDS_PAD;
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    T* ldsWT = (T*)smem;  // [nf16*16][K + PAD]
+
+    const T* __restrict__ X = (const T*)p.X;
+    const T* __restrict__ W = (const T*)p.W;
+    T* __restrict__ O = (T*)p.OUT;
+
+    const int tid = threadIdx.x;
+    // ---- stage W transposed: ldsWT[n][k] = W[k][n] (one time per block) ----
+    for (int idx = tid; idx < p.K * p.N; idx += 256) {
+        const int k = idx / p.N, n = idx % p.N;
+        ldsWT[n * KP + k] = W[(long)k * p.N + n];
+    }
+    // zero-pad the tail fragment rows (n >= N) and k tail
+    for (int idx = tid; idx < nf16 * 16 * KP; idx += 256) {
+        const int n = idx / KP, k = idx % KP;
+        if (n >= p.N || k >= p.K) ldsWT[n * KP + k] = (T)0.f;
+    }
+    __syncthreads();
+
+    const int w = tid / WAVE, lane = tid % WAVE;
+    const int lrow = lane & 15, kgrp = lane >> 4;
+    const int kfrags = (p.K + MT::MFMA_K - 1) / MT::MFMA_K;
+
+    for (long r0 = (long)blockIdx.x * 64 + w * 16; r0 < p.R;
+         r0 += (long)gridDim.x * 64) {
+        const long row = r0 + lrow;
+        const bool row_ok = row < p.R;
+        f32x4 acc[8] = {};  // up to 8 column fragments (N <= 128)
+
+        for (int kf = 0; kf < kfrags; ++kf) {
+            const int k = kf * MT::MFMA_K + kgrp * MT::FRAG_ELEMS;
+            typename MT::frag_t af;
+            if (row_ok && k + MT::FRAG_ELEMS <= p.K && p.x_vec) {
+                af = *(const typename MT::frag_t*)&X[row * p.K + k];
+            } else {
+                alignas(16) T tmp[MT::FRAG_ELEMS];
+#pragma unroll
+                for (int i = 0; i < MT::FRAG_ELEMS; ++i)
+                    tmp[i] = (row_ok && k + i < p.K) ? X[row * p.K + k + i] : (T)0.f;
+                af = *(const typename MT::frag_t*)tmp;
+            }
+            for (int nf = 0; nf < nf16; ++nf) {
+                const typename MT::frag_t bf = *(const typename MT::frag_t*)
+                    &ldsWT[(nf * 16 + lrow) * KP + kf * MT::MFMA_K +
+                           kgrp * MT::FRAG_ELEMS];
+                acc[nf] = MT::mfma(af, bf, acc[nf]);
+            }
+        }
+
+        for (int nf = 0; nf < nf16; ++nf) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const long m = r0 + kgrp * 4 + r;
+                const int n = nf * 16 + lrow;
+                if (m < p.R && n < p.N) {
+                    float v = acc[nf][r];
+                    if (p.bias) v += p.bias[n];
+                    if (p.relu) v = fmaxf(v, 0.f);
+                    O[m * p.o_row + p.o_off + n] = from_f32<T>(v);
+                }
+            }
+        }
+    }
+}
+
+extern "C" void row_gemm_launch(RowGemmParams p, int is_f32, hipStream_t stream) {
+    const int nf16 = (p.N + 15) / 16;
+    const int elem = is_f32 ? 4 : 2;
+    const int pad = is_f32 ? 4 : 8;
+    const size_t smem = (size_t)nf16 * 16 * (p.K + pad) * elem + 64;  // +64 B slack: B-frag k-tail reads may overrun the last LDS row
+    long tiles = (p.R + 63) / 64;
+    if (tiles > 16384) tiles = 16384;
+    dim3 grid((unsigned)tiles), block(256);
+    if (!is_f32)
+        row_gemm_kernel<__bf16><<<grid, block, smem, stream>>>(p, nf16);
+    else
+        row_gemm_kernel<float><<<grid, block, smem, stream>>>(p, nf16);
+}

@@ -1,0 +1,38 @@
+"""Metrics vs reference definitions (Metrics.py:5-26), incl. MAPE eps=1.0."""
+
+import numpy as np
+import torch
+
+from mpgcn_amd.train import metrics as M
+
+
+def test_values():
+    rng = np.random.default_rng(0)
+    p = rng.normal(size=(4, 5, 5)).astype(np.float64)
+    t = rng.normal(size=(4, 5, 5)).astype(np.float64) + 2.0
+    assert np.isclose(M.MSE(p, t), np.mean((p - t) ** 2))
+    assert np.isclose(M.RMSE(p, t), np.sqrt(np.mean((p - t) ** 2)))
+    assert np.isclose(M.MAE(p, t), np.mean(np.abs(p - t)))
+    assert np.isclose(M.MAPE(p, t), np.mean(np.abs(p - t) / (t + 1.0)))
+    assert np.isclose(M.PCC(p, t), np.corrcoef(p.flatten(), t.flatten())[0, 1])
+
+
+def test_evaluate_returns_four(capsys):
+    p = np.ones((3, 3))
+    t = np.full((3, 3), 2.0)
+    res = M.evaluate(p, t)
+    assert len(res) == 4
+    out = capsys.readouterr().out
+    assert "PCC:" in out  # printed but not returned (Metrics.py:10-11)
+
+
+def test_torch_matches_numpy():
+    rng = np.random.default_rng(1)
+    p = rng.normal(size=(100,)).astype(np.float32)
+    t = rng.normal(size=(100,)).astype(np.float32) + 3.0
+    mse, rmse, mae, mape, pcc = M.evaluate_torch(torch.from_numpy(p), torch.from_numpy(t))
+    assert np.isclose(mse, M.MSE(p, t), rtol=1e-5)
+    assert np.isclose(rmse, M.RMSE(p, t), rtol=1e-5)
+    assert np.isclose(mae, M.MAE(p, t), rtol=1e-5)
+    assert np.isclose(mape, M.MAPE(p, t), rtol=1e-5)
+    assert np.isclose(pcc, M.PCC(p, t), rtol=1e-4)
