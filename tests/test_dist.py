@@ -1,0 +1,98 @@
+"""Distributed logic on CPU: 2-rank gloo gradient all-reduce equivalence —
+2-rank DP on half batches must produce the same averaged gradients (and the
+same post-step weights) as 1 process on the full batch."""
+
+import os
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from mpgcn_amd.graph import build_supports
+from mpgcn_amd.models import MPGCN
+from mpgcn_amd.parallel import DistContext, GradAllReducer
+
+N, K, H, B, T = 8, 3, 16, 4, 5
+
+
+def _make_inputs(seed=0):
+    torch.manual_seed(seed)
+    x = torch.rand(B, T, N, N, 1)
+    y = torch.rand(B, 1, N, N, 1)
+    flow = torch.rand(B, N, N)
+    Gs = build_supports(torch.rand(1, N, N), "random_walk_diffusion", K - 1)[0]
+    Go = build_supports(flow, "random_walk_diffusion", K - 1)
+    Gd = build_supports(flow.transpose(-2, -1), "random_walk_diffusion", K - 1)
+    return x, y, Gs, Go, Gd
+
+
+def _make_model(seed=1):
+    torch.manual_seed(seed)
+    return MPGCN(M=2, K=K, input_dim=1, lstm_hidden_dim=H, lstm_num_layers=1,
+                 gcn_hidden_dim=H, gcn_num_layers=2, num_nodes=N)
+
+
+def _single_process_grads():
+    model = _make_model()
+    x, y, Gs, Go, Gd = _make_inputs()
+    out = model(x, [Gs, (Go, Gd)])
+    torch.nn.functional.mse_loss(out, y).backward()
+    return {n: p.grad.clone() for n, p in model.named_parameters()}
+
+
+def _rank_worker(rank, world, file_name, out_file):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world), LOCAL_RANK=str(rank))
+    dist.init_process_group("gloo", init_method=f"file://{file_name}",
+                            rank=rank, world_size=world)
+    ctx = DistContext(rank=rank, world_size=world, local_rank=rank, backend="gloo")
+    model = _make_model(seed=100 + rank)  # divergent init: broadcast must fix it
+    reducer = GradAllReducer(model, ctx)
+
+    x, y, Gs, Go, Gd = _make_inputs()
+    half = B // world
+    sl = slice(rank * half, (rank + 1) * half)
+    out = model(x[sl], [Gs, (Go[sl], Gd[sl])])
+    torch.nn.functional.mse_loss(out, y[sl]).backward()
+    reducer.finalize()
+    if rank == 0:
+        torch.save({n: p.grad.clone() for n, p in model.named_parameters()}, out_file)
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_two_rank_dp_grads_match_full_batch(tmp_path):
+    # rank-0 init is broadcast, so use the same seed stream as single-process:
+    # _make_model(seed=100) on rank 0 => compare against that model's grads
+    file_name = str(tmp_path / "pg_init")
+    out_file = str(tmp_path / "rank0_grads.pt")
+    ctxm = mp.get_context("spawn")
+    procs = [ctxm.Process(target=_rank_worker, args=(r, 2, file_name, out_file))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=240)
+        assert p.exitcode == 0
+    dp_grads = torch.load(out_file, weights_only=True)
+
+    # single-process full-batch reference with rank-0's init seed
+    model = _make_model(seed=100)
+    x, y, Gs, Go, Gd = _make_inputs()
+    out = model(x, [Gs, (Go, Gd)])
+    torch.nn.functional.mse_loss(out, y).backward()
+
+    for n, p in model.named_parameters():
+        # DP averages the two half-batch means; MSE over equal halves averages
+        # to the full-batch mean, so grads must match to fp tolerance
+        assert torch.allclose(dp_grads[n], p.grad, atol=1e-5), n
+
+
+def test_noop_context_without_env():
+    ctx = DistContext()
+    model = _make_model()
+    reducer = GradAllReducer(model, ctx)
+    x, y, Gs, Go, Gd = _make_inputs()
+    out = model(x, [Gs, (Go, Gd)])
+    torch.nn.functional.mse_loss(out, y).backward()
+    reducer.finalize()  # must be a no-op, not raise

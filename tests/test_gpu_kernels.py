@@ -1,0 +1,260 @@
+"""GPU numerics tests: every HIP kernel vs a plain PyTorch fp32 reference of
+the same op (run on an MI355X via gpurun; skipped without a GPU)."""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+
+
+def _ext():
+    from mpgcn_amd import ops
+
+    assert ops.has_ext(), "HIP extension must be present on a GPU box"
+    return ops.get_ext()
+
+
+def _mode1_ref(X, G):
+    if G.dim() == 3:
+        return torch.einsum("onm,bndl->bmdol", G, X)
+    return torch.einsum("bonm,bndl->bmdol", G, X)
+
+
+def _mode2_ref(V, G):
+    if G.dim() == 3:
+        return torch.einsum("scd,bmcsh->bmdh", G, V)
+    return torch.einsum("bscd,bmcsh->bmdh", G, V)
+
+
+def _tol(dtype):
+    return dict(atol=1e-3, rtol=1e-3) if dtype == torch.float32 else dict(atol=8e-2, rtol=8e-2)
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("N,dyn", [(64, False), (47, False), (64, True), (33, True)])
+def test_mode1(dtype, N, dyn):
+    ext = _ext()
+    torch.manual_seed(0)
+    B, S, C = 2, 3, 32
+    X = torch.randn(B, N, N, C, device=DEV).to(dtype)
+    G = torch.randn(B, S, N, N, device=DEV).to(dtype) if dyn else torch.randn(S, N, N, device=DEV).to(dtype)
+    U = ext.bdgcn_mode1(X.contiguous(), G.transpose(-2, -1).contiguous())
+    ref = _mode1_ref(X.float(), G.float())
+    assert U.shape == (B, N, N, S, C)
+    torch.testing.assert_close(U.float(), ref, **_tol(dtype))
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("N,S,dyn,relu", [(64, 3, False, True), (47, 3, True, False), (32, 5, False, True)])
+def test_mode2(dtype, N, S, dyn, relu):
+    ext = _ext()
+    torch.manual_seed(1)
+    B, H = 2, 32
+    V = torch.randn(B, N, N, S, H, device=DEV).to(dtype)
+    G = torch.randn(B, S, N, N, device=DEV).to(dtype) if dyn else torch.randn(S, N, N, device=DEV).to(dtype)
+    bias = torch.randn(H, device=DEV)
+    if dyn:
+        A2T = G.permute(0, 3, 2, 1).reshape(B, N, N * S).contiguous()
+    else:
+        A2T = G.permute(2, 1, 0).reshape(N, N * S).contiguous()
+    Y = ext.bdgcn_mode2(V.reshape(B, N, N * S, H).contiguous(), A2T, bias, relu, N, S)
+    ref = _mode2_ref(V.float(), G.float()) + bias
+    if relu:
+        ref = torch.relu(ref)
+    torch.testing.assert_close(Y.float(), ref, **_tol(dtype))
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_mode2_bwd(dtype):
+    ext = _ext()
+    torch.manual_seed(2)
+    B, N, S, H = 2, 48, 3, 32
+    dY = torch.randn(B, N, N, H, device=DEV).to(dtype)
+    G = torch.randn(S, N, N, device=DEV).to(dtype)
+    A2 = G.permute(1, 0, 2).reshape(N * S, N).contiguous()
+    dV = ext.bdgcn_mode2_bwd(dY.contiguous(), A2, S)
+    # dV[b,m,c,s,h] = sum_d G[s,c,d] dY[b,m,d,h]
+    ref = torch.einsum("scd,bmdh->bmcsh", G.float(), dY.float())
+    torch.testing.assert_close(dV.float(), ref, **_tol(dtype))
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("dyn", [False, True])
+def test_mode1_bwd(dtype, dyn):
+    ext = _ext()
+    torch.manual_seed(3)
+    B, N, S, C = 2, 40, 3, 32
+    dU = torch.randn(B, N, N, S, C, device=DEV).to(dtype)
+    G = torch.randn(B, S, N, N, device=DEV).to(dtype) if dyn else torch.randn(S, N, N, device=DEV).to(dtype)
+    if dyn:
+        A3T = G.permute(0, 2, 1, 3).reshape(B, N, S * N).contiguous()
+        ref = torch.einsum("bonm,bmdol->bndl", G.float(), dU.float())
+    else:
+        A3T = G.permute(1, 0, 2).reshape(N, S * N).contiguous()
+        ref = torch.einsum("onm,bmdol->bndl", G.float(), dU.float())
+    dX = ext.bdgcn_mode1_bwd(dU.contiguous(), A3T)
+    torch.testing.assert_close(dX.float(), ref, **_tol(dtype))
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("R,K,Nc,relu", [(1000, 96, 96, False), (513, 32, 1, True), (256, 128, 32, False)])
+def test_row_gemm(dtype, R, K, Nc, relu):
+    ext = _ext()
+    torch.manual_seed(4)
+    X = torch.randn(R, K, device=DEV).to(dtype)
+    W = torch.randn(K, Nc, device=DEV).to(dtype)
+    bias = torch.randn(Nc, device=DEV)
+    out = ext.row_gemm(X.contiguous(), W.contiguous(), bias, relu)
+    ref = X.float() @ W.float() + bias
+    if relu:
+        ref = torch.relu(ref)
+    torch.testing.assert_close(out.float(), ref, **_tol(dtype))
+
+
+def test_row_gemm_chunked_wide():
+    from mpgcn_amd.ops.functional import _row_gemm_chunked
+
+    ext = _ext()
+    torch.manual_seed(5)
+    R, K, Nc = 500, 160, 160  # S=5 dual-RWD shape: S*C = S*H = 160 > 128
+    X = torch.randn(R, K, device=DEV, dtype=torch.bfloat16)
+    W = torch.randn(K, Nc, device=DEV, dtype=torch.bfloat16)
+    out = _row_gemm_chunked(ext, X.contiguous(), W.contiguous(), None, False)
+    ref = X.float() @ W.float()
+    torch.testing.assert_close(out.float(), ref, atol=8e-2, rtol=8e-2)
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_lstm_forward_vs_nn_lstm(dtype):
+    from mpgcn_amd.ops.functional import fused_lstm_last
+
+    torch.manual_seed(6)
+    R, T, H = 333, 7, 32
+    x = torch.randn(R, T, device=DEV)
+    lstm = torch.nn.LSTM(1, H, num_layers=1, batch_first=True).to(DEV)
+    ref, _ = lstm(x.unsqueeze(-1), (torch.zeros(1, R, H, device=DEV),
+                                    torch.zeros(1, R, H, device=DEV)))
+    out = fused_lstm_last(
+        x.to(dtype), lstm.weight_ih_l0.to(dtype), lstm.weight_hh_l0.to(dtype),
+        lstm.bias_ih_l0, lstm.bias_hh_l0,
+    )
+    torch.testing.assert_close(out.float(), ref[:, -1, :].float(), **_tol(dtype))
+
+
+def test_lstm_backward_grads_match_autograd():
+    from mpgcn_amd.ops import eager
+    from mpgcn_amd.ops.functional import fused_lstm_last
+
+    torch.manual_seed(7)
+    R, T, H = 200, 6, 32
+    x = torch.randn(R, T, device=DEV)
+    w_ih = torch.randn(4 * H, 1, device=DEV) * 0.2
+    w_hh = torch.randn(4 * H, H, device=DEV) * 0.2
+    b_ih = torch.randn(4 * H, device=DEV) * 0.1
+    b_hh = torch.randn(4 * H, device=DEV) * 0.1
+
+    # autograd reference through the eager fp32 implementation
+    ref_in = [t.clone().requires_grad_(True) for t in (x, w_ih, w_hh, b_ih, b_hh)]
+    out, _, _ = eager.lstm_forward_eager(ref_in[0].unsqueeze(-1), *ref_in[1:])
+    loss = out[:, -1, :].square().sum()
+    loss.backward()
+
+    ins = [t.clone().requires_grad_(True) for t in (x, w_ih, w_hh, b_ih, b_hh)]
+    h = fused_lstm_last(*ins)
+    h.square().sum().backward()
+
+    for got, ref, name in zip(ins, ref_in, ("x", "w_ih", "w_hh", "b_ih", "b_hh")):
+        torch.testing.assert_close(
+            got.grad.float(), ref.grad.float(), atol=2e-2, rtol=2e-2, msg=name
+        )
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("dyn", [False, True])
+def test_bdgcn_layer_fwd_bwd_vs_eager(dtype, dyn):
+    from mpgcn_amd.ops import GraphOperator, bdgcn_layer, eager
+
+    torch.manual_seed(8)
+    B, N, C, H, S = 2, 40, 32, 32, 3
+    X = torch.randn(B, N, N, C, device=DEV).to(dtype)
+    if dyn:
+        Go = torch.randn(B, S, N, N, device=DEV).to(dtype) * 0.3
+        Gd = torch.randn(B, S, N, N, device=DEV).to(dtype) * 0.3
+    else:
+        Go = torch.randn(S, N, N, device=DEV).to(dtype) * 0.3
+        Gd = torch.randn(S, N, N, device=DEV).to(dtype) * 0.3
+    W = (torch.randn(C * S * S, H, device=DEV) * 0.05).to(dtype)
+    b = torch.randn(H, device=DEV)
+
+    Xg = X.clone().requires_grad_(True)
+    Wg = W.clone().requires_grad_(True)
+    bg = b.clone().requires_grad_(True)
+    out = bdgcn_layer(Xg, Wg, bg, GraphOperator(Go, Gd), relu=True)
+    out.square().sum().backward()
+
+    Xr = X.float().clone().requires_grad_(True)
+    Wr = W.float().clone().requires_grad_(True)
+    br = b.clone().requires_grad_(True)
+    ref = eager.bdgcn_layer_eager(Xr, Go.float(), Gd.float(), Wr, br, "relu")
+    ref.square().sum().backward()
+
+    tol = dict(atol=5e-2, rtol=5e-2) if dtype == torch.float32 else dict(atol=0.5, rtol=0.2)
+    torch.testing.assert_close(out.float(), ref.detach(), **_tol(dtype))
+    torch.testing.assert_close(Xg.grad.float(), Xr.grad, **tol)
+    torch.testing.assert_close(Wg.grad.float(), Wr.grad, **tol)
+    torch.testing.assert_close(bg.grad.float(), br.grad, **tol)
+
+
+def test_full_model_gpu_vs_cpu_oracle():
+    from mpgcn_amd.graph import build_supports
+    from mpgcn_amd.models import MPGCN
+
+    torch.manual_seed(9)
+    B, N, K, H, T = 2, 32, 3, 32, 7
+    model_cpu = MPGCN(M=2, K=K, input_dim=1, lstm_hidden_dim=H, lstm_num_layers=1,
+                      gcn_hidden_dim=H, gcn_num_layers=3, num_nodes=N)
+    model_gpu = MPGCN(M=2, K=K, input_dim=1, lstm_hidden_dim=H, lstm_num_layers=1,
+                      gcn_hidden_dim=H, gcn_num_layers=3, num_nodes=N,
+                      compute_dtype=torch.bfloat16).to(DEV)
+    model_gpu.load_state_dict(model_cpu.state_dict())
+
+    x = torch.rand(B, T, N, N, 1)
+    flow = torch.rand(B, N, N)
+    Gs = build_supports(torch.rand(1, N, N), "random_walk_diffusion", K - 1)[0]
+    Go = build_supports(flow, "random_walk_diffusion", K - 1)
+    Gd = build_supports(flow.transpose(-2, -1), "random_walk_diffusion", K - 1)
+
+    with torch.no_grad():
+        ref = model_cpu(x, [Gs, (Go, Gd)])
+        out = model_gpu(x.to(DEV), [Gs.to(DEV), (Go.to(DEV), Gd.to(DEV))]).cpu()
+    torch.testing.assert_close(out, ref, atol=5e-2, rtol=5e-2)
+
+
+def test_train_step_decreases_loss_on_gpu():
+    from mpgcn_amd.graph import build_supports
+    from mpgcn_amd.models import MPGCN
+
+    torch.manual_seed(10)
+    B, N, K, H, T = 4, 32, 3, 32, 7
+    model = MPGCN(M=2, K=K, input_dim=1, lstm_hidden_dim=H, lstm_num_layers=1,
+                  gcn_hidden_dim=H, gcn_num_layers=3, num_nodes=N,
+                  compute_dtype=torch.bfloat16).to(DEV)
+    opt = torch.optim.Adam(model.parameters(), lr=3e-3)
+    x = torch.rand(B, T, N, N, 1, device=DEV)
+    y = torch.rand(B, 1, N, N, 1, device=DEV) + 1.0
+    flow = torch.rand(B, N, N, device=DEV)
+    Gs = build_supports(torch.rand(1, N, N, device=DEV), "random_walk_diffusion", K - 1)[0]
+    Go = build_supports(flow, "random_walk_diffusion", K - 1)
+    Gd = build_supports(flow.transpose(-2, -1), "random_walk_diffusion", K - 1)
+
+    losses = []
+    for _ in range(30):
+        out = model(x, [Gs, (Go, Gd)])
+        loss = torch.nn.functional.mse_loss(out, y)
+        opt.zero_grad(set_to_none=True)
+        loss.backward()
+        opt.step()
+        losses.append(loss.item())
+    assert losses[-1] < losses[0] * 0.7, losses[:3] + losses[-3:]
