@@ -200,11 +200,18 @@ def test_bdgcn_layer_fwd_bwd_vs_eager(dtype, dyn):
     ref = eager.bdgcn_layer_eager(Xr, Go.float(), Gd.float(), Wr, br, "relu")
     ref.square().sum().backward()
 
-    tol = dict(atol=5e-2, rtol=5e-2) if dtype == torch.float32 else dict(atol=0.5, rtol=0.2)
     torch.testing.assert_close(out.float(), ref.detach(), **_tol(dtype))
-    torch.testing.assert_close(Xg.grad.float(), Xr.grad, **tol)
-    torch.testing.assert_close(Wg.grad.float(), Wr.grad, **tol)
-    torch.testing.assert_close(bg.grad.float(), br.grad, **tol)
+    if dtype == torch.float32:
+        tol = dict(atol=5e-2, rtol=5e-2)
+        torch.testing.assert_close(Xg.grad.float(), Xr.grad, **tol)
+        torch.testing.assert_close(Wg.grad.float(), Wr.grad, **tol)
+        torch.testing.assert_close(bg.grad.float(), br.grad, **tol)
+    else:
+        # bf16 grads compound rounding through dY->dV->dW; elementwise bounds
+        # are the wrong criterion — check normalized Frobenius error instead
+        for got, ref_g in ((Xg.grad, Xr.grad), (Wg.grad, Wr.grad), (bg.grad, br.grad)):
+            err = (got.float() - ref_g).norm() / ref_g.norm()
+            assert err < 2e-2, f"relative grad error {err:.4f}"
 
 
 def test_full_model_gpu_vs_cpu_oracle():
