@@ -136,9 +136,10 @@ class _BDGCNLayerFn(torch.autograd.Function):
         dV = ext.bdgcn_mode2_bwd(dY, gop.A2, S)  # (B,N,N,S,H)
         R = B * N * N
         dVflat = dV.reshape(R, S * Hdim)
-        # dWre: plain library reduction GEMM over R rows (rocBLAS), f32 accum
-        dWre = U.reshape(R, S * C).t() @ dVflat
-        dW = dWre.view(S, C, S, Hdim).permute(0, 2, 1, 3).reshape(S * S * C, Hdim)
+        # dWre^T = dV^T @ U via the fused reduction kernel (f32 accumulate)
+        dWreT, _, _ = ext.red_gemm(dVflat, U.reshape(R, S * C), False, None, 0, 0)
+        dWre = dWreT.t().to(dH.dtype)
+        dW = dWre.reshape(S, C, S, Hdim).permute(0, 2, 1, 3).reshape(S * S * C, Hdim)
         dU = _row_gemm_chunked(ext, dVflat, Wre.t().contiguous(), None, False)
         dX = ext.bdgcn_mode1_bwd(dU.view(B, N, N, S, C), gop.A3T)
         db = dbias if ctx.has_bias else None
@@ -198,20 +199,26 @@ class _FusedLSTMLastFn(torch.autograd.Function):
         dwhh = torch.zeros(4 * Hd, Hd, device=xc.device, dtype=torch.float32)
         dbias = torch.zeros(4 * Hd, device=xc.device, dtype=torch.float32)
         dwih = torch.zeros(4 * Hd, device=xc.device, dtype=torch.float32)
-        dx = torch.empty_like(xc)
+        need_dx = ctx.needs_input_grad[0]
+        dx = torch.zeros_like(xc)
         wih_c = wih_f.to(xc.dtype).view(-1, 1).contiguous()
         for t in range(T - 1, -1, -1):
             dgates, dc = ext.lstm_step_bwd(dh, dc, gs[t], cs[t], cs[t + 1])
-            # weight grads: plain library reduction GEMMs (rocBLAS), f32 accum
-            dwhh += (dgates.t() @ hs[t]).float()
-            dbias += dgates.sum(dim=0).float()
-            dwih += (dgates.t() @ xc[:, t].unsqueeze(-1)).squeeze(-1).float()
+            # fused reduction: dW_hh partial + dbias(colsum) + dw_ih(xdot) in
+            # one pass over dgates
+            w_part, b_part, x_part = ext.red_gemm(
+                dgates, hs[t], True, xc, T, t
+            )
+            dwhh += w_part
+            dbias += b_part
+            dwih += x_part
             dh = ext.row_gemm(dgates, whh, None, False)  # dgates @ W_hh -> dh_prev
-            dx[:, t] = ext.row_gemm(dgates, wih_c, None, False).view(-1)
+            if need_dx:
+                dx[:, t] = ext.row_gemm(dgates, wih_c, None, False).view(-1)
         ctx.saved_lists = None
         wdt = whh.dtype
         return (
-            dx,
+            dx if need_dx else None,
             dwih.view(-1, 1).to(wdt),
             dwhh.to(wdt),
             dbias,

@@ -209,6 +209,38 @@ void row_gemm_out(torch::Tensor X, torch::Tensor W,
     row_gemm_launch(p, is_f32(X), stream());
 }
 
+// Reduction GEMM: out = X^T @ Y (f32), plus optional colsum(X) and
+// xdot[k] = sum_r X[r,k]*xv[r]. Returns (out, colsum, xdot); colsum/xdot are
+// empty tensors when not requested. xv is addressed xv[r*xv_stride + xv_off].
+std::vector<torch::Tensor> red_gemm(torch::Tensor X, torch::Tensor Y,
+                                    bool want_colsum,
+                                    c10::optional<torch::Tensor> xvec,
+                                    long xv_stride, long xv_off) {
+    check_in(X, "X");
+    check_in(Y, "Y");
+    const long R = X.size(0), K = X.size(1), N = Y.size(1);
+    TORCH_CHECK(Y.size(0) == R, "row mismatch");
+    auto f32 = X.options().dtype(torch::kFloat);
+    auto out = torch::zeros({K, N}, f32);
+    auto colsum = want_colsum ? torch::zeros({K}, f32) : torch::Tensor();
+    const bool has_xv = xvec.has_value() && xvec->defined();
+    auto xdot = has_xv ? torch::zeros({K}, f32) : torch::Tensor();
+    RedGemmParams p{};
+    p.X = X.data_ptr();
+    p.Y = Y.data_ptr();
+    p.xvec = has_xv ? xvec->data_ptr() : nullptr;
+    p.xv_stride = xv_stride; p.xv_off = xv_off;
+    p.out = out.data_ptr<float>();
+    p.colsum = want_colsum ? colsum.data_ptr<float>() : nullptr;
+    p.xdot = has_xv ? xdot.data_ptr<float>() : nullptr;
+    p.R = R; p.K = (int)K; p.N = (int)N;
+    const int ch = chunk_elems(X);
+    p.x_vec = (K % ch == 0);
+    p.y_vec = (N % ch == 0);
+    red_gemm_launch(p, is_f32(X), stream());
+    return {out, colsum, xdot};
+}
+
 // One fused LSTM step. x: any tensor whose flat layout gives the step input at
 // x[r * x_stride + x_off]; h_prev: (R, H); c_prev: (R, H) f32; whh: (4H, H);
 // wih, bias: (4H) f32. Returns (h, c, gates_postact).
@@ -269,6 +301,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("bdgcn_mode2_bwd", &bdgcn_mode2_bwd, "backward dV of mode2");
     m.def("bdgcn_mode1_bwd", &bdgcn_mode1_bwd, "backward dX of mode1");
     m.def("row_gemm", &row_gemm, "fused row GEMM + bias + act (K3/K6)");
+    m.def("red_gemm", &red_gemm, "fused reduction GEMM X^T@Y + colsum + xdot");
     m.def("row_gemm_out", &row_gemm_out, "row GEMM into strided output slice");
     m.def("lstm_step_fwd", &lstm_step_fwd, "fused LSTM cell forward step (K4)");
     m.def("lstm_step_bwd", &lstm_step_bwd, "LSTM cell backward pointwise step");

@@ -61,8 +61,22 @@ struct LstmBwdParams {
     int H;
 };
 
+struct RedGemmParams {
+    const void* X;      // (R, K) row-major T
+    const void* Y;      // (R, N) row-major T
+    const void* xvec;   // per-row scalar T at xvec[r*xv_stride + xv_off]; nullable
+    long xv_stride, xv_off;
+    float* out;         // (K, N) f32, accumulated (caller zeroes)
+    float* colsum;      // (K,) f32 or nullptr
+    float* xdot;        // (K,) f32 or nullptr (requires xvec)
+    long R;
+    int K, N;
+    int x_vec, y_vec;
+};
+
 extern "C" {
 void axis_gemm_launch(AxisGemmParams p, int instances, int is_f32, hipStream_t s);
+void red_gemm_launch(RedGemmParams p, int is_f32, hipStream_t s);
 void row_gemm_launch(RowGemmParams p, int is_f32, hipStream_t s);
 void lstm_step_fwd_launch(LstmStepParams p, int is_f32, hipStream_t s);
 void lstm_step_bwd_launch(LstmBwdParams p, int is_f32, hipStream_t s);

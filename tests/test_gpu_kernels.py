@@ -113,6 +113,24 @@ def test_row_gemm(dtype, R, K, Nc, relu):
     torch.testing.assert_close(out.float(), ref, **_tol(dtype))
 
 
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("R,K,Nc", [(5000, 96, 96), (2048, 128, 32), (777, 160, 160)])
+def test_red_gemm(dtype, R, K, Nc):
+    ext = _ext()
+    torch.manual_seed(11)
+    X = (torch.randn(R, K, device=DEV) * 0.1).to(dtype)
+    Y = (torch.randn(R, Nc, device=DEV) * 0.1).to(dtype)
+    xv = torch.randn(R, 3, device=DEV).to(dtype)
+    out, colsum, xdot = ext.red_gemm(X.contiguous(), Y.contiguous(), True,
+                                     xv.contiguous(), 3, 1)
+    ref = X.float().t() @ Y.float()
+    rel = (out - ref).norm() / ref.norm()
+    assert rel < (1e-5 if dtype == torch.float32 else 1e-2), rel
+    torch.testing.assert_close(colsum, X.float().sum(0), atol=1e-1, rtol=1e-2)
+    ref_xdot = (X.float() * xv[:, 1].float().unsqueeze(1)).sum(0)
+    torch.testing.assert_close(xdot, ref_xdot, atol=1e-1, rtol=2e-2)
+
+
 def test_row_gemm_chunked_wide():
     from mpgcn_amd.ops.functional import _row_gemm_chunked
 
