@@ -171,54 +171,57 @@ class _FusedLSTMLastFn(torch.autograd.Function):
         wih_f = w_ih.reshape(-1).float().contiguous()
         bias_f = (b_ih.float() + b_hh.float()).contiguous()
 
-        h = torch.zeros(R, Hd, device=dev, dtype=x.dtype)
-        c = torch.zeros(R, Hd, device=dev, dtype=torch.float32)
-        hs, cs, gs = [h], [c], []
+        # T-slab state buffers: H_buf[t] is h BEFORE step t (h_0 = 0), so the
+        # whole-sequence weight-grad reduction in backward is ONE red_gemm pass
+        # over (T*R) rows instead of T separate kernel sweeps.
+        H_buf = torch.zeros(T + 1, R, Hd, device=dev, dtype=x.dtype)
+        C_buf = torch.zeros(T + 1, R, Hd, device=dev, dtype=torch.float32)
+        G_buf = torch.empty(T, R, 4 * Hd, device=dev, dtype=x.dtype)
         xc = x.contiguous()
         for t in range(T):
-            h, c, gates = ext.lstm_step_fwd(xc, T, t, h, c, whh, wih_f, bias_f)
-            hs.append(h)
-            cs.append(c)
-            gs.append(gates)
-        ctx.saved_lists = (hs, cs, gs)
-        ctx.save_for_backward(xc, whh, wih_f)
+            ext.lstm_step_fwd(xc, T, t, H_buf[t], C_buf[t], whh, wih_f, bias_f,
+                              H_buf[t + 1], C_buf[t + 1], G_buf[t])
+        ctx.save_for_backward(xc, whh, wih_f, H_buf, C_buf, G_buf)
         ctx.T = T
-        return h
+        return H_buf[T]
 
     @staticmethod
     def backward(ctx, dh_last):
         ext = _ops.get_ext()
-        xc, whh, wih_f = ctx.saved_tensors
-        hs, cs, gs = ctx.saved_lists
+        xc, whh, wih_f, H_buf, C_buf, G_buf = ctx.saved_tensors
         T = ctx.T
         R = xc.shape[0]
         Hd = whh.shape[1]
+        dev = xc.device
 
         dh = dh_last.contiguous()
         dc = None
-        dwhh = torch.zeros(4 * Hd, Hd, device=xc.device, dtype=torch.float32)
-        dbias = torch.zeros(4 * Hd, device=xc.device, dtype=torch.float32)
-        dwih = torch.zeros(4 * Hd, device=xc.device, dtype=torch.float32)
-        need_dx = ctx.needs_input_grad[0]
-        dx = torch.zeros_like(xc)
-        wih_c = wih_f.to(xc.dtype).view(-1, 1).contiguous()
+        dc_buf = [torch.empty(R, Hd, device=dev, dtype=torch.float32) for _ in range(2)]
+        dG_buf = torch.empty(T, R, 4 * Hd, device=dev, dtype=xc.dtype)
         for t in range(T - 1, -1, -1):
-            dgates, dc = ext.lstm_step_bwd(dh, dc, gs[t], cs[t], cs[t + 1])
-            # fused reduction: dW_hh partial + dbias(colsum) + dw_ih(xdot) in
-            # one pass over dgates
-            w_part, b_part, x_part = ext.red_gemm(
-                dgates, hs[t], True, xc, T, t
-            )
-            dwhh += w_part
-            dbias += b_part
-            dwih += x_part
-            dh = ext.row_gemm(dgates, whh, None, False)  # dgates @ W_hh -> dh_prev
-            if need_dx:
-                dx[:, t] = ext.row_gemm(dgates, wih_c, None, False).view(-1)
-        ctx.saved_lists = None
+            dc_out = dc_buf[t % 2]
+            ext.lstm_step_bwd(dh, dc, G_buf[t], C_buf[t], C_buf[t + 1],
+                              dG_buf[t], dc_out)
+            dc = dc_out
+            if t > 0:
+                dh = ext.row_gemm(dG_buf[t], whh, None, False)  # dgates @ W_hh
+
+        # one fused reduction over all T steps: dW_hh = dG^T @ h_prev,
+        # dbias = colsum(dG), dw_ih = dG^T x
+        xT = xc.t().contiguous()  # (T, R): row t*R+r aligns with dG/h slabs
+        dwhh, dbias, dwih = ext.red_gemm(
+            dG_buf.view(T * R, 4 * Hd), H_buf[:T].reshape(T * R, Hd),
+            True, xT.view(-1, 1), 1, 0,
+        )
+        need_dx = ctx.needs_input_grad[0]
+        dx = None
+        if need_dx:
+            wih_c = wih_f.to(xc.dtype).view(-1, 1).contiguous()
+            dx_all = ext.row_gemm(dG_buf.view(T * R, 4 * Hd), wih_c, None, False)
+            dx = dx_all.view(T, R).t().contiguous()
         wdt = whh.dtype
         return (
-            dx if need_dx else None,
+            dx,
             dwih.view(-1, 1).to(wdt),
             dwhh.to(wdt),
             dbias,
@@ -254,10 +257,14 @@ class _LinearActFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dOut):
         X2d, weight, out = ctx.saved_tensors
-        dY = dOut * (out > 0).to(dOut.dtype) if ctx.relu else dOut
+        dY = (dOut * (out > 0).to(dOut.dtype) if ctx.relu else dOut).contiguous()
         dX = dY @ weight.to(dY.dtype)
-        dW = dY.t() @ X2d
-        db = dY.sum(0).to(torch.float32) if ctx.has_bias else None
+        # dW = dY^T @ X via the fused reduction kernel (rocBLAS is ~25x off
+        # roofline on this 1-column tall reduction); colsum(dY) gives dbias
+        ext = _ops.get_ext()
+        dWt, db_cs, _ = ext.red_gemm(dY, X2d, ctx.has_bias, None, 0, 0)
+        dW = dWt.view(weight.shape).to(weight.dtype)
+        db = db_cs if ctx.has_bias else None
         return dX, dW, db, None
 
 

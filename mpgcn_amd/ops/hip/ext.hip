@@ -241,22 +241,25 @@ std::vector<torch::Tensor> red_gemm(torch::Tensor X, torch::Tensor Y,
     return {out, colsum, xdot};
 }
 
-// One fused LSTM step. x: any tensor whose flat layout gives the step input at
-// x[r * x_stride + x_off]; h_prev: (R, H); c_prev: (R, H) f32; whh: (4H, H);
-// wih, bias: (4H) f32. Returns (h, c, gates_postact).
-std::vector<torch::Tensor> lstm_step_fwd(torch::Tensor x, long x_stride, long x_off,
-                                         torch::Tensor h_prev, torch::Tensor c_prev,
-                                         torch::Tensor whh, torch::Tensor wih,
-                                         torch::Tensor bias) {
+// One fused LSTM step, writing into caller-provided output slices (lets the
+// Python layer keep per-timestep states in contiguous T-slab buffers so the
+// whole sequence's weight-grad reduction is a single red_gemm pass).
+// x[r * x_stride + x_off]; h_prev/h_out: (R, H); c_prev/c_out: (R, H) f32;
+// whh: (4H, H); wih, bias: (4H) f32; gates_out: (R, 4H) post-activation.
+void lstm_step_fwd(torch::Tensor x, long x_stride, long x_off,
+                   torch::Tensor h_prev, torch::Tensor c_prev,
+                   torch::Tensor whh, torch::Tensor wih, torch::Tensor bias,
+                   torch::Tensor h_out, torch::Tensor c_out,
+                   torch::Tensor gates_out) {
     check_in(x, "x");
     check_in(h_prev, "h_prev");
     check_in(whh, "whh");
+    check_in(h_out, "h_out");
+    check_in(gates_out, "gates_out");
     TORCH_CHECK(c_prev.scalar_type() == torch::kFloat && c_prev.is_contiguous());
+    TORCH_CHECK(c_out.scalar_type() == torch::kFloat && c_out.is_contiguous());
     TORCH_CHECK(wih.scalar_type() == torch::kFloat && bias.scalar_type() == torch::kFloat);
     const long R = h_prev.size(0), H = h_prev.size(1);
-    auto h = torch::empty_like(h_prev);
-    auto c = torch::empty_like(c_prev);
-    auto gates = torch::empty({R, 4 * H}, h_prev.options());
     LstmStepParams p{};
     p.x = x.data_ptr(); p.x_stride = x_stride; p.x_off = x_off;
     p.h_prev = h_prev.data_ptr();
@@ -264,35 +267,31 @@ std::vector<torch::Tensor> lstm_step_fwd(torch::Tensor x, long x_stride, long x_
     p.whh = whh.data_ptr();
     p.wih = wih.data_ptr<float>();
     p.bias = bias.data_ptr<float>();
-    p.h_out = h.data_ptr();
-    p.c_out = c.data_ptr<float>();
-    p.gates_out = gates.data_ptr();
+    p.h_out = h_out.data_ptr();
+    p.c_out = c_out.data_ptr<float>();
+    p.gates_out = gates_out.data_ptr();
     p.R = R; p.H = (int)H;
     lstm_step_fwd_launch(p, is_f32(h_prev), stream());
-    return {h, c, gates};
 }
 
-// One LSTM backward step (pointwise part). Returns (dgates_preact, dc_prev).
-std::vector<torch::Tensor> lstm_step_bwd(torch::Tensor dh,
-                                         c10::optional<torch::Tensor> dc_in,
-                                         torch::Tensor gates, torch::Tensor c_prev,
-                                         torch::Tensor c) {
+// One LSTM backward step (pointwise part), writing into provided buffers.
+void lstm_step_bwd(torch::Tensor dh, c10::optional<torch::Tensor> dc_in,
+                   torch::Tensor gates, torch::Tensor c_prev, torch::Tensor c,
+                   torch::Tensor dgates_out, torch::Tensor dc_prev_out) {
     check_in(dh, "dh");
     check_in(gates, "gates");
+    check_in(dgates_out, "dgates_out");
     const long R = dh.size(0), H = dh.size(1);
-    auto dgates = torch::empty({R, 4 * H}, dh.options());
-    auto dc_prev = torch::empty({R, H}, c.options());
     LstmBwdParams p{};
     p.dh = dh.data_ptr();
     p.dc_in = (dc_in.has_value() && dc_in->defined()) ? dc_in->data_ptr<float>() : nullptr;
     p.gates = gates.data_ptr();
     p.c_prev = c_prev.data_ptr<float>();
     p.c = c.data_ptr<float>();
-    p.dgates = dgates.data_ptr();
-    p.dc_prev = dc_prev.data_ptr<float>();
+    p.dgates = dgates_out.data_ptr();
+    p.dc_prev = dc_prev_out.data_ptr<float>();
     p.R = R; p.H = (int)H;
     lstm_step_bwd_launch(p, is_f32(dh), stream());
-    return {dgates, dc_prev};
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {

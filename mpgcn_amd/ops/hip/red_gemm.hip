@@ -20,7 +20,7 @@ template <typename T, int AKF, int ANF>
 __launch_bounds__(256) __global__ void red_gemm_kernel(RedGemmParams p) {
     using MT = MfmaTraits<T>;
     constexpr int CH = 16 / sizeof(T);
-    constexpr int RCH = 64;  // rows per chunk
+    constexpr int RCH = sizeof(T) == 2 ? 128 : 64;  // rows per chunk (LDS budget)
     constexpr int PAD = MT::LDS_PAD;
     constexpr int KMAX = AKF * 32, NMAX = ANF * 32;
 
@@ -100,15 +100,23 @@ __launch_bounds__(256) __global__ void red_gemm_kernel(RedGemmParams p) {
                     acc[kf][nf] = MT::mfma(af[kf], bf[nf], acc[kf][nf]);
         }
 
-        if ((p.colsum || XV) && tid < p.K) {
-            float c1 = 0.f, x1 = 0.f;
-            for (int r = 0; r < RCH; ++r) {
-                const float v = to_f32(ldsXT[tid][r]);
-                c1 += v;
-                if (XV) x1 += v * to_f32(ldsXV[r]);
+        if (p.colsum || XV) {
+            // split the row range across thread groups of KMAX: thread t
+            // handles k = t % KMAX over its group's row slice — keeps (most of)
+            // the 256 threads busy; partials combine in the final atomics
+            constexpr int NSPLIT = (256 / KMAX >= 2) ? 2 : 1;
+            const int k = tid % KMAX;
+            const int part = tid / KMAX;
+            if (k < p.K && part < NSPLIT) {
+                float c1 = 0.f, x1 = 0.f;
+                for (int r = part * (RCH / NSPLIT); r < (part + 1) * (RCH / NSPLIT); ++r) {
+                    const float v = to_f32(ldsXT[k][r]);
+                    c1 += v;
+                    if (XV) x1 += v * to_f32(ldsXV[r]);
+                }
+                cs += c1;
+                xd += x1;
             }
-            cs += c1;
-            xd += x1;
         }
         __syncthreads();
     }
@@ -127,15 +135,20 @@ __launch_bounds__(256) __global__ void red_gemm_kernel(RedGemmParams p) {
             }
         }
     }
-    if (tid < p.K) {
-        if (p.colsum) atomicAdd(&p.colsum[tid], cs);
-        if (XV) atomicAdd(&p.xdot[tid], xd);
+    {
+        constexpr int NSPLIT = (256 / KMAX >= 2) ? 2 : 1;
+        const int k = tid % KMAX;
+        if (k < p.K && tid / KMAX < NSPLIT) {
+            if (p.colsum) atomicAdd(&p.colsum[k], cs);
+            if (XV) atomicAdd(&p.xdot[k], xd);
+        }
     }
 }
 
 extern "C" void red_gemm_launch(RedGemmParams p, int is_f32, hipStream_t s) {
-    long chunks = (p.R + 63) / 64;
-    long blocks = chunks < 304 ? chunks : 304;
+    const int rch = is_f32 ? 64 : 128;
+    long chunks = (p.R + rch - 1) / rch;
+    long blocks = chunks < 608 ? chunks : 608;
     dim3 grid((unsigned)blocks), block(256);
 #define DISPATCH(TT)                                                        \
     do {                                                                    \
