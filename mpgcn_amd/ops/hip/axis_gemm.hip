@@ -100,8 +100,14 @@ __launch_bounds__(256) __global__ void axis_gemm_kernel(AxisGemmParams p) {
                         if (q0 + i < p.L) tmp[i] = X[rbase + xcol_off(p, q0 + i)];
                 }
             }
+            // lane-rotated write order de-conflicts the transpose (see
+            // red_gemm.hip): stride between lanes' rows is ~0 mod 32 banks,
+            // rotating the element index spreads each instruction's writes
 #pragma unroll
-            for (int i = 0; i < CH; ++i) ldsB[qc * CH + i][krow] = tmp[i];
+            for (int i = 0; i < CH; ++i) {
+                const int j = (i + tid) % CH;
+                ldsB[qc * CH + j][krow] = tmp[j];
+            }
         }
         __syncthreads();
 

@@ -55,8 +55,13 @@ __launch_bounds__(256) __global__ void red_gemm_kernel(RedGemmParams p) {
                     for (int i = 0; i < CH; ++i)
                         if (k0 + i < p.K) tmp[i] = X[row * p.K + k0 + i];
             }
+            // lane-rotated write order: consecutive lanes write different
+            // LDS rows per instruction (banks 4*{0..7}, no 16-way conflict)
 #pragma unroll
-            for (int i = 0; i < CH; ++i) ldsXT[k0 + i][r] = tmp[i];
+            for (int i = 0; i < CH; ++i) {
+                const int j = (i + tid) % CH;
+                ldsXT[k0 + j][r] = tmp[j];
+            }
         }
         // stage Y chunk transposed: ldsYT[n][r]
         for (int idx = tid; idx < RCH * (NMAX / CH); idx += 256) {
@@ -73,7 +78,10 @@ __launch_bounds__(256) __global__ void red_gemm_kernel(RedGemmParams p) {
                         if (n0 + i < p.N) tmp[i] = Y[row * p.N + n0 + i];
             }
 #pragma unroll
-            for (int i = 0; i < CH; ++i) ldsYT[n0 + i][r] = tmp[i];
+            for (int i = 0; i < CH; ++i) {
+                const int j = (i + tid) % CH;
+                ldsYT[n0 + j][r] = tmp[j];
+            }
         }
         if (XV) {
             for (int r = tid; r < RCH; r += 256) {
