@@ -142,7 +142,7 @@ __launch_bounds__(256) __global__ void axis_gemm_kernel(AxisGemmParams p) {
                 const int q = l0 + wn + nf * 16 + lrow;
                 if (m < p.M && q < p.L) {
                     float v = acc[mf][nf][r];
-                    if (p.bias) v += p.bias[q];
+                    if (p.bias) v += p.bias[p.bias_mod ? q % p.bias_mod : q];
                     if (p.relu) v = fmaxf(v, 0.f);
                     O[o_base + (long)m * p.o_row + ocol_off(p, q)] = from_f32<T>(v);
                 }

@@ -80,24 +80,28 @@ torch::Tensor bdgcn_mode2(torch::Tensor V, torch::Tensor A2T,
     TORCH_CHECK(B * N <= 65535, "too many instances");
     auto Y = torch::empty({B, N, N, H}, V.options());
 
+    // One GEMM per batch element: OUT[d, (m,h)] = sum_cs A2T[d,cs] V[b,m,cs,h]
+    // — folding the m axis into the L (column) dimension gives each staged
+    // A-tile (the graph) 256x more MFMA work than per-(b,m) instances.
     AxisGemmParams p{};
     p.AT = A2T.data_ptr();
     p.X = V.data_ptr();
     p.OUT = Y.data_ptr();
     p.bias = bias_ptr(bias);
-    p.M = (int)N; p.K = (int)(N * S); p.L = (int)H;
-    p.a_div = dyn ? (int)N : 1; p.a_bs1 = dyn ? N * N * S : 0; p.a_bs2 = 0;
-    p.x_div = 1; p.x_bs1 = N * S * H; p.x_bs2 = 0;
-    p.o_div = 1; p.o_bs1 = N * H; p.o_bs2 = 0;
+    p.M = (int)N; p.K = (int)(N * S); p.L = (int)(N * H);
+    p.a_div = 1; p.a_bs1 = dyn ? N * N * S : 0; p.a_bs2 = 0;
+    p.x_div = 1; p.x_bs1 = N * N * S * H; p.x_bs2 = 0;
+    p.o_div = 1; p.o_bs1 = N * N * H; p.o_bs2 = 0;
     p.kdiv = 1; p.k_lo = H;
-    p.qdiv = 0;
-    p.o_row = H;
-    p.ogdiv = 0;
+    p.qdiv = (int)H; p.q_hi = N * S * H;   // q = (m, h)
+    p.o_row = H;                           // out row = d
+    p.ogdiv = (int)H; p.og_hi = N * H;
     p.relu = relu ? 1 : 0;
+    p.bias_mod = (int)H;
     const int ch = chunk_elems(V);
     p.a_vec = ((N * S) % ch == 0);
     p.x_vec = (H % ch == 0);
-    axis_gemm_launch(p, (int)(B * N), is_f32(V), stream());
+    axis_gemm_launch(p, (int)B, is_f32(V), stream());
     return Y;
 }
 
@@ -112,22 +116,24 @@ torch::Tensor bdgcn_mode2_bwd(torch::Tensor dY, torch::Tensor A2, long S) {
     TORCH_CHECK(B * N <= 65535, "too many instances");
     auto dV = torch::empty({B, N, N, S, H}, dY.options());
 
+    // One GEMM per batch element (m folded into L, as in bdgcn_mode2):
+    // dV[cs, (m,h)] = sum_d A2[cs,d] dY[b,m,d,h]
     AxisGemmParams p{};
     p.AT = A2.data_ptr();
     p.X = dY.data_ptr();
     p.OUT = dV.data_ptr();
-    p.M = (int)(N * S); p.K = (int)N; p.L = (int)H;
-    p.a_div = dyn ? (int)N : 1; p.a_bs1 = dyn ? N * S * N : 0; p.a_bs2 = 0;
-    p.x_div = 1; p.x_bs1 = N * H; p.x_bs2 = 0;
-    p.o_div = 1; p.o_bs1 = N * S * H; p.o_bs2 = 0;
+    p.M = (int)(N * S); p.K = (int)N; p.L = (int)(N * H);
+    p.a_div = 1; p.a_bs1 = dyn ? N * S * N : 0; p.a_bs2 = 0;
+    p.x_div = 1; p.x_bs1 = N * N * H; p.x_bs2 = 0;
+    p.o_div = 1; p.o_bs1 = N * N * S * H; p.o_bs2 = 0;
     p.kdiv = 1; p.k_lo = H;
-    p.qdiv = 0;
-    p.o_row = H;
-    p.ogdiv = 0;
+    p.qdiv = (int)H; p.q_hi = N * H;       // q = (m, h)
+    p.o_row = H;                           // out row = cs
+    p.ogdiv = (int)H; p.og_hi = N * S * H;
     const int ch = chunk_elems(dY);
     p.a_vec = (N % ch == 0);
     p.x_vec = (H % ch == 0);
-    axis_gemm_launch(p, (int)(B * N), is_f32(dY), stream());
+    axis_gemm_launch(p, (int)B, is_f32(dY), stream());
     return dV;
 }
 

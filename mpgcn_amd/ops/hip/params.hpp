@@ -18,6 +18,7 @@ struct AxisGemmParams {
     long o_row;
     int ogdiv; long og_hi;
     int relu;
+    int bias_mod;  // bias index = bias_mod ? q % bias_mod : q
     int a_vec, x_vec;
     int tiles_l;
 };
