@@ -229,9 +229,45 @@ class _FusedLSTMLastFn(torch.autograd.Function):
         )
 
 
+class _RegLSTMFn(torch.autograd.Function):
+    """Fully register-resident LSTM (bf16, H=32, T<=8): forward computes h_T
+    with states in registers; backward recomputes the forward in-kernel and
+    produces all gradients in one pass — no per-step state tensors at all."""
+
+    @staticmethod
+    def forward(ctx, x, w_ih, w_hh, b_ih, b_hh):
+        ext = _ops.get_ext()
+        whh = w_hh.contiguous()
+        wih_f = w_ih.reshape(-1).float().contiguous()
+        bias_f = (b_ih.float() + b_hh.float()).contiguous()
+        xc = x.contiguous()
+        h = ext.lstm_fused_fwd(xc, whh, wih_f, bias_f)
+        ctx.save_for_backward(xc, whh, wih_f, bias_f)
+        return h
+
+    @staticmethod
+    def backward(ctx, dh):
+        ext = _ops.get_ext()
+        xc, whh, wih_f, bias_f = ctx.saved_tensors
+        need_dx = ctx.needs_input_grad[0]
+        dwhh, dbias, dwih, dx = ext.lstm_fused_bwd(
+            xc, whh, whh.t().contiguous(), wih_f, bias_f, dh.contiguous(), need_dx
+        )
+        wdt = whh.dtype
+        return (
+            dx if need_dx else None,
+            dwih.view(-1, 1).to(wdt),
+            dwhh.to(wdt),
+            dbias,
+            dbias.clone(),
+        )
+
+
 def fused_lstm_last(x, w_ih, w_hh, b_ih, b_hh):
     """Last hidden state of a 1-layer batch-first LSTM over (R, T) scalar inputs."""
     Hd = w_hh.shape[1]
+    if x.is_cuda and Hd == 32 and x.dtype == torch.bfloat16 and x.shape[1] <= 8:
+        return _RegLSTMFn.apply(x, w_ih, w_hh, b_ih, b_hh)
     kernel_ok = Hd == 32 or (Hd == 16 and x.dtype == torch.float32)
     if x.is_cuda and kernel_ok:
         return _FusedLSTMLastFn.apply(x, w_ih, w_hh, b_ih, b_hh)

@@ -300,6 +300,62 @@ void lstm_step_bwd(torch::Tensor dh, c10::optional<torch::Tensor> dc_in,
     lstm_step_bwd_launch(p, is_f32(dh), stream());
 }
 
+// Fully-fused register-resident LSTM forward: x (R, T) bf16 -> h_T (R, H=32).
+torch::Tensor lstm_fused_fwd(torch::Tensor x, torch::Tensor whh,
+                             torch::Tensor wih, torch::Tensor bias) {
+    check_in(x, "x");
+    check_in(whh, "whh");
+    TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "fused LSTM is bf16-only");
+    TORCH_CHECK(whh.size(1) == 32 && whh.size(0) == 128, "fused LSTM needs H=32");
+    const long R = x.size(0);
+    const int T = (int)x.size(1);
+    TORCH_CHECK(T >= 1 && T <= 8, "fused LSTM needs T <= 8");
+    auto h = torch::empty({R, 32}, x.options());
+    LstmFusedParams p{};
+    p.x = x.data_ptr();
+    p.whh = whh.data_ptr();
+    p.wih = wih.data_ptr<float>();
+    p.bias = bias.data_ptr<float>();
+    p.h_out = h.data_ptr();
+    p.R = R; p.T = T;
+    lstm_fused_fwd_launch(p, stream());
+    return h;
+}
+
+// Fused backward with in-kernel forward recompute. Returns
+// {dwhh (4H,H) f32, dbias (4H) f32, dwih (4H) f32, dx (R,T) bf16 or undefined}.
+std::vector<torch::Tensor> lstm_fused_bwd(torch::Tensor x, torch::Tensor whh,
+                                          torch::Tensor whh2, torch::Tensor wih,
+                                          torch::Tensor bias, torch::Tensor dh,
+                                          bool need_dx) {
+    check_in(x, "x");
+    check_in(whh, "whh");
+    check_in(whh2, "whh2");
+    check_in(dh, "dh");
+    const long R = x.size(0);
+    const int T = (int)x.size(1);
+    const int nb = lstm_fused_bwd_blocks(R);
+    auto f32 = x.options().dtype(torch::kFloat);
+    auto ws_dw = torch::empty({nb, 128, 32}, f32);
+    auto ws_db = torch::empty({nb, 128}, f32);
+    auto ws_dwih = torch::empty({nb, 128}, f32);
+    auto dx = need_dx ? torch::empty({R, (long)T}, x.options()) : torch::Tensor();
+    LstmFusedParams p{};
+    p.x = x.data_ptr();
+    p.whh = whh.data_ptr();
+    p.whh2 = whh2.data_ptr();
+    p.wih = wih.data_ptr<float>();
+    p.bias = bias.data_ptr<float>();
+    p.dh = dh.data_ptr();
+    p.ws_dw = ws_dw.data_ptr<float>();
+    p.ws_db = ws_db.data_ptr<float>();
+    p.ws_dwih = ws_dwih.data_ptr<float>();
+    p.dx = need_dx ? dx.data_ptr() : nullptr;
+    p.R = R; p.T = T;
+    lstm_fused_bwd_launch(p, stream());
+    return {ws_dw.sum(0), ws_db.sum(0), ws_dwih.sum(0), dx};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("bdgcn_mode1", &bdgcn_mode1, "origin-axis graph product (K1)");
     m.def("bdgcn_mode2", &bdgcn_mode2, "dest-axis graph product + bias + act (K2)");
@@ -309,5 +365,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("red_gemm", &red_gemm, "fused reduction GEMM X^T@Y + colsum + xdot");
     m.def("row_gemm_out", &row_gemm_out, "row GEMM into strided output slice");
     m.def("lstm_step_fwd", &lstm_step_fwd, "fused LSTM cell forward step (K4)");
+    m.def("lstm_fused_fwd", &lstm_fused_fwd, "register-resident fused LSTM forward");
+    m.def("lstm_fused_bwd", &lstm_fused_bwd, "register-resident fused LSTM backward");
     m.def("lstm_step_bwd", &lstm_step_bwd, "LSTM cell backward pointwise step");
 }

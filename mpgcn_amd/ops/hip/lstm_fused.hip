@@ -1,0 +1,405 @@
+// Fully-fused register-resident LSTM (gfx950, bf16, H = 32, T <= 8).
+//
+// The MPGCN temporal encoder runs R = batch*N^2 independent scalar-input
+// sequences of length T <= 8 and only the LAST hidden state is consumed
+// (MPGCN.py:103-104). That makes the whole sequence small enough to live in
+// registers per 16-row wave tile:
+//
+//   forward:  h/c stay in registers across all T steps; the ONLY global
+//             traffic is x (R*T scalars) in and h_T out — no per-step state
+//             round-trips through HBM (the slab variant in lstm.hip moves
+//             ~9 GB/step at the flagship config; this moves ~0.2 GB).
+//   backward: recomputes the forward states in registers (bit-identical MFMA
+//             sequence), then walks t = T-1..0 computing the gate gradients,
+//             the dh chain (MFMA vs W_hh), and the dW_hh / dbias / dw_ih
+//             partials in-register; partials land in a per-block f32
+//             workspace reduced by one torch sum (no atomics, no dgates
+//             materialization).
+//
+// Weight-grad MFMAs pair two timesteps per K=32 contraction (16 rows each).
+// Per-wave LDS images: dg_img [2*16][4H] (dh A-operand, vector reads; dW
+// A-operand via strided scalar reads), hT_img [H][2*16] (dW B-operand,
+// vector reads). W_hh staged twice: [4H][H] (gate GEMM B) and [H][4H]
+// (dh-chain B).
+#include "common.hpp"
+#include "params.hpp"
+
+#define LF_H 32
+#define LF_G4 128
+// per-wave image paddings
+#define DG_LD (LF_G4 + 8)  // dg_img row length (rows = 32 paired rows)
+#define HT_LD (32 + 8)     // hT_img row length (rows = 32 k values)
+
+
+typedef __attribute__((ext_vector_type(8))) __bf16 lf_frag;
+
+// gate activation order: i, f, g, o (torch chunk order)
+__device__ __forceinline__ float lf_act(float v, int gate) {
+    return (gate == 2) ? tanhf(v) : 1.f / (1.f + __expf(-v));
+}
+
+// one gate GEMM: acc[nf] = h_frag @ whhT fragments (8 MFMAs, K = 32)
+__device__ __forceinline__ void lf_gate_mfma(const __bf16* ldsW, lf_frag h_frag,
+                                             int lrow, int kgrp, f32x4 acc[8]) {
+#pragma unroll
+    for (int nf = 0; nf < 8; ++nf) {
+        const lf_frag bf = *(const lf_frag*)
+            &ldsW[(nf * 16 + lrow) * (LF_H + 8) + kgrp * 8];
+        acc[nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(h_frag, bf, acc[nf], 0, 0, 0);
+    }
+}
+
+template <int T>
+__launch_bounds__(256) __global__ void lstm_fused_fwd_kernel(LstmFusedParams p) {
+    __shared__ __bf16 ldsW[LF_G4 * (LF_H + 8)];
+    __shared__ __bf16 ldsH[4][16 * (LF_H + 8)];  // per-wave h transpose tile
+
+    const __bf16* __restrict__ X = (const __bf16*)p.x;
+    const __bf16* __restrict__ Whh = (const __bf16*)p.whh;
+    __bf16* __restrict__ Ho = (__bf16*)p.h_out;
+
+    const int tid = threadIdx.x;
+    for (int i = tid; i < LF_G4 * (LF_H + 8); i += 256) {
+        const int n = i / (LF_H + 8), k = i % (LF_H + 8);
+        ldsW[i] = (k < LF_H) ? Whh[n * LF_H + k] : (__bf16)0.f;
+    }
+    __syncthreads();
+
+    const int w = tid / WAVE, lane = tid % WAVE;
+    const int lrow = lane & 15, kgrp = lane >> 4;
+    __bf16* myH = &ldsH[w][0];
+
+    const long ntiles = (p.R + 63) / 64;
+    for (long tile = blockIdx.x; tile < ntiles; tile += gridDim.x) {
+        const long r0 = tile * 64 + w * 16;
+        lf_frag h_frag = {};
+        float c[2][4] = {};  // c[jf][r] for j = jf*16 + lrow, row = r0 + kgrp*4 + r
+#pragma unroll
+        for (int t = 0; t < T; ++t) {
+            f32x4 acc[8] = {};
+            lf_gate_mfma(ldsW, h_frag, lrow, kgrp, acc);
+            // activations + state update (C-layout)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const long m = r0 + kgrp * 4 + r;
+                const float xv = (m < p.R) ? to_f32(X[m * T + t]) : 0.f;
+                float gv[8];
+#pragma unroll
+                for (int nf = 0; nf < 8; ++nf) {
+                    const int n = nf * 16 + lrow;
+                    gv[nf] = lf_act(acc[nf][r] + xv * p.wih[n] + p.bias[n], n / LF_H);
+                }
+#pragma unroll
+                for (int jf = 0; jf < 2; ++jf) {
+                    const float c_new = gv[2 + jf] * c[jf][r] + gv[0 + jf] * gv[4 + jf];
+                    c[jf][r] = c_new;
+                    // h in C-layout -> per-wave LDS tile for transposition
+                    myH[(kgrp * 4 + r) * (LF_H + 8) + jf * 16 + lrow] =
+                        (__bf16)(gv[6 + jf] * tanhf(c_new));
+                }
+            }
+            __syncthreads();
+            h_frag = *(const lf_frag*)&myH[lrow * (LF_H + 8) + kgrp * 8];
+            __syncthreads();
+        }
+        // store h_T, vectorized: lane holds row lrow's k-run
+        const long row = r0 + lrow;
+        if (row < p.R) *(lf_frag*)&Ho[row * LF_H + kgrp * 8] = h_frag;
+    }
+}
+
+// NOTE on gv indexing above: gate order i,f,g,o over n = 0..127 means
+// nf 0..1 = i, 2..3 = f, 4..5 = g, 6..7 = o; jf = n/16 % 2 selects the half.
+
+template <int T>
+__launch_bounds__(256) __global__ void lstm_fused_bwd_kernel(LstmFusedParams p) {
+    __shared__ __bf16 ldsW[LF_G4 * (LF_H + 8)];   // [n][k] gate-GEMM B image
+    __shared__ __bf16 ldsW2[LF_H * (LF_G4 + 8)];  // [k][n] dh-chain B image
+    __shared__ __bf16 ldsDG[4][32 * DG_LD];       // per-wave paired dgates [row'][n]
+    __shared__ __bf16 ldsHT[4][LF_H * HT_LD];     // per-wave paired h_prev^T [k][row']
+    // c states live in LDS (they would otherwise cost 8*T VGPRs per lane):
+    // CST[w][t][row][j], row stride 33 breaks write conflicts
+    __shared__ float ldsC[4][T * 16 * 33];
+
+    const __bf16* __restrict__ X = (const __bf16*)p.x;
+    const __bf16* __restrict__ Whh = (const __bf16*)p.whh;
+    const __bf16* __restrict__ Whh2 = (const __bf16*)p.whh2;
+    const __bf16* __restrict__ DH = (const __bf16*)p.dh;
+    __bf16* __restrict__ DX = (__bf16*)p.dx;
+
+    const int tid = threadIdx.x;
+    for (int i = tid; i < LF_G4 * (LF_H + 8); i += 256) {
+        const int n = i / (LF_H + 8), k = i % (LF_H + 8);
+        ldsW[i] = (k < LF_H) ? Whh[n * LF_H + k] : (__bf16)0.f;
+    }
+    for (int i = tid; i < LF_H * (LF_G4 + 8); i += 256) {
+        const int k = i / (LF_G4 + 8), n = i % (LF_G4 + 8);
+        ldsW2[i] = (n < LF_G4) ? Whh2[k * LF_G4 + n] : (__bf16)0.f;
+    }
+    __syncthreads();
+
+    const int w = tid / WAVE, lane = tid % WAVE;
+    const int lrow = lane & 15, kgrp = lane >> 4;
+    __bf16* myDG = &ldsDG[w][0];
+    __bf16* myHT = &ldsHT[w][0];
+    __bf16* myH = &ldsHT[w][0];  // phase-A transpose scratch aliases the hT image
+    float* myC = &ldsC[w][0];
+
+    // per-wave dW accumulator: D[i=n][j=k], 8 m-frags x 2 j-frags
+    f32x4 dw_acc[8][2] = {};
+    float db_acc[8] = {}, dwih_acc[8] = {};
+
+    const long ntiles = (p.R + 63) / 64;
+    for (long tile = blockIdx.x; tile < ntiles; tile += gridDim.x) {
+        const long r0 = tile * 64 + w * 16;
+        // ---- phase A: forward recompute; h states in registers, c in LDS ----
+        lf_frag h_states[T];  // h AFTER step t (A-frag layout)
+#define CST(t, r, jf) myC[(t) * 16 * 33 + (kgrp * 4 + (r)) * 33 + (jf) * 16 + lrow]
+        {
+            lf_frag h_frag = {};
+            float c[2][4] = {};
+#pragma unroll
+            for (int t = 0; t < T; ++t) {
+                f32x4 acc[8] = {};
+                lf_gate_mfma(ldsW, h_frag, lrow, kgrp, acc);
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    const long m = r0 + kgrp * 4 + r;
+                    const float xv = (m < p.R) ? to_f32(X[m * T + t]) : 0.f;
+                    float gv[8];
+#pragma unroll
+                    for (int nf = 0; nf < 8; ++nf) {
+                        const int n = nf * 16 + lrow;
+                        gv[nf] = lf_act(acc[nf][r] + xv * p.wih[n] + p.bias[n], n / LF_H);
+                    }
+#pragma unroll
+                    for (int jf = 0; jf < 2; ++jf) {
+                        const float c_new = gv[2 + jf] * c[jf][r] + gv[0 + jf] * gv[4 + jf];
+                        c[jf][r] = c_new;
+                        CST(t, r, jf) = c_new;
+                        myH[(kgrp * 4 + r) * (LF_H + 8) + jf * 16 + lrow] =
+                            (__bf16)(gv[6 + jf] * tanhf(c_new));
+                    }
+                }
+                __syncthreads();
+                h_frag = *(const lf_frag*)&myH[lrow * (LF_H + 8) + kgrp * 8];
+                __syncthreads();
+                h_states[t] = h_frag;
+            }
+        }
+
+        // ---- phase B: backward t = T-1 .. 0 ----
+        // dh, dc in C-layout registers
+        float dh[2][4], dc[2][4] = {};
+#pragma unroll
+        for (int jf = 0; jf < 2; ++jf)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const long m = r0 + kgrp * 4 + r;
+                dh[jf][r] = (m < p.R) ? to_f32(DH[m * LF_H + jf * 16 + lrow]) : 0.f;
+            }
+
+#pragma unroll
+        for (int tt = 0; tt < T; ++tt) {
+            const int t = T - 1 - tt;
+            const int slot = tt % 2;
+            // recompute this step's gates from h_prev
+            lf_frag hp = {};
+            if (t > 0) hp = h_states[t - 1];
+            f32x4 acc[8] = {};
+            lf_gate_mfma(ldsW, hp, lrow, kgrp, acc);
+
+            float dgp[8][4];  // pre-activation gate grads, C-layout
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const long m = r0 + kgrp * 4 + r;
+                const bool ok = m < p.R;
+                const float xv = ok ? to_f32(X[m * T + t]) : 0.f;
+                float gv[8];
+#pragma unroll
+                for (int nf = 0; nf < 8; ++nf) {
+                    const int n = nf * 16 + lrow;
+                    gv[nf] = lf_act(acc[nf][r] + xv * p.wih[n] + p.bias[n], n / LF_H);
+                }
+#pragma unroll
+                for (int jf = 0; jf < 2; ++jf) {
+                    const float i_g = gv[0 + jf], f_g = gv[2 + jf];
+                    const float g_g = gv[4 + jf], o_g = gv[6 + jf];
+                    const float c_t = CST(t, r, jf);
+                    const float c_prev = (t > 0) ? CST(t - 1, r, jf) : 0.f;
+                    const float tc = tanhf(c_t);
+                    float d_c = dc[jf][r] + dh[jf][r] * o_g * (1.f - tc * tc);
+                    const float d_i = d_c * g_g, d_g = d_c * i_g, d_f = d_c * c_prev;
+                    const float d_o = dh[jf][r] * tc;
+                    const float da_i = ok ? d_i * i_g * (1.f - i_g) : 0.f;
+                    const float da_f = ok ? d_f * f_g * (1.f - f_g) : 0.f;
+                    const float da_g = ok ? d_g * (1.f - g_g * g_g) : 0.f;
+                    const float da_o = ok ? d_o * o_g * (1.f - o_g) : 0.f;
+                    dgp[0 + jf][r] = da_i;
+                    dgp[2 + jf][r] = da_f;
+                    dgp[4 + jf][r] = da_g;
+                    dgp[6 + jf][r] = da_o;
+                    dc[jf][r] = d_c * f_g;
+                    db_acc[0 + jf] += da_i; db_acc[2 + jf] += da_f;
+                    db_acc[4 + jf] += da_g; db_acc[6 + jf] += da_o;
+                    dwih_acc[0 + jf] += da_i * xv; dwih_acc[2 + jf] += da_f * xv;
+                    dwih_acc[4 + jf] += da_g * xv; dwih_acc[6 + jf] += da_o * xv;
+                }
+            }
+
+            // optional dx[row, t] = sum_n dgp[row, n] * wih[n]
+            if (DX) {
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    float part = 0.f;
+#pragma unroll
+                    for (int nf = 0; nf < 8; ++nf)
+                        part += dgp[nf][r] * p.wih[nf * 16 + lrow];
+                    // reduce over the 16 lrow lanes (same row across lrow)
+#pragma unroll
+                    for (int s = 1; s < 16; s <<= 1)
+                        part += __shfl_xor(part, s);
+                    const long m = r0 + kgrp * 4 + r;
+                    if (lrow == 0 && m < p.R) DX[m * T + t] = (__bf16)part;
+                }
+            }
+
+            // stage dgates [row'][n] and h_prev^T [k][row'] into this slot
+#pragma unroll
+            for (int r = 0; r < 4; ++r)
+#pragma unroll
+                for (int nf = 0; nf < 8; ++nf)
+                    myDG[(slot * 16 + kgrp * 4 + r) * DG_LD + nf * 16 + lrow] =
+                        (__bf16)dgp[nf][r];
+            {
+                // hp is A-frag layout: lane holds h_prev[row=lrow][k=kgrp*8+j]
+#pragma unroll
+                for (int j = 0; j < 8; ++j)
+                    myHT[(kgrp * 8 + j) * HT_LD + slot * 16 + lrow] = hp[j];
+            }
+            __syncthreads();
+
+            // dh chain: dh_prev[row][k] = sum_n dgp[row][n] * Whh[n][k]
+            if (t > 0) {
+                f32x4 dh_acc[2] = {};
+#pragma unroll
+                for (int kf = 0; kf < 4; ++kf) {
+                    const lf_frag af = *(const lf_frag*)
+                        &myDG[(slot * 16 + lrow) * DG_LD + kf * 32 + kgrp * 8];
+#pragma unroll
+                    for (int jf = 0; jf < 2; ++jf) {
+                        const lf_frag bf = *(const lf_frag*)
+                            &ldsW2[(jf * 16 + lrow) * (LF_G4 + 8) + kf * 32 + kgrp * 8];
+                        dh_acc[jf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            af, bf, dh_acc[jf], 0, 0, 0);
+                    }
+                }
+#pragma unroll
+                for (int jf = 0; jf < 2; ++jf)
+#pragma unroll
+                    for (int r = 0; r < 4; ++r) dh[jf][r] = dh_acc[jf][r];
+            }
+
+            // dW pair MFMA when both slots are filled (or at the last step)
+            if (slot == 1 || t == 0) {
+                if (slot == 0) {  // odd T tail: zero the unused slot 1
+#pragma unroll
+                    for (int r = 0; r < 4; ++r)
+#pragma unroll
+                        for (int nf = 0; nf < 8; ++nf)
+                            myDG[(16 + kgrp * 4 + r) * DG_LD + nf * 16 + lrow] = (__bf16)0.f;
+#pragma unroll
+                    for (int j = 0; j < 8; ++j)
+                        myHT[(kgrp * 8 + j) * HT_LD + 16 + lrow] = (__bf16)0.f;
+                    __syncthreads();
+                }
+                // dW[n][k] += sum_{row'} dg[row'][n] * h_prev[row'][k]
+#pragma unroll
+                for (int mf = 0; mf < 8; ++mf) {
+                    // A-frag via strided scalar reads of dg_img (column n)
+                    alignas(16) __bf16 a_sc[8];
+                    const int n = mf * 16 + lrow;
+#pragma unroll
+                    for (int jj = 0; jj < 8; ++jj)
+                        a_sc[jj] = myDG[(kgrp * 8 + jj) * DG_LD + n];
+                    const lf_frag af = *(const lf_frag*)a_sc;
+#pragma unroll
+                    for (int jf = 0; jf < 2; ++jf) {
+                        const lf_frag bf = *(const lf_frag*)
+                            &myHT[(jf * 16 + lrow) * HT_LD + kgrp * 8];
+                        dw_acc[mf][jf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            af, bf, dw_acc[mf][jf], 0, 0, 0);
+                    }
+                }
+            }
+            __syncthreads();
+        }
+    }
+
+    // ---- write per-block partials to the workspace ----
+    // block-level reduce across the 4 waves via LDS (reuse ldsDG as f32 scratch)
+    float* red = (float*)&ldsDG[0][0];  // 4H * H floats = 16 KB
+    for (int i = tid; i < LF_G4 * LF_H; i += 256) red[i] = 0.f;
+    __syncthreads();
+    for (int ww = 0; ww < 4; ++ww) {
+        if (w == ww) {
+#pragma unroll
+            for (int mf = 0; mf < 8; ++mf)
+#pragma unroll
+                for (int jf = 0; jf < 2; ++jf)
+#pragma unroll
+                    for (int r = 0; r < 4; ++r) {
+                        const int n = mf * 16 + kgrp * 4 + r;
+                        const int k = jf * 16 + lrow;
+                        red[n * LF_H + k] += dw_acc[mf][jf][r];
+                    }
+        }
+        __syncthreads();
+    }
+    float* ws = p.ws_dw + (long)blockIdx.x * LF_G4 * LF_H;
+    for (int i = tid; i < LF_G4 * LF_H; i += 256) ws[i] = red[i];
+
+    // dbias / dwih: lane covers n = nf*16 + lrow for its 8 nf; rows are summed
+    // already; reduce over kgrp lanes (same n) then across waves via LDS
+    float* redb = (float*)&ldsHT[0][0];  // 2 * 4H floats
+    for (int i = tid; i < 2 * LF_G4; i += 256) redb[i] = 0.f;
+    __syncthreads();
+    // every lane LDS-atomically adds its partial into the shared row
+#pragma unroll
+    for (int nf = 0; nf < 8; ++nf) {
+        atomicAdd(&redb[nf * 16 + lrow], db_acc[nf]);
+        atomicAdd(&redb[LF_G4 + nf * 16 + lrow], dwih_acc[nf]);
+    }
+    __syncthreads();
+    for (int i = tid; i < LF_G4; i += 256) {
+        p.ws_db[(long)blockIdx.x * LF_G4 + i] = redb[i];
+        p.ws_dwih[(long)blockIdx.x * LF_G4 + i] = redb[LF_G4 + i];
+    }
+}
+
+extern "C" void lstm_fused_fwd_launch(LstmFusedParams p, hipStream_t s) {
+    long tiles = (p.R + 63) / 64;
+    long blocks = tiles < 8192 ? tiles : 8192;
+    dim3 grid((unsigned)blocks), block(256);
+    switch (p.T) {
+#define CASE(TT) case TT: lstm_fused_fwd_kernel<TT><<<grid, block, 0, s>>>(p); break;
+        CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
+#undef CASE
+        default: fprintf(stderr, "lstm_fused_fwd: T=%d unsupported\n", p.T); abort();
+    }
+}
+
+extern "C" int lstm_fused_bwd_blocks(long R) {
+    long tiles = (R + 63) / 64;
+    return (int)(tiles < 512 ? tiles : 512);
+}
+
+extern "C" void lstm_fused_bwd_launch(LstmFusedParams p, hipStream_t s) {
+    dim3 grid(lstm_fused_bwd_blocks(p.R)), block(256);
+    switch (p.T) {
+#define CASE(TT) case TT: lstm_fused_bwd_kernel<TT><<<grid, block, 0, s>>>(p); break;
+        CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
+#undef CASE
+        default: fprintf(stderr, "lstm_fused_bwd: T=%d unsupported\n", p.T); abort();
+    }
+}

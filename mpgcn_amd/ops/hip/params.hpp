@@ -75,10 +75,29 @@ struct RedGemmParams {
     int x_vec, y_vec;
 };
 
+struct LstmFusedParams {
+    const void* x;  // (R, T) T
+    const void* whh;   // (4H, H)
+    const void* whh2;  // (H, 4H) = whh^T (backward only)
+    const float* wih;  // (4H)
+    const float* bias; // (4H)
+    void* h_out;       // (R, H) — forward output h_T
+    const void* dh;    // (R, H) — backward input
+    float* ws_dw;      // (nblocks, 4H, H) f32 workspace
+    float* ws_db;      // (nblocks, 4H)
+    float* ws_dwih;    // (nblocks, 4H)
+    void* dx;          // (R, T) or nullptr
+    long R;
+    int T;
+};
+
 extern "C" {
 void axis_gemm_launch(AxisGemmParams p, int instances, int is_f32, hipStream_t s);
 void red_gemm_launch(RedGemmParams p, int is_f32, hipStream_t s);
 void row_gemm_launch(RowGemmParams p, int is_f32, hipStream_t s);
 void lstm_step_fwd_launch(LstmStepParams p, int is_f32, hipStream_t s);
 void lstm_step_bwd_launch(LstmBwdParams p, int is_f32, hipStream_t s);
+void lstm_fused_fwd_launch(LstmFusedParams p, hipStream_t s);
+void lstm_fused_bwd_launch(LstmFusedParams p, hipStream_t s);
+int lstm_fused_bwd_blocks(long R);
 }

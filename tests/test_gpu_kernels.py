@@ -189,6 +189,35 @@ def test_lstm_backward_grads_match_autograd():
         )
 
 
+def test_reg_lstm_bf16_grads_match_autograd():
+    """The register-resident bf16 LSTM (forward + recompute-backward) vs the
+    eager fp32 autograd reference."""
+    from mpgcn_amd.ops import eager
+    from mpgcn_amd.ops.functional import _RegLSTMFn
+
+    torch.manual_seed(12)
+    R, T, H = 3000, 7, 32
+    x = torch.randn(R, T, device=DEV) * 0.5
+    w_ih = torch.randn(4 * H, 1, device=DEV) * 0.2
+    w_hh = torch.randn(4 * H, H, device=DEV) * 0.2
+    b_ih = torch.randn(4 * H, device=DEV) * 0.1
+    b_hh = torch.randn(4 * H, device=DEV) * 0.1
+
+    ref_in = [t.clone().requires_grad_(True) for t in (x, w_ih, w_hh, b_ih, b_hh)]
+    out, _, _ = eager.lstm_forward_eager(ref_in[0].unsqueeze(-1), *ref_in[1:])
+    out[:, -1, :].square().sum().backward()
+
+    ins = [x.bfloat16(), w_ih.bfloat16(), w_hh.bfloat16(), b_ih, b_hh]
+    ins = [t.clone().requires_grad_(True) for t in ins]
+    h = _RegLSTMFn.apply(*ins)
+    torch.testing.assert_close(h.float(), out[:, -1, :].detach(), atol=5e-2, rtol=5e-2)
+    h.float().square().sum().backward()
+
+    for got, ref, name in zip(ins, ref_in, ("x", "w_ih", "w_hh", "b_ih", "b_hh")):
+        err = (got.grad.float() - ref.grad).norm() / (ref.grad.norm() + 1e-8)
+        assert err < 3e-2, f"{name}: rel grad err {err:.4f}"
+
+
 @pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
 @pytest.mark.parametrize("dyn", [False, True])
 def test_bdgcn_layer_fwd_bwd_vs_eager(dtype, dyn):
