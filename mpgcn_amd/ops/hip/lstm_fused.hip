@@ -33,6 +33,14 @@
 
 typedef __attribute__((ext_vector_type(8))) __bf16 lf_frag;
 
+// Wave-local LDS fence: the transpose tiles are PER-WAVE, so cross-lane
+// visibility needs only this wave's ds ops committed (lgkmcnt(0)), not a
+// block barrier — at 1 block/CU a __syncthreads here would serialize all
+// four independent waves on every step.
+__device__ __forceinline__ void lds_wave_fence() {
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+}
+
 // gate activation order: i, f, g, o (torch chunk order)
 __device__ __forceinline__ float lf_act(float v, int gate) {
     return (gate == 2) ? tanhf(v) : 1.f / (1.f + __expf(-v));
@@ -98,9 +106,9 @@ __launch_bounds__(256) __global__ void lstm_fused_fwd_kernel(LstmFusedParams p) 
                         (__bf16)(gv[6 + jf] * tanhf(c_new));
                 }
             }
-            __syncthreads();
+            lds_wave_fence();
             h_frag = *(const lf_frag*)&myH[lrow * (LF_H + 8) + kgrp * 8];
-            __syncthreads();
+            lds_wave_fence();
         }
         // store h_T, vectorized: lane holds row lrow's k-run
         const long row = r0 + lrow;
@@ -181,9 +189,9 @@ __launch_bounds__(256) __global__ void lstm_fused_bwd_kernel(LstmFusedParams p) 
                             (__bf16)(gv[6 + jf] * tanhf(c_new));
                     }
                 }
-                __syncthreads();
+                lds_wave_fence();
                 h_frag = *(const lf_frag*)&myH[lrow * (LF_H + 8) + kgrp * 8];
-                __syncthreads();
+                lds_wave_fence();
                 h_states[t] = h_frag;
             }
         }
@@ -277,7 +285,7 @@ __launch_bounds__(256) __global__ void lstm_fused_bwd_kernel(LstmFusedParams p) 
                 for (int j = 0; j < 8; ++j)
                     myHT[(kgrp * 8 + j) * HT_LD + slot * 16 + lrow] = hp[j];
             }
-            __syncthreads();
+            lds_wave_fence();
 
             // dh chain: dh_prev[row][k] = sum_n dgp[row][n] * Whh[n][k]
             if (t > 0) {
@@ -311,7 +319,7 @@ __launch_bounds__(256) __global__ void lstm_fused_bwd_kernel(LstmFusedParams p) 
 #pragma unroll
                     for (int j = 0; j < 8; ++j)
                         myHT[(kgrp * 8 + j) * HT_LD + 16 + lrow] = (__bf16)0.f;
-                    __syncthreads();
+                    lds_wave_fence();
                 }
                 // dW[n][k] += sum_{row'} dg[row'][n] * h_prev[row'][k]
 #pragma unroll
@@ -332,9 +340,10 @@ __launch_bounds__(256) __global__ void lstm_fused_bwd_kernel(LstmFusedParams p) 
                     }
                 }
             }
-            __syncthreads();
+            lds_wave_fence();
         }
     }
+    __syncthreads();  // before reusing per-wave LDS as the block-reduce scratch
 
     // ---- write per-block partials to the workspace ----
     // block-level reduce across the 4 waves via LDS (reuse ldsDG as f32 scratch)
