@@ -54,6 +54,18 @@ __device__ __forceinline__ float from_f32<float>(float v) { return v; }
 __device__ __forceinline__ float to_f32(__bf16 v) { return (float)v; }
 __device__ __forceinline__ float to_f32(float v) { return v; }
 
+// Fast transcendentals: v_rcp_f32-based sigmoid/tanh (no IEEE division
+// sequences, no libm branches — hipcc otherwise emits v_div_scale/div_fixup
+// chains and branchy tanhf that dominate the LSTM kernels' issue time).
+// Accuracy ~1 ulp of rcp (~1e-7 rel) — far below bf16 resolution.
+__device__ __forceinline__ float fast_sigmoid(float v) {
+    return __builtin_amdgcn_rcpf(1.f + __expf(-v));
+}
+__device__ __forceinline__ float fast_tanh(float v) {
+    // tanh(x) = 1 - 2/(exp(2x)+1); saturates correctly at +-inf
+    return 1.f - 2.f * __builtin_amdgcn_rcpf(__expf(2.f * v) + 1.f);
+}
+
 // 16-byte raw copy chunk (8 bf16 / 4 f32).
 struct alignas(16) Chunk16 { int v[4]; };
 

@@ -81,7 +81,7 @@ __launch_bounds__(256) __global__ void lstm_step_fwd_kernel(LstmStepParams p) {
                 const int n = nf * 16 + lrow;
                 float v = acc[nf][r] + xv * p.wih[n] + p.bias[n];
                 const int gate = n / H;
-                v = (gate == 2) ? tanhf(v) : 1.f / (1.f + __expf(-v));
+                v = (gate == 2) ? fast_tanh(v) : fast_sigmoid(v);
                 gv[nf] = v;
                 Go[m * G4 + n] = from_f32<T>(v);
             }
@@ -93,7 +93,7 @@ __launch_bounds__(256) __global__ void lstm_step_fwd_kernel(LstmStepParams p) {
                 const float g_g = gv[2 * h16 + jf], o_g = gv[3 * h16 + jf];
                 const float c_new = f_g * p.c_prev[m * H + j] + i_g * g_g;
                 p.c_out[m * H + j] = c_new;
-                Ho[m * H + j] = from_f32<T>(o_g * tanhf(c_new));
+                Ho[m * H + j] = from_f32<T>(o_g * fast_tanh(c_new));
             }
         }
     }
@@ -115,7 +115,7 @@ __launch_bounds__(256) __global__ void lstm_step_bwd_kernel(LstmBwdParams p) {
         const float g_g = to_f32(gates[r * G4 + 2 * H + j]);
         const float o_g = to_f32(gates[r * G4 + 3 * H + j]);
         const float cv = p.c[idx];
-        const float tc = tanhf(cv);
+        const float tc = fast_tanh(cv);
         const float dhv = to_f32(dh[idx]);
         float dc = dhv * o_g * (1.f - tc * tc);
         if (p.dc_in) dc += p.dc_in[idx];

@@ -43,7 +43,7 @@ __device__ __forceinline__ void lds_wave_fence() {
 
 // gate activation order: i, f, g, o (torch chunk order)
 __device__ __forceinline__ float lf_act(float v, int gate) {
-    return (gate == 2) ? tanhf(v) : 1.f / (1.f + __expf(-v));
+    return (gate == 2) ? fast_tanh(v) : fast_sigmoid(v);
 }
 
 // one gate GEMM: acc[nf] = h_frag @ whhT fragments (8 MFMAs, K = 32)
@@ -103,7 +103,7 @@ __launch_bounds__(256) __global__ void lstm_fused_fwd_kernel(LstmFusedParams p) 
                     c[jf][r] = c_new;
                     // h in C-layout -> per-wave LDS tile for transposition
                     myH[(kgrp * 4 + r) * (LF_H + 8) + jf * 16 + lrow] =
-                        (__bf16)(gv[6 + jf] * tanhf(c_new));
+                        (__bf16)(gv[6 + jf] * fast_tanh(c_new));
                 }
             }
             lds_wave_fence();
@@ -186,7 +186,7 @@ __launch_bounds__(256) __global__ void lstm_fused_bwd_kernel(LstmFusedParams p) 
                         c[jf][r] = c_new;
                         CST(t, r, jf) = c_new;
                         myH[(kgrp * 4 + r) * (LF_H + 8) + jf * 16 + lrow] =
-                            (__bf16)(gv[6 + jf] * tanhf(c_new));
+                            (__bf16)(gv[6 + jf] * fast_tanh(c_new));
                     }
                 }
                 lds_wave_fence();
@@ -235,7 +235,7 @@ __launch_bounds__(256) __global__ void lstm_fused_bwd_kernel(LstmFusedParams p) 
                     const float g_g = gv[4 + jf], o_g = gv[6 + jf];
                     const float c_t = CST(t, r, jf);
                     const float c_prev = (t > 0) ? CST(t - 1, r, jf) : 0.f;
-                    const float tc = tanhf(c_t);
+                    const float tc = fast_tanh(c_t);
                     float d_c = dc[jf][r] + dh[jf][r] * o_g * (1.f - tc * tc);
                     const float d_i = d_c * g_g, d_g = d_c * i_g, d_f = d_c * c_prev;
                     const float d_o = dh[jf][r] * tc;
