@@ -240,18 +240,22 @@ class _RegLSTMFn(torch.autograd.Function):
         whh = w_hh.contiguous()
         wih_f = w_ih.reshape(-1).float().contiguous()
         bias_f = (b_ih.float() + b_hh.float()).contiguous()
-        xc = x.contiguous()
-        h = ext.lstm_fused_fwd(xc, whh, wih_f, bias_f)
-        ctx.save_for_backward(xc, whh, wih_f, bias_f)
+        T = x.shape[1]
+        # kernel reads each row's x as one 16-byte vector: pad to 8 columns
+        xp = torch.nn.functional.pad(x, (0, 8 - T)).contiguous() if T < 8 else x.contiguous()
+        h = ext.lstm_fused_fwd(xp, T, whh, wih_f, bias_f)
+        ctx.save_for_backward(xp, whh, wih_f, bias_f)
+        ctx.T = T
         return h
 
     @staticmethod
     def backward(ctx, dh):
         ext = _ops.get_ext()
-        xc, whh, wih_f, bias_f = ctx.saved_tensors
+        xp, whh, wih_f, bias_f = ctx.saved_tensors
         need_dx = ctx.needs_input_grad[0]
         dwhh, dbias, dwih, dx = ext.lstm_fused_bwd(
-            xc, whh, whh.t().contiguous(), wih_f, bias_f, dh.contiguous(), need_dx
+            xp, ctx.T, whh, whh.t().contiguous(), wih_f, bias_f, dh.contiguous(),
+            need_dx
         )
         wdt = whh.dtype
         return (

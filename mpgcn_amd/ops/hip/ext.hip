@@ -301,14 +301,15 @@ void lstm_step_bwd(torch::Tensor dh, c10::optional<torch::Tensor> dc_in,
 }
 
 // Fully-fused register-resident LSTM forward: x (R, T) bf16 -> h_T (R, H=32).
-torch::Tensor lstm_fused_fwd(torch::Tensor x, torch::Tensor whh,
+torch::Tensor lstm_fused_fwd(torch::Tensor x, long T_logical, torch::Tensor whh,
                              torch::Tensor wih, torch::Tensor bias) {
     check_in(x, "x");
     check_in(whh, "whh");
     TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "fused LSTM is bf16-only");
     TORCH_CHECK(whh.size(1) == 32 && whh.size(0) == 128, "fused LSTM needs H=32");
+    TORCH_CHECK(x.size(1) == 8, "x must be zero-padded to 8 columns");
     const long R = x.size(0);
-    const int T = (int)x.size(1);
+    const int T = (int)T_logical;
     TORCH_CHECK(T >= 1 && T <= 8, "fused LSTM needs T <= 8");
     auto h = torch::empty({R, 32}, x.options());
     LstmFusedParams p{};
@@ -324,7 +325,8 @@ torch::Tensor lstm_fused_fwd(torch::Tensor x, torch::Tensor whh,
 
 // Fused backward with in-kernel forward recompute. Returns
 // {dwhh (4H,H) f32, dbias (4H) f32, dwih (4H) f32, dx (R,T) bf16 or undefined}.
-std::vector<torch::Tensor> lstm_fused_bwd(torch::Tensor x, torch::Tensor whh,
+std::vector<torch::Tensor> lstm_fused_bwd(torch::Tensor x, long T_logical,
+                                          torch::Tensor whh,
                                           torch::Tensor whh2, torch::Tensor wih,
                                           torch::Tensor bias, torch::Tensor dh,
                                           bool need_dx) {
@@ -332,8 +334,9 @@ std::vector<torch::Tensor> lstm_fused_bwd(torch::Tensor x, torch::Tensor whh,
     check_in(whh, "whh");
     check_in(whh2, "whh2");
     check_in(dh, "dh");
+    TORCH_CHECK(x.size(1) == 8, "x must be zero-padded to 8 columns");
     const long R = x.size(0);
-    const int T = (int)x.size(1);
+    const int T = (int)T_logical;
     const int nb = lstm_fused_bwd_blocks(R);
     auto f32 = x.options().dtype(torch::kFloat);
     auto ws_dw = torch::empty({nb, 128, 32}, f32);

@@ -41,6 +41,8 @@ __device__ __forceinline__ void lds_wave_fence() {
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 }
 
+#define LF_XCOLS 8  // x is padded to 8 columns so each row is one 16-B load
+
 // gate activation order: i, f, g, o (torch chunk order)
 __device__ __forceinline__ float lf_act(float v, int gate) {
     return (gate == 2) ? fast_tanh(v) : fast_sigmoid(v);
@@ -76,10 +78,25 @@ __launch_bounds__(256) __global__ void lstm_fused_fwd_kernel(LstmFusedParams p) 
     const int w = tid / WAVE, lane = tid % WAVE;
     const int lrow = lane & 15, kgrp = lane >> 4;
     __bf16* myH = &ldsH[w][0];
+    // hoist the per-lane w_ih/bias values out of the step loops (they are
+    // otherwise re-read from global memory every step)
+    float wih_r[8], bias_r[8];
+#pragma unroll
+    for (int nf = 0; nf < 8; ++nf) {
+        wih_r[nf] = p.wih[nf * 16 + lrow];
+        bias_r[nf] = p.bias[nf * 16 + lrow];
+    }
 
     const long ntiles = (p.R + 63) / 64;
     for (long tile = blockIdx.x; tile < ntiles; tile += gridDim.x) {
         const long r0 = tile * 64 + w * 16;
+        alignas(16) __bf16 xrow[4][LF_XCOLS];
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            const long m = r0 + kgrp * 4 + r;
+            if (m < p.R) *(Chunk16*)&xrow[r][0] = *(const Chunk16*)&X[m * LF_XCOLS];
+            else for (int t2 = 0; t2 < LF_XCOLS; ++t2) xrow[r][t2] = (__bf16)0.f;
+        }
         lf_frag h_frag = {};
         float c[2][4] = {};  // c[jf][r] for j = jf*16 + lrow, row = r0 + kgrp*4 + r
 #pragma unroll
@@ -89,14 +106,12 @@ __launch_bounds__(256) __global__ void lstm_fused_fwd_kernel(LstmFusedParams p) 
             // activations + state update (C-layout)
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
-                const long m = r0 + kgrp * 4 + r;
-                const float xv = (m < p.R) ? to_f32(X[m * T + t]) : 0.f;
+                const float xv = to_f32(xrow[r][t]);
                 float gv[8];
 #pragma unroll
-                for (int nf = 0; nf < 8; ++nf) {
-                    const int n = nf * 16 + lrow;
-                    gv[nf] = lf_act(acc[nf][r] + xv * p.wih[n] + p.bias[n], n / LF_H);
-                }
+                for (int nf = 0; nf < 8; ++nf)
+                    gv[nf] = lf_act(acc[nf][r] + xv * wih_r[nf] + bias_r[nf],
+                                    (nf * 16 + lrow) / LF_H);
 #pragma unroll
                 for (int jf = 0; jf < 2; ++jf) {
                     const float c_new = gv[2 + jf] * c[jf][r] + gv[0 + jf] * gv[4 + jf];
@@ -152,6 +167,12 @@ __launch_bounds__(256) __global__ void lstm_fused_bwd_kernel(LstmFusedParams p) 
     __bf16* myHT = &ldsHT[w][0];
     __bf16* myH = &ldsHT[w][0];  // phase-A transpose scratch aliases the hT image
     float* myC = &ldsC[w][0];
+    float wih_r[8], bias_r[8];
+#pragma unroll
+    for (int nf = 0; nf < 8; ++nf) {
+        wih_r[nf] = p.wih[nf * 16 + lrow];
+        bias_r[nf] = p.bias[nf * 16 + lrow];
+    }
 
     // per-wave dW accumulator: D[i=n][j=k], 8 m-frags x 2 j-frags
     f32x4 dw_acc[8][2] = {};
@@ -160,6 +181,13 @@ __launch_bounds__(256) __global__ void lstm_fused_bwd_kernel(LstmFusedParams p) 
     const long ntiles = (p.R + 63) / 64;
     for (long tile = blockIdx.x; tile < ntiles; tile += gridDim.x) {
         const long r0 = tile * 64 + w * 16;
+        alignas(16) __bf16 xrow[4][LF_XCOLS];
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            const long m = r0 + kgrp * 4 + r;
+            if (m < p.R) *(Chunk16*)&xrow[r][0] = *(const Chunk16*)&X[m * LF_XCOLS];
+            else for (int t2 = 0; t2 < LF_XCOLS; ++t2) xrow[r][t2] = (__bf16)0.f;
+        }
         // ---- phase A: forward recompute; h states in registers, c in LDS ----
         lf_frag h_states[T];  // h AFTER step t (A-frag layout)
 #define CST(t, r, jf) myC[(t) * 16 * 33 + (kgrp * 4 + (r)) * 33 + (jf) * 16 + lrow]
@@ -172,14 +200,12 @@ __launch_bounds__(256) __global__ void lstm_fused_bwd_kernel(LstmFusedParams p) 
                 lf_gate_mfma(ldsW, h_frag, lrow, kgrp, acc);
 #pragma unroll
                 for (int r = 0; r < 4; ++r) {
-                    const long m = r0 + kgrp * 4 + r;
-                    const float xv = (m < p.R) ? to_f32(X[m * T + t]) : 0.f;
+                    const float xv = to_f32(xrow[r][t]);
                     float gv[8];
 #pragma unroll
-                    for (int nf = 0; nf < 8; ++nf) {
-                        const int n = nf * 16 + lrow;
-                        gv[nf] = lf_act(acc[nf][r] + xv * p.wih[n] + p.bias[n], n / LF_H);
-                    }
+                    for (int nf = 0; nf < 8; ++nf)
+                        gv[nf] = lf_act(acc[nf][r] + xv * wih_r[nf] + bias_r[nf],
+                                        (nf * 16 + lrow) / LF_H);
 #pragma unroll
                     for (int jf = 0; jf < 2; ++jf) {
                         const float c_new = gv[2 + jf] * c[jf][r] + gv[0 + jf] * gv[4 + jf];
@@ -222,13 +248,12 @@ __launch_bounds__(256) __global__ void lstm_fused_bwd_kernel(LstmFusedParams p) 
             for (int r = 0; r < 4; ++r) {
                 const long m = r0 + kgrp * 4 + r;
                 const bool ok = m < p.R;
-                const float xv = ok ? to_f32(X[m * T + t]) : 0.f;
+                const float xv = to_f32(xrow[r][t]);
                 float gv[8];
 #pragma unroll
-                for (int nf = 0; nf < 8; ++nf) {
-                    const int n = nf * 16 + lrow;
-                    gv[nf] = lf_act(acc[nf][r] + xv * p.wih[n] + p.bias[n], n / LF_H);
-                }
+                for (int nf = 0; nf < 8; ++nf)
+                    gv[nf] = lf_act(acc[nf][r] + xv * wih_r[nf] + bias_r[nf],
+                                    (nf * 16 + lrow) / LF_H);
 #pragma unroll
                 for (int jf = 0; jf < 2; ++jf) {
                     const float i_g = gv[0 + jf], f_g = gv[2 + jf];
@@ -262,7 +287,7 @@ __launch_bounds__(256) __global__ void lstm_fused_bwd_kernel(LstmFusedParams p) 
                     float part = 0.f;
 #pragma unroll
                     for (int nf = 0; nf < 8; ++nf)
-                        part += dgp[nf][r] * p.wih[nf * 16 + lrow];
+                        part += dgp[nf][r] * wih_r[nf];
                     // reduce over the 16 lrow lanes (same row across lrow)
 #pragma unroll
                     for (int s = 1; s < 16; s <<= 1)
