@@ -5,18 +5,18 @@
 // kernel serves every contraction of the factored BDGCN layer (see
 // mpgcn_amd/ops/eager.py for the algebra; reference call sites MPGCN.py:28-50):
 //
-//   mode-1 fwd : U[b,m,d,o,l] = sum_n  Go^T[m,n]      X[b,n,d,l]       (K1)
-//   mode-2 fwd : Y[b,m,d,h]   = sum_cs A2T[d,cs]      V[b,m,cs,h]      (K2, +bias+ReLU)
-//   bwd dV     : dV[b,m,cs,h] = sum_d  A2[cs,d]       dY[b,m,d,h]
-//   bwd dX     : dX[b,n,d,l]  = sum_om A3T[n,om]      dU[b,m,d,o,l]
+//   mode-1 fwd : U[b,m,d,o,l]  = sum_n  Go^T[m,n]   X[b,n,d,l]       (K1)
+//   mode-2 fwd : Y[b,m,d,h]    = sum_cs A2T[d,cs]   V[b,m,cs,h]      (K2, +bias+ReLU)
+//   bwd dV     : dV[b,m,cs,h]  = sum_d  A2[cs,d]    dY[b,m,d,h]
+//   bwd dX     : dX[b,n,d,l]   = sum_om A3T[n,om]   dU[b,m,d,o,l]
 //
-// The graph operand AT is always passed (M, K) row-major (k contiguous) — the
-// Python layer pre-permutes the tiny graph tensors — so the A-tile stages into
-// LDS as a straight vectorized copy; the X-tile is transposed into an
-// [n][k]-major LDS image during staging so MFMA fragments on both operands are
-// contiguous ds_read_b128 (bf16) reads. LDS rows are padded by 16 B: the
-// resulting 144 B row stride makes 16-lane b128 fragment reads bank-conflict-free
-// (row*36 mod 64 covers all 64 banks).
+// Structure: double-buffered LDS tiles with next-tile staging issued BEFORE
+// the current tile's MFMAs and ONE barrier per K-tile (the 2-phase schedule of
+// the CDNA GEMM playbook). The graph operand AT is (M, K) row-major (the
+// Python layer pre-permutes the tiny graph tensors), so the A-tile stages as a
+// straight vectorized copy; the X-tile transposes into an [n][k]-major image
+// with lane-rotated element order (spreads write banks over 4*{0..7}). LDS
+// rows are padded 16 B so 16-lane b128 fragment reads are conflict-free.
 #include "common.hpp"
 #include "params.hpp"
 
@@ -34,14 +34,14 @@ __device__ __forceinline__ long ocol_off(const AxisGemmParams& p, int q) {
 template <typename T, int BM, int BN, int BK, int WVM, int WVN>
 __launch_bounds__(256) __global__ void axis_gemm_kernel(AxisGemmParams p) {
     using MT = MfmaTraits<T>;
-    constexpr int CH = 16 / sizeof(T);  // elements per 16-byte chunk
+    constexpr int CH = 16 / sizeof(T);
     constexpr int PAD = MT::LDS_PAD;
     constexpr int WM = BM / WVM, WN = BN / WVN;
     constexpr int AM = WM / 16, AN = WN / 16;
     static_assert(WVM * WVN == 4, "4 waves per block");
 
-    __shared__ T ldsA[BM][BK + PAD];
-    __shared__ T ldsB[BN][BK + PAD];
+    __shared__ T ldsA[2][BM][BK + PAD];
+    __shared__ T ldsB[2][BN][BK + PAD];
 
     const T* __restrict__ A = (const T*)p.AT;
     const T* __restrict__ X = (const T*)p.X;
@@ -64,9 +64,9 @@ __launch_bounds__(256) __global__ void axis_gemm_kernel(AxisGemmParams p) {
 
     f32x4 acc[AM][AN] = {};
 
-    for (int kt = 0; kt < p.K; kt += BK) {
-        // ---- stage A tile (straight copy, k contiguous) ----
-        constexpr int A_CPR = BK / CH;  // chunks per row
+    auto stage = [&](int buf, int kt) {
+        // ---- A tile: straight vectorized copy (k contiguous) ----
+        constexpr int A_CPR = BK / CH;
         for (int idx = tid; idx < BM * A_CPR; idx += 256) {
             const int row = idx / A_CPR, cc = idx % A_CPR;
             const int m = m0 + row, k = kt + cc * CH;
@@ -80,9 +80,9 @@ __launch_bounds__(256) __global__ void axis_gemm_kernel(AxisGemmParams p) {
                     for (int i = 0; i < CH; ++i) d[i] = (k + i < p.K) ? src[i] : (T)0.f;
                 }
             }
-            *(Chunk16*)&ldsA[row][cc * CH] = val;
+            *(Chunk16*)&ldsA[buf][row][cc * CH] = val;
         }
-        // ---- stage X tile, transposed to [n][k] ----
+        // ---- X tile, transposed to [n][k], lane-rotated write order ----
         constexpr int B_CPR = BN / CH;
         for (int idx = tid; idx < BK * B_CPR; idx += 256) {
             const int krow = idx / B_CPR, qc = idx % B_CPR;
@@ -93,35 +93,38 @@ __launch_bounds__(256) __global__ void axis_gemm_kernel(AxisGemmParams p) {
             if (k < p.K) {
                 const long rbase = x_base + xrow_off(p, k);
                 if (p.x_vec && q0 + CH <= p.L) {
-                    Chunk16 c = *(const Chunk16*)(X + rbase + xcol_off(p, q0));
-                    *(Chunk16*)tmp = c;
+                    *(Chunk16*)tmp = *(const Chunk16*)(X + rbase + xcol_off(p, q0));
                 } else {
                     for (int i = 0; i < CH; ++i)
                         if (q0 + i < p.L) tmp[i] = X[rbase + xcol_off(p, q0 + i)];
                 }
             }
-            // lane-rotated write order de-conflicts the transpose (see
-            // red_gemm.hip): stride between lanes' rows is ~0 mod 32 banks,
-            // rotating the element index spreads each instruction's writes
 #pragma unroll
             for (int i = 0; i < CH; ++i) {
                 const int j = (i + tid) % CH;
-                ldsB[qc * CH + j][krow] = tmp[j];
+                ldsB[buf][qc * CH + j][krow] = tmp[j];
             }
         }
-        __syncthreads();
+    };
 
-        // ---- MFMA over the tile ----
+    stage(0, 0);
+    __syncthreads();
+
+    const int ktiles = (p.K + BK - 1) / BK;
+    int cur = 0;
+    for (int t = 0; t < ktiles; ++t) {
+        if (t + 1 < ktiles) stage(cur ^ 1, (t + 1) * BK);  // issue next tile first
+#pragma unroll
         for (int kk = 0; kk < BK; kk += MT::MFMA_K) {
             typename MT::frag_t af[AM], bf[AN];
 #pragma unroll
             for (int mf = 0; mf < AM; ++mf)
                 af[mf] = *(const typename MT::frag_t*)
-                    &ldsA[wm + mf * 16 + lrow][kk + kgrp * MT::FRAG_ELEMS];
+                    &ldsA[cur][wm + mf * 16 + lrow][kk + kgrp * MT::FRAG_ELEMS];
 #pragma unroll
             for (int nf = 0; nf < AN; ++nf)
                 bf[nf] = *(const typename MT::frag_t*)
-                    &ldsB[wn + nf * 16 + lrow][kk + kgrp * MT::FRAG_ELEMS];
+                    &ldsB[cur][wn + nf * 16 + lrow][kk + kgrp * MT::FRAG_ELEMS];
 #pragma unroll
             for (int mf = 0; mf < AM; ++mf)
 #pragma unroll
@@ -129,6 +132,7 @@ __launch_bounds__(256) __global__ void axis_gemm_kernel(AxisGemmParams p) {
                     acc[mf][nf] = MT::mfma(af[mf], bf[nf], acc[mf][nf]);
         }
         __syncthreads();
+        cur ^= 1;
     }
 
     // ---- epilogue: bias + activation + strided store ----
@@ -154,19 +158,23 @@ __launch_bounds__(256) __global__ void axis_gemm_kernel(AxisGemmParams p) {
 extern "C" void axis_gemm_launch(AxisGemmParams p, int instances, int is_f32,
                                  hipStream_t stream) {
     constexpr int BM = 128, BK = 64;
-    const int BN = (p.L >= 48) ? 64 : 32;
+    int BN = (p.L >= 96 && p.M >= 96) ? 128 : (p.L >= 48 ? 64 : 32);
+    if (is_f32 && BN > 64) BN = 64;  // f32 LDS budget caps the tile
     const int tiles_m = (p.M + BM - 1) / BM;
     p.tiles_l = (p.L + BN - 1) / BN;
     dim3 grid(tiles_m * p.tiles_l, instances), block(256);
     if (!is_f32) {
-        if (BN == 64)
+        if (BN == 128)
+            axis_gemm_kernel<__bf16, BM, 128, BK, 2, 2><<<grid, block, 0, stream>>>(p);
+        else if (BN == 64)
             axis_gemm_kernel<__bf16, BM, 64, BK, 2, 2><<<grid, block, 0, stream>>>(p);
         else
             axis_gemm_kernel<__bf16, BM, 32, BK, 4, 1><<<grid, block, 0, stream>>>(p);
     } else {
-        if (BN == 64)
-            axis_gemm_kernel<float, BM, 64, BK, 2, 2><<<grid, block, 0, stream>>>(p);
+        // f32: halve BK to keep the double-buffered LDS within budget
+        if (BN >= 64)
+            axis_gemm_kernel<float, BM, 64, 32, 2, 2><<<grid, block, 0, stream>>>(p);
         else
-            axis_gemm_kernel<float, BM, 32, BK, 4, 1><<<grid, block, 0, stream>>>(p);
+            axis_gemm_kernel<float, BM, 32, 32, 4, 1><<<grid, block, 0, stream>>>(p);
     }
 }
