@@ -83,17 +83,25 @@ __launch_bounds__(256) __global__ void axis_gemm_kernel(AxisGemmParams p) {
             *(Chunk16*)&ldsA[buf][row][cc * CH] = val;
         }
         // ---- X tile, transposed to [n][k], lane-rotated write order ----
+        // 256 % B_CPR == 0, so each thread's column chunk qc is CONSTANT across
+        // its grid-stride chunks: the q-side division in xcol_off hoists out of
+        // the loop entirely (it otherwise dominates issue time as a magic-
+        // number division sequence per chunk per K-tile).
         constexpr int B_CPR = BN / CH;
+        static_assert(256 % B_CPR == 0);
+        const int qc = tid % B_CPR;
+        const int q0 = l0 + qc * CH;
+        const bool qvec = p.x_vec && q0 + CH <= p.L;
+        const long xq = qvec ? xcol_off(p, q0) : 0;
         for (int idx = tid; idx < BK * B_CPR; idx += 256) {
-            const int krow = idx / B_CPR, qc = idx % B_CPR;
+            const int krow = idx / B_CPR;
             const int k = kt + krow;
-            const int q0 = l0 + qc * CH;
             alignas(16) T tmp[CH];
             for (int i = 0; i < CH; ++i) tmp[i] = (T)0.f;
             if (k < p.K) {
                 const long rbase = x_base + xrow_off(p, k);
-                if (p.x_vec && q0 + CH <= p.L) {
-                    *(Chunk16*)tmp = *(const Chunk16*)(X + rbase + xcol_off(p, q0));
+                if (qvec) {
+                    *(Chunk16*)tmp = *(const Chunk16*)(X + rbase + xq);
                 } else {
                     for (int i = 0; i < CH; ++i)
                         if (q0 + i < p.L) tmp[i] = X[rbase + xcol_off(p, q0 + i)];
@@ -136,19 +144,23 @@ __launch_bounds__(256) __global__ void axis_gemm_kernel(AxisGemmParams p) {
     }
 
     // ---- epilogue: bias + activation + strided store ----
+    // per-lane q depends only on nf: hoist the column-offset division and the
+    // bias load out of the (mf, r) loops
 #pragma unroll
-    for (int mf = 0; mf < AM; ++mf) {
+    for (int nf = 0; nf < AN; ++nf) {
+        const int q = l0 + wn + nf * 16 + lrow;
+        if (q >= p.L) continue;
+        const long oc = o_base + ocol_off(p, q);
+        const float bv = p.bias ? p.bias[p.bias_mod ? q % p.bias_mod : q] : 0.f;
 #pragma unroll
-        for (int nf = 0; nf < AN; ++nf) {
+        for (int mf = 0; mf < AM; ++mf) {
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 const int m = m0 + wm + mf * 16 + kgrp * 4 + r;
-                const int q = l0 + wn + nf * 16 + lrow;
-                if (m < p.M && q < p.L) {
-                    float v = acc[mf][nf][r];
-                    if (p.bias) v += p.bias[p.bias_mod ? q % p.bias_mod : q];
+                if (m < p.M) {
+                    float v = acc[mf][nf][r] + bv;
                     if (p.relu) v = fmaxf(v, 0.f);
-                    O[o_base + (long)m * p.o_row + ocol_off(p, q)] = from_f32<T>(v);
+                    O[oc + (long)m * p.o_row] = from_f32<T>(v);
                 }
             }
         }
