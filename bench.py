@@ -34,6 +34,9 @@ def main():
     ap.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "float32"])
     ap.add_argument("--device", type=str, default=None,
                     help="override device (cpu for debug)")
+    ap.add_argument("--impl", type=str, default="native", choices=["native", "eager"],
+                    help="'eager' runs the reference-math fp32 transcription "
+                         "(stock torch ops, K^2-pair formulation) as the baseline")
     args = ap.parse_args()
 
     import torch.distributed as dist
@@ -63,9 +66,16 @@ def main():
     S = K_order + 1
     cdtype = torch.bfloat16 if (args.dtype == "bf16" and is_cuda) else torch.float32
 
-    model = MPGCN(M=2, K=S, input_dim=1, lstm_hidden_dim=H, lstm_num_layers=1,
-                  gcn_hidden_dim=H, gcn_num_layers=3, num_nodes=N,
-                  compute_dtype=cdtype).to(device)
+    if args.impl == "eager":
+        from mpgcn_amd.models.reference_eager import MPGCNReference
+
+        cdtype = torch.float32  # the reference implementation is fp32 eager
+        model = MPGCNReference(M=2, K=S, input_dim=1, hidden=H, gcn_layers=3,
+                               num_nodes=N).to(device)
+    else:
+        model = MPGCN(M=2, K=S, input_dim=1, lstm_hidden_dim=H, lstm_num_layers=1,
+                      gcn_hidden_dim=H, gcn_num_layers=3, num_nodes=N,
+                      compute_dtype=cdtype).to(device)
     opt = torch.optim.Adam(model.parameters(), lr=1e-4)
     reducer = GradAllReducer(model, ctx)
     criterion = torch.nn.MSELoss()
@@ -133,7 +143,7 @@ def main():
             "dtype": "bf16" if cdtype == torch.bfloat16 else "float32",
             "data": "synthetic",
             "config": {
-                "model": "MPGCN",
+                "model": "MPGCN" if args.impl == "native" else "MPGCN-reference-eager",
                 "regions": N,
                 "global_batch": global_batch,
                 "seq_len": T,
