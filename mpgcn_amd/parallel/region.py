@@ -1,0 +1,168 @@
+"""Region partition: shard the N x N OD activation grid across GPUs.
+
+The reference is single-device; at 4096 regions a single
+(B, N, N, C) activation is ~4 GB and the per-layer intermediates several times
+that, so the activation grid itself must shard (weights are tiny and stay
+replicated — the model is activation-heavy).
+
+Scheme (the all-to-all formulation of SURVEY.md §7 "hard parts"):
+  * activations live DESTINATION-sharded: X_p = X[:, :, d_p : d_p + N/P, :]
+    — each rank holds all origin rows for a slice of destination columns;
+  * mode-1 (contract the ORIGIN axis) is then fully local;
+  * the projection GEMM is row-local;
+  * before mode-2 (contract the DESTINATION axis) an all-to-all re-shards
+    from destination-sharded to ORIGIN-sharded (each rank: all destinations
+    for a slice of origin rows) — mode-2 becomes local;
+  * a second all-to-all re-shards back for the next layer's mode-1.
+  Per BDGCN layer: 2 all-to-alls of O(B*N^2*C/P) bytes per rank over xGMI —
+  the bandwidth-critical collective of this workload.
+
+The LSTM/FC stages are pointwise over (origin, destination) pairs and run on
+any sharding. Losses are computed shard-locally; weight gradients all-reduce
+through the usual GradAllReducer (weights replicated). The all-to-alls are
+autograd-aware (backward = the inverse all-to-all), so the whole sharded
+forward trains with plain autograd.
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.distributed as dist
+
+from mpgcn_amd.ops import eager, fused_lstm_last, linear_act
+
+
+class _AllToAllShard(torch.autograd.Function):
+    """Differentiable all-to-all that converts a destination-sharded tensor
+    (B, N, Nl, F) into an origin-sharded one (B, Nl, N, F) or back.
+
+    Forward 'd2o': input  (B, N, N/P, F)  ->  output (B, N/P, N, F)
+    Forward 'o2d': input  (B, N/P, N, F)  ->  output (B, N, N/P, F)
+    Backward is the opposite direction (all-to-all is self-adjoint up to the
+    permutation).
+    """
+
+    @staticmethod
+    def forward(ctx, x, direction: str, group):
+        ctx.direction = direction
+        ctx.group = group
+        return _a2a(x, direction, group)
+
+    @staticmethod
+    def backward(ctx, g):
+        inv = "o2d" if ctx.direction == "d2o" else "d2o"
+        return _a2a(g.contiguous(), inv, ctx.group), None, None
+
+
+def _a2a(x: torch.Tensor, direction: str, group) -> torch.Tensor:
+    P = dist.get_world_size(group)
+    if direction == "d2o":
+        B, N, Nl, F = x.shape
+        assert N == Nl * P, (N, Nl, P)
+        # send chunk q = origin rows [q*Nl, (q+1)*Nl) of the local dest slice
+        send = x.reshape(B, P, Nl, Nl, F).permute(1, 0, 2, 3, 4).contiguous()
+        recv = torch.empty_like(send)
+        dist.all_to_all_single(recv, send, group=group)
+        # recv[p] = origin rows (local) x dest cols of peer p
+        out = recv.permute(1, 2, 0, 3, 4).reshape(B, Nl, N, F)
+        return out.contiguous()
+    else:
+        B, Nl, N, F = x.shape
+        assert N == Nl * P, (N, Nl, P)
+        send = x.reshape(B, Nl, P, Nl, F).permute(2, 0, 1, 3, 4).contiguous()
+        recv = torch.empty_like(send)
+        dist.all_to_all_single(recv, send, group=group)
+        out = recv.permute(1, 0, 2, 3, 4).reshape(B, N, Nl, F)
+        return out.contiguous()
+
+
+def dest_to_origin(x, group=None):
+    """(B, N, N/P, F) destination-sharded -> (B, N/P, N, F) origin-sharded."""
+    return _AllToAllShard.apply(x, "d2o", group)
+
+
+def origin_to_dest(x, group=None):
+    """(B, N/P, N, F) origin-sharded -> (B, N, N/P, F) destination-sharded."""
+    return _AllToAllShard.apply(x, "o2d", group)
+
+
+def shard_dest(x_full: torch.Tensor, rank: int, P: int) -> torch.Tensor:
+    """Slice the destination axis of a full (B, ..., N, N, F) tensor."""
+    N = x_full.shape[-2]
+    Nl = N // P
+    return x_full[..., rank * Nl:(rank + 1) * Nl, :].contiguous()
+
+
+def bdgcn_layer_sharded(Xd, Go, Gd, W, bias, group=None, relu=True):
+    """One BDGCN layer on a destination-sharded input.
+
+    Xd: (B, N, N/P, C) destination-sharded; Go/Gd: full (S, N, N) or
+    (B, S, N, N) (graphs are O(S*N^2), replicated); W: (C*S*S, H).
+    Returns the next layer's destination-sharded input (B, N, N/P, H).
+
+    Math identical to mpgcn_amd.ops.eager.bdgcn_layer_eager (verified by
+    tests/test_region.py against the unsharded computation).
+    """
+    S = Go.shape[-3]
+    B, N, Nl, C = Xd.shape
+    Hdim = W.shape[1]
+
+    # mode-1: contract the (full, local) origin axis
+    if Go.dim() == 3:
+        U = torch.einsum("onm,bndl->bmdol", Go, Xd)  # (B, N, Nl, S, C)
+    else:
+        U = torch.einsum("bonm,bndl->bmdol", Go, Xd)
+    # projection (row-local)
+    Wre = eager.reorder_projection_weight(W, S, C)
+    V = (U.reshape(B * N * Nl, S * C) @ Wre).view(B, N, Nl, S * Hdim)
+    # re-shard: destination-sharded -> origin-sharded (full dest axis)
+    Vo = dest_to_origin(V, group)  # (B, Nl, N, S*H)
+    Vo = Vo.view(B, Nl, N, S, Hdim)
+    # mode-2: contract the (full, local) destination axis
+    if Gd.dim() == 3:
+        Y = torch.einsum("scd,bmcsh->bmdh", Gd, Vo)  # (B, Nl, N, H)
+    else:
+        Y = torch.einsum("bscd,bmcsh->bmdh", Gd, Vo)
+    if bias is not None:
+        Y = Y + bias.to(Y.dtype)
+    if relu:
+        Y = torch.relu(Y)
+    # re-shard back for the next layer's mode-1
+    return origin_to_dest(Y, group)  # (B, N, Nl, H)
+
+
+def mpgcn_forward_sharded(model, x_seq_shard, G_list, group=None):
+    """Full MPGCN forward on destination-sharded inputs.
+
+    model: an mpgcn_amd.models.MPGCN (weights replicated across ranks);
+    x_seq_shard: (B, T, N, N/P, 1) — this rank's destination slice;
+    G_list: full graphs, the usual [static (S,N,N), (O_dyn, D_dyn)] contract.
+    Returns the destination-sharded prediction (B, 1, N, N/P, 1).
+    """
+    B, T, N, Nl, _ = x_seq_shard.shape
+    gops = model._graph_operators(G_list)
+    cd = model.compute_dtype
+    lstm_in = (
+        x_seq_shard.to(cd).permute(0, 2, 3, 1, 4).reshape(B * N * Nl, T).contiguous()
+    )
+    outs = []
+    for m in range(model.M):
+        branch = model.branch_models[m]
+        h = fused_lstm_last(
+            lstm_in,
+            branch["temporal"].weight_ih_l0.to(cd),
+            branch["temporal"].weight_hh_l0.to(cd),
+            branch["temporal"].bias_ih_l0,
+            branch["temporal"].bias_hh_l0,
+        )
+        X = h.reshape(B, N, Nl, model.lstm_hidden_dim)
+        gop = gops[m]
+        for layer in branch["spatial"]:
+            X = bdgcn_layer_sharded(
+                X, gop.Go, gop.Gd, layer.W.to(cd), layer.b, group, relu=layer.relu
+            )
+        fc = branch["fc"][0]
+        out = linear_act(X.reshape(B * N * Nl, -1), fc.weight.to(cd), fc.bias, True)
+        outs.append(out.view(B, N, Nl, 1))
+    ens = torch.mean(torch.stack(outs, dim=-1), dim=-1)
+    return ens.float().unsqueeze(1)
