@@ -60,12 +60,19 @@ def build_parser() -> argparse.ArgumentParser:
     parser.add_argument("-shuffle", "--shuffle", action="store_true",
                         help="Shuffle training batches (reference default: off)")
     parser.add_argument("-seed", "--seed", type=int, default=0)
+    parser.add_argument("-partition", "--partition", type=str,
+                        choices=["dp", "region"], default="dp",
+                        help="multi-GPU strategy: data parallel (default) or "
+                             "region partition (shard the N x N activation grid)")
+    parser.add_argument("-resume", "--resume", action="store_true",
+                        help="resume training from the extended checkpoint "
+                             "({model}_od.resume.pkl) if present")
     return parser
 
 
 def main():
     from mpgcn_amd.data import DataGenerator, DataInput
-    from mpgcn_amd.parallel import init_distributed
+    from mpgcn_amd.parallel import init_distributed, rank_watchdog
     from mpgcn_amd.train import ModelTrainer
 
     params = build_parser().parse_args().__dict__
@@ -93,18 +100,23 @@ def main():
         obs_len=params["obs_len"], pred_len=params["pred_len"],
         data_split_ratio=params["split_ratio"],
     )
+    # region partition shards the activation grid, not the sample axis:
+    # every rank sees the full batch stream
+    dl_rank = ctx.rank if params["partition"] == "dp" else 0
+    dl_world = ctx.world_size if params["partition"] == "dp" else 1
     data_loader = data_generator.get_data_loader(
         data=data, params=params, device=device,
-        rank=ctx.rank, world_size=ctx.world_size,
+        rank=dl_rank, world_size=dl_world,
     )
 
     trainer = ModelTrainer(params=params, data=data,
                            data_container=data_input, dist_ctx=ctx)
 
-    if params["mode"] == "train":
-        trainer.train(data_loader=data_loader, modes=["train", "validate"])
-    else:
-        trainer.test(data_loader=data_loader, modes=["train", "test"])
+    with rank_watchdog(ctx):
+        if params["mode"] == "train":
+            trainer.train(data_loader=data_loader, modes=["train", "validate"])
+        else:
+            trainer.test(data_loader=data_loader, modes=["train", "test"])
 
 
 if __name__ == "__main__":

@@ -1,3 +1,3 @@
-from mpgcn_amd.parallel.ddp import DistContext, GradAllReducer, init_distributed
+from mpgcn_amd.parallel.ddp import (DistContext, GradAllReducer, init_distributed, rank_watchdog)
 
-__all__ = ["DistContext", "GradAllReducer", "init_distributed"]
+__all__ = ["DistContext", "GradAllReducer", "init_distributed", "rank_watchdog"]

@@ -14,8 +14,12 @@ and a single flat bucket per backward is the right shape.
 
 from __future__ import annotations
 
+import contextlib
 import os
+import sys
+import traceback
 from dataclasses import dataclass
+from datetime import timedelta
 
 import torch
 import torch.distributed as dist
@@ -37,9 +41,13 @@ class DistContext:
         return self.world_size > 1
 
 
-def init_distributed(device: str = "cuda", backend: str | None = None) -> DistContext:
+def init_distributed(device: str = "cuda", backend: str | None = None,
+                     timeout_s: int = 300) -> DistContext:
     """Initialize from torchrun env vars (RANK/WORLD_SIZE/LOCAL_RANK); no-op
-    single-process context when they are absent."""
+    single-process context when they are absent. `timeout_s` bounds every
+    collective: a dead or wedged rank turns into a clean RCCL timeout error
+    instead of an indefinite hang (the failure-detection scope SURVEY.md §5
+    prescribes for single-node 8-GPU)."""
     if "RANK" not in os.environ or "WORLD_SIZE" not in os.environ:
         return DistContext()
     rank = int(os.environ["RANK"])
@@ -52,8 +60,30 @@ def init_distributed(device: str = "cuda", backend: str | None = None) -> DistCo
     if not dist.is_initialized():
         if backend == "nccl":
             torch.cuda.set_device(local)
-        dist.init_process_group(backend=backend)
+        dist.init_process_group(backend=backend, timeout=timedelta(seconds=timeout_s))
     return DistContext(rank=rank, world_size=world, local_rank=local, backend=backend)
+
+
+@contextlib.contextmanager
+def rank_watchdog(ctx: DistContext):
+    """Clean-abort wrapper for distributed runs: on any exception, log the
+    failing rank with its traceback, tear the process group down (so peers get
+    a fast connection error instead of waiting out the collective timeout) and
+    exit non-zero."""
+    try:
+        yield
+    except Exception:
+        sys.stderr.write(
+            f"[mpgcn] rank {ctx.rank}/{ctx.world_size} failed:\n"
+            + traceback.format_exc()
+        )
+        sys.stderr.flush()
+        if ctx.enabled and dist.is_initialized():
+            with contextlib.suppress(Exception):
+                dist.destroy_process_group()
+        if ctx.enabled:
+            os._exit(1)
+        raise
 
 
 class GradAllReducer:

@@ -15,6 +15,7 @@ MI355X deltas (deliberate):
 
 from __future__ import annotations
 
+import os
 import time
 from datetime import datetime
 
@@ -26,6 +27,8 @@ from mpgcn_amd.graph import build_supports, get_support_K
 from mpgcn_amd.models import MPGCN
 from mpgcn_amd.parallel import DistContext, GradAllReducer
 from mpgcn_amd.train import metrics as metrics_mod
+from mpgcn_amd.utils import checkpoint as ckpt_io
+from mpgcn_amd.utils.profiling import ThroughputMeter, trace_range
 
 
 class ModelTrainer:
@@ -41,6 +44,7 @@ class ModelTrainer:
             else torch.float32
         )
 
+        self.partition = params.get("partition", "dp")
         self.K = get_support_K(params["kernel_type"], params["cheby_order"])
         self.G = self.preprocess_adj(data["adj"])
         self.model = self.get_model().to(self.device)
@@ -107,22 +111,52 @@ class ModelTrainer:
         return self.params["output_dir"] + f"/{self.params['model']}_prediction_scores.txt"
 
     def _forward(self, x_seq, O_dyn_G, D_dyn_G):
-        dyn = (
-            self.preprocess_dynamic_graph(O_dyn_G),
-            self.preprocess_dynamic_graph(D_dyn_G),
-        )
-        return self.model(x_seq=x_seq, G_list=[self.G, dyn])
+        with trace_range("dyn_supports"):
+            dyn = (
+                self.preprocess_dynamic_graph(O_dyn_G),
+                self.preprocess_dynamic_graph(D_dyn_G),
+            )
+        with trace_range("forward"):
+            return self.model(x_seq=x_seq, G_list=[self.G, dyn])
+
+    def _step_forward(self, x_seq, y_true, O_dyn_G, D_dyn_G):
+        """Forward + matching target; region partition shards the destination
+        axis across ranks (mpgcn_amd/parallel/region.py) — predictions and
+        targets come back shard-local, losses average to the full-batch loss."""
+        if self.partition == "region" and self.ctx.enabled:
+            from mpgcn_amd.parallel.region import mpgcn_forward_sharded, shard_dest
+
+            with trace_range("dyn_supports"):
+                dyn = (
+                    self.preprocess_dynamic_graph(O_dyn_G),
+                    self.preprocess_dynamic_graph(D_dyn_G),
+                )
+            xs = shard_dest(x_seq, self.ctx.rank, self.ctx.world_size)
+            ys = shard_dest(y_true, self.ctx.rank, self.ctx.world_size)
+            with trace_range("forward"):
+                return mpgcn_forward_sharded(self.model, xs, [self.G, dyn]), ys
+        return self._forward(x_seq, O_dyn_G, D_dyn_G), y_true
 
     # -- training loop (Model_Trainer.py:87-142 equivalent) --
     def train(self, data_loader: dict, modes: list, early_stop_patience: int = 10):
         checkpoint = {"epoch": 0, "state_dict": self.model.state_dict()}
         val_loss = np.inf
         patience_count = early_stop_patience
+        start_epoch = 1
+        if self.params.get("resume"):
+            rp = ckpt_io.resume_path(self.params["output_dir"], self.params["model"])
+            if os.path.exists(rp):
+                start_epoch, val_loss, patience_count = ckpt_io.load_resume(
+                    rp, self.model, self.optimizer
+                )
+                if self.ctx.is_main:
+                    print(f"[mpgcn] resumed from {rp} at epoch {start_epoch}")
         log = print if self.ctx.is_main else (lambda *a, **k: None)
+        meter = ThroughputMeter()
 
         log("\n", datetime.now().strftime("%Y/%m/%d %H:%M:%S"))
         log(f'     {self.params["model"]} model training begins:')
-        for epoch in range(1, 1 + self.params["num_epochs"]):
+        for epoch in range(start_epoch, 1 + self.params["num_epochs"]):
             running_loss = {mode: 0.0 for mode in modes}
             epoch_samples = 0
             t0 = time.time()
@@ -133,18 +167,24 @@ class ModelTrainer:
                 step = 0
                 for x_seq, y_true, O_dyn_G, D_dyn_G in data_loader[mode]:
                     with torch.set_grad_enabled(mode == "train"):
-                        y_pred = self._forward(x_seq, O_dyn_G, D_dyn_G)
-                        loss = self.criterion(y_pred, y_true)
+                        y_pred, y_tgt = self._step_forward(
+                            x_seq, y_true, O_dyn_G, D_dyn_G
+                        )
+                        loss = self.criterion(y_pred, y_tgt)
                         if mode == "train":
                             self.optimizer.zero_grad(set_to_none=True)
-                            loss.backward()
-                            self.reducer.finalize()
-                            self.optimizer.step()
+                            with trace_range("backward"):
+                                loss.backward()
+                            with trace_range("grad_allreduce"):
+                                self.reducer.finalize()
+                            with trace_range("optimizer"):
+                                self.optimizer.step()
                     bs = y_true.shape[0]
                     running_loss[mode] += loss.item() * bs
                     step += bs
                     if mode == "train":
                         epoch_samples += bs
+                        meter.step(bs * self.ctx.world_size)
 
                 if mode == "validate":
                     if self.ctx.enabled:
@@ -171,6 +211,14 @@ class ModelTrainer:
                         if self.ctx.is_main:
                             torch.save(checkpoint, self._ckpt_path())
                         patience_count = early_stop_patience
+                        if self.ctx.is_main and self.params.get("resume"):
+                            ckpt_io.save_resume(
+                                ckpt_io.resume_path(
+                                    self.params["output_dir"], self.params["model"]
+                                ),
+                                epoch, self.model, self.optimizer, val_loss,
+                                patience_count,
+                            )
                     else:
                         log(
                             f"Epoch {epoch}, validation loss does not improve from "
