@@ -82,46 +82,35 @@ __launch_bounds__(256) __global__ void axis_gemm_kernel(AxisGemmParams p) {
             }
             *(Chunk16*)&ldsA[buf][row][cc * CH] = val;
         }
-        // ---- X tile, transposed to [n][k] via in-register CHxCH block
-        // transposes: coalesced 16-B loads (CH rows of k at this thread's
-        // fixed q chunk), register transpose, then full b128 column writes
-        // with lane-rotated order (conflict-free banks). The q-side division
-        // in xcol_off hoists out entirely (thread's q chunk is constant).
-        constexpr int QB = BN / CH;
-        static_assert(256 % QB == 0);
-        const int qc = tid % QB;
+        // ---- X tile, transposed to [n][k], lane-rotated write order ----
+        // 256 % B_CPR == 0, so each thread's column chunk qc is CONSTANT across
+        // its grid-stride chunks: the q-side division in xcol_off hoists out of
+        // the loop entirely (it otherwise dominates issue time as a magic-
+        // number division sequence per chunk per K-tile).
+        constexpr int B_CPR = BN / CH;
+        static_assert(256 % B_CPR == 0);
+        const int qc = tid % B_CPR;
         const int q0 = l0 + qc * CH;
         const bool qvec = p.x_vec && q0 + CH <= p.L;
         const long xq = qvec ? xcol_off(p, q0) : 0;
-        for (int b = tid; b < (BK / CH) * QB; b += 256) {
-            const int kb = b / QB;  // this thread's qc is b % QB == tid % QB
-            const int k0 = kt + kb * CH;
-            Chunk16 rows[CH], cols[CH];
-#pragma unroll
-            for (int i = 0; i < CH; ++i) {
-                const int k = k0 + i;
-                if (k < p.K && qvec) {
-                    rows[i] = *(const Chunk16*)(X + x_base + xrow_off(p, k) + xq);
+        for (int idx = tid; idx < BK * B_CPR; idx += 256) {
+            const int krow = idx / B_CPR;
+            const int k = kt + krow;
+            alignas(16) T tmp[CH];
+            for (int i = 0; i < CH; ++i) tmp[i] = (T)0.f;
+            if (k < p.K) {
+                const long rbase = x_base + xrow_off(p, k);
+                if (qvec) {
+                    *(Chunk16*)tmp = *(const Chunk16*)(X + rbase + xq);
                 } else {
-                    T* d = (T*)&rows[i];
-                    if (k < p.K) {
-                        const long rbase = x_base + xrow_off(p, k);
-#pragma unroll
-                        for (int j = 0; j < CH; ++j)
-                            d[j] = (q0 + j < p.L) ? X[rbase + xcol_off(p, q0 + j)]
-                                                  : (T)0.f;
-                    } else {
-#pragma unroll
-                        for (int j = 0; j < CH; ++j) d[j] = (T)0.f;
-                    }
+                    for (int i = 0; i < CH; ++i)
+                        if (q0 + i < p.L) tmp[i] = X[rbase + xcol_off(p, q0 + i)];
                 }
             }
-            if constexpr (sizeof(T) == 2) transpose8x8_bf16(rows, cols);
-            else transpose4x4_f32(rows, cols);
 #pragma unroll
             for (int i = 0; i < CH; ++i) {
-                const int c = (i + tid) % CH;  // rotation de-conflicts banks
-                *(Chunk16*)&ldsB[buf][qc * CH + c][kb * CH] = cols[c];
+                const int j = (i + tid) % CH;
+                ldsB[buf][qc * CH + j][krow] = tmp[j];
             }
         }
     };

@@ -69,33 +69,6 @@ __device__ __forceinline__ float fast_tanh(float v) {
 // 16-byte raw copy chunk (8 bf16 / 4 f32).
 struct alignas(16) Chunk16 { int v[4]; };
 
-// In-register 8x8 bf16 transpose: in[r] holds row r (8 bf16 = 4 dwords),
-// out[c] gets column c. Each output dword merges the 16-bit halves of two
-// input rows (~2 VALU ops) — replaces per-element LDS scalar writes (which
-// bank-conflict heavily on transposed images) with full b128 writes.
-__device__ __forceinline__ void transpose8x8_bf16(const Chunk16 in[8],
-                                                  Chunk16 out[8]) {
-#pragma unroll
-    for (int c = 0; c < 8; ++c) {
-#pragma unroll
-        for (int d2 = 0; d2 < 4; ++d2) {
-            const unsigned A = (unsigned)in[2 * d2].v[c / 2];
-            const unsigned B = (unsigned)in[2 * d2 + 1].v[c / 2];
-            out[c].v[d2] = (int)((c & 1) ? ((A >> 16) | (B & 0xffff0000u))
-                                         : ((A & 0xffffu) | (B << 16)));
-        }
-    }
-}
-
-// 4x4 f32 transpose: pure register renaming (free).
-__device__ __forceinline__ void transpose4x4_f32(const Chunk16 in[4],
-                                                 Chunk16 out[4]) {
-#pragma unroll
-    for (int c = 0; c < 4; ++c)
-#pragma unroll
-        for (int r = 0; r < 4; ++r) out[c].v[r] = in[r].v[c];
-}
-
 #define HIP_CHECK(expr)                                              \
     do {                                                             \
         hipError_t _e = (expr);                                      \

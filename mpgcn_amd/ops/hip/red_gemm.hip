@@ -16,43 +16,6 @@
 #include "common.hpp"
 #include "params.hpp"
 
-
-// Stage a (RCH x K) row-major chunk transposed into lds[KMAX][RCH+PAD] using
-// CHxCH in-register block transposes. Tail rows/columns zero-fill.
-template <typename T, int KMAX, int RCH>
-__device__ __forceinline__ void stage_T(const T* __restrict__ X, int K, long R,
-                                        long r0, int x_vec, T* lds, int tid) {
-    constexpr int CH = 16 / (int)sizeof(T);
-    constexpr int PAD = MfmaTraits<T>::LDS_PAD;
-    constexpr int LD = RCH + PAD;
-    constexpr int KB = KMAX / CH;
-    constexpr int NBLK = (RCH / CH) * KB;
-    for (int b = tid; b < NBLK; b += 256) {
-        const int kb = b % KB, rb = b / KB;  // consecutive threads: consecutive k
-        const int k0 = kb * CH, rl = rb * CH;
-        Chunk16 rows[CH], cols[CH];
-#pragma unroll
-        for (int i = 0; i < CH; ++i) {
-            const long row = r0 + rl + i;
-            if (row < R && x_vec && k0 + CH <= K) {
-                rows[i] = *(const Chunk16*)&X[row * K + k0];
-            } else {
-                T* d = (T*)&rows[i];
-#pragma unroll
-                for (int j = 0; j < CH; ++j)
-                    d[j] = (row < R && k0 + j < K) ? X[row * K + k0 + j] : (T)0.f;
-            }
-        }
-        if constexpr (sizeof(T) == 2) transpose8x8_bf16(rows, cols);
-        else transpose4x4_f32(rows, cols);
-#pragma unroll
-        for (int i = 0; i < CH; ++i) {
-            const int c = (i + tid) % CH;  // rotation de-conflicts write banks
-            *(Chunk16*)&lds[(k0 + c) * LD + rl] = cols[c];
-        }
-    }
-}
-
 template <typename T, int AKF, int ANF>
 __launch_bounds__(256) __global__ void red_gemm_kernel(RedGemmParams p) {
     using MT = MfmaTraits<T>;
@@ -78,12 +41,48 @@ __launch_bounds__(256) __global__ void red_gemm_kernel(RedGemmParams p) {
     float cs = 0.f, xd = 0.f;
 
     for (long r0 = (long)blockIdx.x * RCH; r0 < p.R; r0 += (long)gridDim.x * RCH) {
-        // stage X and Y transposed via in-register CHxCH block transposes:
-        // coalesced 16-B row loads -> register transpose -> full b128 column
-        // writes with lane-rotated order (banks 4*{0..CH-1}, conflict-free) —
-        // no per-element LDS scatter.
-        stage_T<T, KMAX, RCH>(X, p.K, p.R, r0, p.x_vec, &ldsXT[0][0], tid);
-        stage_T<T, NMAX, RCH>(Y, p.N, p.R, r0, p.y_vec, &ldsYT[0][0], tid);
+        // stage X chunk transposed: ldsXT[k][r]
+        for (int idx = tid; idx < RCH * (KMAX / CH); idx += 256) {
+            const int r = idx / (KMAX / CH), c = idx % (KMAX / CH);
+            const int k0 = c * CH;
+            const long row = r0 + r;
+            alignas(16) T tmp[CH];
+            for (int i = 0; i < CH; ++i) tmp[i] = (T)0.f;
+            if (row < p.R) {
+                if (p.x_vec && k0 + CH <= p.K)
+                    *(Chunk16*)tmp = *(const Chunk16*)&X[row * p.K + k0];
+                else
+                    for (int i = 0; i < CH; ++i)
+                        if (k0 + i < p.K) tmp[i] = X[row * p.K + k0 + i];
+            }
+            // lane-rotated write order: consecutive lanes write different
+            // LDS rows per instruction (banks 4*{0..7}, no 16-way conflict)
+#pragma unroll
+            for (int i = 0; i < CH; ++i) {
+                const int j = (i + tid) % CH;
+                ldsXT[k0 + j][r] = tmp[j];
+            }
+        }
+        // stage Y chunk transposed: ldsYT[n][r]
+        for (int idx = tid; idx < RCH * (NMAX / CH); idx += 256) {
+            const int r = idx / (NMAX / CH), c = idx % (NMAX / CH);
+            const int n0 = c * CH;
+            const long row = r0 + r;
+            alignas(16) T tmp[CH];
+            for (int i = 0; i < CH; ++i) tmp[i] = (T)0.f;
+            if (row < p.R) {
+                if (p.y_vec && n0 + CH <= p.N)
+                    *(Chunk16*)tmp = *(const Chunk16*)&Y[row * p.N + n0];
+                else
+                    for (int i = 0; i < CH; ++i)
+                        if (n0 + i < p.N) tmp[i] = Y[row * p.N + n0 + i];
+            }
+#pragma unroll
+            for (int i = 0; i < CH; ++i) {
+                const int j = (i + tid) % CH;
+                ldsYT[n0 + j][r] = tmp[j];
+            }
+        }
         if (XV) {
             for (int r = tid; r < RCH; r += 256) {
                 const long row = r0 + r;
