@@ -31,7 +31,7 @@ __device__ __forceinline__ long ocol_off(const AxisGemmParams& p, int q) {
     return p.ogdiv ? (long)(q / p.ogdiv) * p.og_hi + (long)(q % p.ogdiv) : (long)q;
 }
 
-template <typename T, int BM, int BN, int BK, int WVM, int WVN>
+template <typename T, int BM, int BN, int BK, int WVM, int WVN, int BUFS = 2>
 __launch_bounds__(256) __global__ void axis_gemm_kernel(AxisGemmParams p) {
     using MT = MfmaTraits<T>;
     constexpr int CH = 16 / sizeof(T);
@@ -40,8 +40,8 @@ __launch_bounds__(256) __global__ void axis_gemm_kernel(AxisGemmParams p) {
     constexpr int AM = WM / 16, AN = WN / 16;
     static_assert(WVM * WVN == 4, "4 waves per block");
 
-    __shared__ T ldsA[2][BM][BK + PAD];
-    __shared__ T ldsB[2][BN][BK + PAD];
+    __shared__ T ldsA[BUFS][BM][BK + PAD];
+    __shared__ T ldsB[BUFS][BN][BK + PAD];
 
     const T* __restrict__ A = (const T*)p.AT;
     const T* __restrict__ X = (const T*)p.X;
@@ -115,13 +115,19 @@ __launch_bounds__(256) __global__ void axis_gemm_kernel(AxisGemmParams p) {
         }
     };
 
-    stage(0, 0);
-    __syncthreads();
-
     const int ktiles = (p.K + BK - 1) / BK;
     int cur = 0;
+    if (BUFS == 2) {
+        stage(0, 0);
+        __syncthreads();
+    }
     for (int t = 0; t < ktiles; ++t) {
-        if (t + 1 < ktiles) stage(cur ^ 1, (t + 1) * BK);  // issue next tile first
+        if (BUFS == 2) {
+            if (t + 1 < ktiles) stage(cur ^ 1, (t + 1) * BK);  // prefetch next
+        } else {
+            stage(0, t * BK);
+            __syncthreads();
+        }
 #pragma unroll
         for (int kk = 0; kk < BK; kk += MT::MFMA_K) {
             typename MT::frag_t af[AM], bf[AN];
@@ -140,7 +146,7 @@ __launch_bounds__(256) __global__ void axis_gemm_kernel(AxisGemmParams p) {
                     acc[mf][nf] = MT::mfma(af[mf], bf[nf], acc[mf][nf]);
         }
         __syncthreads();
-        cur ^= 1;
+        if (BUFS == 2) cur ^= 1;
     }
 
     // ---- epilogue: bias + activation + strided store ----
@@ -176,12 +182,15 @@ extern "C" void axis_gemm_launch(AxisGemmParams p, int instances, int is_f32,
     p.tiles_l = (p.L + BN - 1) / BN;
     dim3 grid(tiles_m * p.tiles_l, instances), block(256);
     if (!is_f32) {
+        // single-buffer: half the LDS -> twice the blocks/CU; these kernels
+        // are global-latency-bound (PMC: 67% WAIT_ANY), so occupancy beats
+        // the 2-buffer pipeline here
         if (BN == 128)
-            axis_gemm_kernel<__bf16, BM, 128, BK, 2, 2><<<grid, block, 0, stream>>>(p);
+            axis_gemm_kernel<__bf16, BM, 128, BK, 2, 2, 1><<<grid, block, 0, stream>>>(p);
         else if (BN == 64)
-            axis_gemm_kernel<__bf16, BM, 64, BK, 2, 2><<<grid, block, 0, stream>>>(p);
+            axis_gemm_kernel<__bf16, BM, 64, BK, 2, 2, 1><<<grid, block, 0, stream>>>(p);
         else
-            axis_gemm_kernel<__bf16, BM, 32, BK, 4, 1><<<grid, block, 0, stream>>>(p);
+            axis_gemm_kernel<__bf16, BM, 32, BK, 4, 1, 1><<<grid, block, 0, stream>>>(p);
     } else {
         // f32: halve BK to keep the double-buffered LDS within budget
         if (BN >= 64)
