@@ -32,13 +32,14 @@ __device__ __forceinline__ long ocol_off(const AxisGemmParams& p, int q) {
 }
 
 template <typename T, int BM, int BN, int BK, int WVM, int WVN, int BUFS = 2>
-__launch_bounds__(256) __global__ void axis_gemm_kernel(AxisGemmParams p) {
+__launch_bounds__(WVM * WVN * 64) __global__ void axis_gemm_kernel(AxisGemmParams p) {
     using MT = MfmaTraits<T>;
     constexpr int CH = 16 / sizeof(T);
     constexpr int PAD = MT::LDS_PAD;
     constexpr int WM = BM / WVM, WN = BN / WVN;
     constexpr int AM = WM / 16, AN = WN / 16;
-    static_assert(WVM * WVN == 4, "4 waves per block");
+    constexpr int NT = WVM * WVN * 64;  // block threads (4 or 8 waves)
+    static_assert(WVM * WVN == 4 || WVM * WVN == 8);
 
     __shared__ T ldsA[BUFS][BM][BK + PAD];
     __shared__ T ldsB[BUFS][BN][BK + PAD];
@@ -67,7 +68,7 @@ __launch_bounds__(256) __global__ void axis_gemm_kernel(AxisGemmParams p) {
     auto stage = [&](int buf, int kt) {
         // ---- A tile: straight vectorized copy (k contiguous) ----
         constexpr int A_CPR = BK / CH;
-        for (int idx = tid; idx < BM * A_CPR; idx += 256) {
+        for (int idx = tid; idx < BM * A_CPR; idx += NT) {
             const int row = idx / A_CPR, cc = idx % A_CPR;
             const int m = m0 + row, k = kt + cc * CH;
             Chunk16 val = {};
@@ -88,12 +89,12 @@ __launch_bounds__(256) __global__ void axis_gemm_kernel(AxisGemmParams p) {
         // the loop entirely (it otherwise dominates issue time as a magic-
         // number division sequence per chunk per K-tile).
         constexpr int B_CPR = BN / CH;
-        static_assert(256 % B_CPR == 0);
+        static_assert(NT % B_CPR == 0);
         const int qc = tid % B_CPR;
         const int q0 = l0 + qc * CH;
         const bool qvec = p.x_vec && q0 + CH <= p.L;
         const long xq = qvec ? xcol_off(p, q0) : 0;
-        for (int idx = tid; idx < BK * B_CPR; idx += 256) {
+        for (int idx = tid; idx < BK * B_CPR; idx += NT) {
             const int krow = idx / B_CPR;
             const int k = kt + krow;
             alignas(16) T tmp[CH];
@@ -175,9 +176,11 @@ __launch_bounds__(256) __global__ void axis_gemm_kernel(AxisGemmParams p) {
 
 extern "C" void axis_gemm_launch(AxisGemmParams p, int instances, int is_f32,
                                  hipStream_t stream) {
-    constexpr int BM = 128, BK = 64;
+    constexpr int BK = 64;
     int BN = (p.L >= 96 && p.M >= 96) ? 128 : (p.L >= 48 ? 64 : 32);
+    if (!is_f32 && p.M >= 192 && p.L >= 192) BN = 256;
     if (is_f32 && BN > 64) BN = 64;  // f32 LDS budget caps the tile
+    const int BM = (BN == 256) ? 256 : 128;
     const int tiles_m = (p.M + BM - 1) / BM;
     p.tiles_l = (p.L + BN - 1) / BN;
     dim3 grid(tiles_m * p.tiles_l, instances), block(256);
@@ -185,17 +188,22 @@ extern "C" void axis_gemm_launch(AxisGemmParams p, int instances, int is_f32,
         // single-buffer: half the LDS -> twice the blocks/CU; these kernels
         // are global-latency-bound (PMC: 67% WAIT_ANY), so occupancy beats
         // the 2-buffer pipeline here
-        if (BN == 128)
-            axis_gemm_kernel<__bf16, BM, 128, BK, 2, 2, 1><<<grid, block, 0, stream>>>(p);
+        if (BN == 256) {
+            // 256x256 tile (8 waves): halves BOTH operands' total load bytes
+            // (the binding resource is the per-CU load path, ~10 B/cyc/CU)
+            axis_gemm_kernel<__bf16, 256, 256, BK, 4, 2, 1>
+                <<<grid, dim3(512), 0, stream>>>(p);
+        } else if (BN == 128)
+            axis_gemm_kernel<__bf16, 128, 128, BK, 2, 2, 1><<<grid, block, 0, stream>>>(p);
         else if (BN == 64)
-            axis_gemm_kernel<__bf16, BM, 64, BK, 2, 2, 1><<<grid, block, 0, stream>>>(p);
+            axis_gemm_kernel<__bf16, 128, 64, BK, 2, 2, 1><<<grid, block, 0, stream>>>(p);
         else
-            axis_gemm_kernel<__bf16, BM, 32, BK, 4, 1, 1><<<grid, block, 0, stream>>>(p);
+            axis_gemm_kernel<__bf16, 128, 32, BK, 4, 1, 1><<<grid, block, 0, stream>>>(p);
     } else {
         // f32: halve BK to keep the double-buffered LDS within budget
         if (BN >= 64)
-            axis_gemm_kernel<float, BM, 64, 32, 2, 2><<<grid, block, 0, stream>>>(p);
+            axis_gemm_kernel<float, 128, 64, 32, 2, 2><<<grid, block, 0, stream>>>(p);
         else
-            axis_gemm_kernel<float, BM, 32, 32, 4, 1><<<grid, block, 0, stream>>>(p);
+            axis_gemm_kernel<float, 128, 32, 32, 4, 1><<<grid, block, 0, stream>>>(p);
     }
 }
