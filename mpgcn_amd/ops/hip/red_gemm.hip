@@ -156,7 +156,7 @@ __launch_bounds__(256) __global__ void red_gemm_kernel(RedGemmParams p) {
 extern "C" void red_gemm_launch(RedGemmParams p, int is_f32, hipStream_t s) {
     const int rch = is_f32 ? 64 : 128;
     long chunks = (p.R + rch - 1) / rch;
-    long blocks = chunks < 608 ? chunks : 608;
+    long blocks = chunks < 1216 ? chunks : 1216;
     dim3 grid((unsigned)blocks), block(256);
 #define DISPATCH(TT)                                                        \
     do {                                                                    \
