@@ -73,6 +73,7 @@ class MPGCN(nn.Module):
         self.gcn_num_layers = gcn_num_layers
         self.compute_dtype = compute_dtype
 
+        self._streams: list = []  # side streams for branch overlap (GPU)
         self.branch_models = nn.ModuleList()
         for _ in range(M):
             branch = nn.ModuleDict()
@@ -119,8 +120,8 @@ class MPGCN(nn.Module):
             .reshape(B * N * N, T)
             .contiguous()
         )
-        branch_out = []
-        for m in range(self.M):
+
+        def run_branch(m: int) -> torch.Tensor:
             branch = self.branch_models[m]
             h_last = branch["temporal"](lstm_in)  # (B*N*N, H)
             X = h_last.reshape(B, N, N, self.lstm_hidden_dim)
@@ -130,6 +131,29 @@ class MPGCN(nn.Module):
             out = linear_act(
                 X.reshape(B * N * N, -1), fc.weight.to(X.dtype), fc.bias, relu=True
             )
-            branch_out.append(out.view(B, N, N, 1))
+            return out.view(B, N, N, 1)
+
+        if x_seq.is_cuda and self.M > 1:
+            # The M branches are independent until the final mean: run branches
+            # 1..M-1 on side HIP streams so their (latency-bound) kernel chains
+            # overlap branch 0's — autograd replays each op's backward on the
+            # stream it was recorded on, so the overlap holds in backward too.
+            main = torch.cuda.current_stream()
+            while len(self._streams) < self.M - 1:
+                self._streams.append(torch.cuda.Stream())
+            branch_out = [None] * self.M
+            for m in range(1, self.M):
+                s = self._streams[m - 1]
+                s.wait_stream(main)
+                lstm_in.record_stream(s)  # allocated on main, read on s
+                with torch.cuda.stream(s):
+                    branch_out[m] = run_branch(m)
+            branch_out[0] = run_branch(0)
+            for m in range(1, self.M):
+                main.wait_stream(self._streams[m - 1])
+                # keep the side-stream allocations alive for the main stream
+                branch_out[m].record_stream(main)
+        else:
+            branch_out = [run_branch(m) for m in range(self.M)]
         ensemble = torch.mean(torch.stack(branch_out, dim=-1), dim=-1)
         return ensemble.float().unsqueeze(1)
