@@ -74,6 +74,7 @@ class MPGCN(nn.Module):
         self.compute_dtype = compute_dtype
 
         self._streams: list = []  # side streams for branch overlap (GPU)
+        self._gop_cache: dict = {}  # static-graph operand layout cache
         self.branch_models = nn.ModuleList()
         for _ in range(M):
             branch = nn.ModuleDict()
@@ -93,14 +94,21 @@ class MPGCN(nn.Module):
     def _graph_operators(self, G_list) -> list[GraphOperator]:
         """G_list entries: a static (K, N, N) tensor (origin == destination
         graph), or a (O_dyn, D_dyn) tuple of (B, K, N, N) dynamic supports —
-        the reference's contract (MPGCN.py:89-96)."""
+        the reference's contract (MPGCN.py:89-96). Static graphs are identical
+        every step, so their cast + kernel-layout permutes are cached across
+        forwards (dynamic graphs are data-dependent and rebuilt)."""
         if len(G_list) != self.M:
             raise ValueError(f"expected {self.M} graph inputs, got {len(G_list)}")
         gops = []
         for G in G_list:
             if isinstance(G, torch.Tensor):
-                Gc = G.to(self.compute_dtype)
-                gops.append(GraphOperator(Gc, Gc))
+                key = (id(G), G.device, self.compute_dtype)
+                cached = self._gop_cache.get(key)
+                if cached is None or cached[0] is not G:
+                    Gc = G.to(self.compute_dtype)
+                    cached = (G, GraphOperator(Gc, Gc))
+                    self._gop_cache = {key: cached}  # keep one static entry
+                gops.append(cached[1])
             else:
                 Go, Gd = G
                 gops.append(GraphOperator(Go.to(self.compute_dtype),
