@@ -17,6 +17,12 @@ GT = torch.randn(S, N, N, device=dev, dtype=dt)
 R = B * N * N
 Uflat = torch.randn(R, S * C, device=dev, dtype=dt)
 Wre = torch.randn(S * C, S * H, device=dev, dtype=dt)
+xlf = torch.randn(R, 8, device=dev, dtype=dt)
+whh = torch.randn(4 * H, H, device=dev, dtype=dt) * 0.2
+whh2 = whh.t().contiguous()
+wih = torch.randn(4 * H, device=dev)
+bb = torch.randn(4 * H, device=dev)
+dh = torch.randn(R, H, device=dev, dtype=dt)
 for _ in range(10):
     if which == "mode2":
         ext.bdgcn_mode2(V, A2T, bias, True, N, S)
@@ -24,5 +30,9 @@ for _ in range(10):
         ext.bdgcn_mode1(X, GT)
     elif which == "row":
         ext.row_gemm(Uflat, Wre, None, False)
+    elif which == "lstmf":
+        ext.lstm_fused_fwd(xlf, 7, whh, wih, bb)
+    elif which == "lstmb":
+        ext.lstm_fused_bwd(xlf, 7, whh, whh2, wih, bb, dh, False)
 torch.cuda.synchronize()
 print("done", which)
