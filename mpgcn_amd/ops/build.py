@@ -17,7 +17,7 @@ from pathlib import Path
 HIP_DIR = Path(__file__).resolve().parent / "hip"
 BUILD_DIR = Path(__file__).resolve().parent / "_build"
 EXT_NAME = "_mpgcn_hip"
-SOURCES = ["ext.hip", "axis_gemm.hip", "row_gemm.hip", "lstm.hip", "lstm_fused.hip", "red_gemm.hip"]
+SOURCES = ["ext.hip", "axis_gemm.hip", "row_gemm.hip", "lstm.hip", "lstm_fused.hip", "red_gemm.hip", "elemwise.hip"]
 
 
 def so_path() -> Path:

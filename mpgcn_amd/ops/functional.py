@@ -130,8 +130,15 @@ class _BDGCNLayerFn(torch.autograd.Function):
         B, N, S, C, Hdim = ctx.dims
 
         dH = dH.contiguous()
-        dY = dH * (Y > 0).to(dH.dtype) if ctx.relu else dH
-        dbias = dY.sum(dim=(0, 1, 2)).to(torch.float32) if ctx.has_bias else None
+        if dH.dtype == torch.bfloat16 and (Hdim & (Hdim - 1)) == 0 and Hdim >= 8:
+            # one fused pass: mask + bias column-sum
+            dY, dbias = ext.relu_bwd_colsum(dH, Y, ctx.relu)
+            dY = dY.view_as(dH)
+            if not ctx.has_bias:
+                dbias = None
+        else:
+            dY = dH * (Y > 0).to(dH.dtype) if ctx.relu else dH
+            dbias = dY.sum(dim=(0, 1, 2)).to(torch.float32) if ctx.has_bias else None
 
         dV = ext.bdgcn_mode2_bwd(dY, gop.A2, S)  # (B,N,N,S,H)
         R = B * N * N

@@ -359,6 +359,28 @@ std::vector<torch::Tensor> lstm_fused_bwd(torch::Tensor x, long T_logical,
     return {ws_dw.sum(0), ws_db.sum(0), ws_dwih.sum(0), dx};
 }
 
+// Fused ReLU backward + bias-grad column sum: dY = dH * 1[Y>0], dbias=colsum.
+std::vector<torch::Tensor> relu_bwd_colsum(torch::Tensor dH, torch::Tensor Y,
+                                           bool mask) {
+    check_in(dH, "dH");
+    TORCH_CHECK(dH.scalar_type() == torch::kBFloat16, "bf16 only");
+    const long H = dH.size(-1);
+    TORCH_CHECK(H >= 8 && (H & (H - 1)) == 0 && H <= 2048,
+                "H must be a power of two in [8, 2048]");
+    auto dY = torch::empty_like(dH);
+    auto colsum = torch::zeros({H}, dH.options().dtype(torch::kFloat));
+    ReluBwdParams p{};
+    p.dH = dH.data_ptr();
+    p.Y = mask ? Y.data_ptr() : dH.data_ptr();
+    p.dY = dY.data_ptr();
+    p.colsum = colsum.data_ptr<float>();
+    p.total = dH.numel();
+    p.H = (int)H;
+    p.mask = mask ? 1 : 0;
+    relu_bwd_colsum_launch(p, stream());
+    return {dY, colsum};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("bdgcn_mode1", &bdgcn_mode1, "origin-axis graph product (K1)");
     m.def("bdgcn_mode2", &bdgcn_mode2, "dest-axis graph product + bias + act (K2)");
@@ -366,6 +388,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("bdgcn_mode1_bwd", &bdgcn_mode1_bwd, "backward dX of mode1");
     m.def("row_gemm", &row_gemm, "fused row GEMM + bias + act (K3/K6)");
     m.def("red_gemm", &red_gemm, "fused reduction GEMM X^T@Y + colsum + xdot");
+    m.def("relu_bwd_colsum", &relu_bwd_colsum, "fused ReLU bwd mask + bias colsum");
     m.def("row_gemm_out", &row_gemm_out, "row GEMM into strided output slice");
     m.def("lstm_step_fwd", &lstm_step_fwd, "fused LSTM cell forward step (K4)");
     m.def("lstm_fused_fwd", &lstm_fused_fwd, "register-resident fused LSTM forward");

@@ -91,6 +91,16 @@ struct LstmFusedParams {
     int T;
 };
 
+struct ReluBwdParams {
+    const void* dH;  // (R, H) bf16
+    const void* Y;   // (R, H) bf16 (forward output; ignored when !mask)
+    void* dY;        // (R, H) bf16 out
+    float* colsum;   // (H,) f32, zeroed by caller
+    long total;      // R * H (multiple of 8)
+    int H;           // power of two dividing 2048
+    int mask;
+};
+
 extern "C" {
 void axis_gemm_launch(AxisGemmParams p, int instances, int is_f32, hipStream_t s);
 void red_gemm_launch(RedGemmParams p, int is_f32, hipStream_t s);
@@ -100,4 +110,5 @@ void lstm_step_bwd_launch(LstmBwdParams p, int is_f32, hipStream_t s);
 void lstm_fused_fwd_launch(LstmFusedParams p, hipStream_t s);
 void lstm_fused_bwd_launch(LstmFusedParams p, hipStream_t s);
 int lstm_fused_bwd_blocks(long R);
+void relu_bwd_colsum_launch(ReluBwdParams p, hipStream_t s);
 }

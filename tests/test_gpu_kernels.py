@@ -131,6 +131,19 @@ def test_red_gemm(dtype, R, K, Nc):
     torch.testing.assert_close(xdot, ref_xdot, atol=1e-1, rtol=2e-2)
 
 
+@pytest.mark.parametrize("mask", [True, False])
+def test_relu_bwd_colsum(mask):
+    ext = _ext()
+    torch.manual_seed(13)
+    B, N, H = 3, 40, 32
+    dH = torch.randn(B, N, N, H, device=DEV, dtype=torch.bfloat16)
+    Y = torch.randn(B, N, N, H, device=DEV, dtype=torch.bfloat16)
+    dY, colsum = ext.relu_bwd_colsum(dH, Y, mask)
+    ref = dH.float() * (Y.float() > 0) if mask else dH.float()
+    torch.testing.assert_close(dY.float(), ref.view_as(dY.float()), atol=1e-3, rtol=1e-3)
+    torch.testing.assert_close(colsum, ref.reshape(-1, H).sum(0), atol=0.5, rtol=1e-2)
+
+
 def test_row_gemm_chunked_wide():
     from mpgcn_amd.ops.functional import _row_gemm_chunked
 
