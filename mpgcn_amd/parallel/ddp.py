@@ -100,6 +100,7 @@ class GradAllReducer:
     def __init__(self, model: torch.nn.Module, ctx: DistContext,
                  bucket_bytes: int = 16 << 20):
         self.ctx = ctx
+        self.model = model
         self.params = [p for p in model.parameters() if p.requires_grad]
         self.bucket_bytes = bucket_bytes
         self._pending: list[tuple[torch.distributed.Work, list[torch.Tensor], torch.Tensor]] = []
@@ -137,6 +138,13 @@ class GradAllReducer:
 
     def finalize(self):
         """Call between loss.backward() and optimizer.step()."""
+        # torch guarantees grads are usable on the caller's stream once
+        # backward() returns, but the model may have recorded branch work on
+        # side streams (MPGCN branch overlap) — wait on them explicitly so the
+        # ordering holds for any future call pattern.
+        if torch.cuda.is_available():
+            for s in getattr(self.model, "_streams", []):
+                torch.cuda.current_stream().wait_stream(s)
         if not self.ctx.enabled:
             return
         self._flush()
