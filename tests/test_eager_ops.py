@@ -73,3 +73,18 @@ def test_projection_weight_reorder_roundtrip():
             for l in range(C):
                 for h in range(H):
                     assert Wre[o * C + l, s * H + h] == W[(o * S + s) * C + l, h]
+
+
+def test_gcn1d_matches_reference_math():
+    """1-D GCN library op vs the GCN.py:23-43 equations."""
+    from mpgcn_amd.models import GCN
+
+    torch.manual_seed(5)
+    K, N, C, H, B = 3, 11, 4, 6, 2
+    m = GCN(K=K, input_dim=C, hidden_dim=H)
+    G = torch.randn(K, N, N)
+    x = torch.randn(B, N, C)
+    out = m(G, x)
+    sup = torch.cat([torch.einsum("ij,bjp->bip", G[k], x) for k in range(K)], dim=-1)
+    ref = torch.relu(sup @ m.W + m.b)
+    assert torch.allclose(out, ref, atol=1e-5)

@@ -157,12 +157,13 @@ def test_row_gemm_chunked_wide():
     torch.testing.assert_close(out.float(), ref, atol=8e-2, rtol=8e-2)
 
 
-@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
-def test_lstm_forward_vs_nn_lstm(dtype):
+@pytest.mark.parametrize("dtype,T", [(torch.float32, 7), (torch.bfloat16, 7),
+                                      (torch.bfloat16, 10)])  # T=10: slab path
+def test_lstm_forward_vs_nn_lstm(dtype, T):
     from mpgcn_amd.ops.functional import fused_lstm_last
 
     torch.manual_seed(6)
-    R, T, H = 333, 7, 32
+    R, H = 333, 32
     x = torch.randn(R, T, device=DEV)
     lstm = torch.nn.LSTM(1, H, num_layers=1, batch_first=True).to(DEV)
     ref, _ = lstm(x.unsqueeze(-1), (torch.zeros(1, R, H, device=DEV),
