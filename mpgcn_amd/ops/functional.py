@@ -305,7 +305,12 @@ class _LinearActFn(torch.autograd.Function):
     def backward(ctx, dOut):
         X2d, weight, out = ctx.saved_tensors
         dY = (dOut * (out > 0).to(dOut.dtype) if ctx.relu else dOut).contiguous()
-        dX = dY @ weight.to(dY.dtype)
+        if weight.shape[0] == 1:
+            # rank-1 head (the MPGCN FC): the outer product is a broadcast
+            # multiply — rocBLAS spends ~250us + a workspace memset on it
+            dX = dY * weight.to(dY.dtype).reshape(1, -1)
+        else:
+            dX = dY @ weight.to(dY.dtype)
         # dW = dY^T @ X via the fused reduction kernel (rocBLAS is ~25x off
         # roofline on this 1-column tall reduction); colsum(dY) gives dbias
         ext = _ops.get_ext()
