@@ -39,7 +39,7 @@ __launch_bounds__(WVM * WVN * 64) __global__ void axis_gemm_kernel(AxisGemmParam
     constexpr int WM = BM / WVM, WN = BN / WVN;
     constexpr int AM = WM / 16, AN = WN / 16;
     constexpr int NT = WVM * WVN * 64;  // block threads (4 or 8 waves)
-    static_assert(WVM * WVN == 4 || WVM * WVN == 8 || WVM * WVN == 16);
+    static_assert(WVM * WVN == 4 || WVM * WVN == 8);
 
     __shared__ T ldsA[BUFS][BM][BK + PAD];
     __shared__ T ldsB[BUFS][BN][BK + PAD];
@@ -179,7 +179,6 @@ extern "C" void axis_gemm_launch(AxisGemmParams p, int instances, int is_f32,
     constexpr int BK = 64;
     int BN = (p.L >= 96 && p.M >= 96) ? 128 : (p.L >= 48 ? 64 : 32);
     if (!is_f32 && p.M >= 192 && p.L >= 192) BN = 256;
-    if (!is_f32 && p.M >= 192 && p.L >= 2048) BN = 512;
     if (is_f32 && BN > 64) BN = 64;  // f32 LDS budget caps the tile
     const int BM = (BN >= 256) ? 256 : 128;
     const int tiles_m = (p.M + BM - 1) / BM;
@@ -189,12 +188,7 @@ extern "C" void axis_gemm_launch(AxisGemmParams p, int instances, int is_f32,
         // single-buffer: half the LDS -> twice the blocks/CU; these kernels
         // are global-latency-bound (PMC: 67% WAIT_ANY), so occupancy beats
         // the 2-buffer pipeline here
-        if (BN == 512) {
-            // 256x512 tile (16 waves, 1 block/CU): minimizes total load-path
-            // bytes for the large-L contractions
-            axis_gemm_kernel<__bf16, 256, 512, BK, 2, 8, 1>
-                <<<grid, dim3(1024), 0, stream>>>(p);
-        } else if (BN == 256) {
+        if (BN == 256) {
             // 256x256 tile (8 waves): halves BOTH operands' total load bytes
             // (the binding resource is the per-CU load path, ~10 B/cyc/CU)
             axis_gemm_kernel<__bf16, 256, 256, BK, 4, 2, 1>
