@@ -37,6 +37,9 @@ def main():
     ap.add_argument("--impl", type=str, default="native", choices=["native", "eager"],
                     help="'eager' runs the reference-math fp32 transcription "
                          "(stock torch ops, K^2-pair formulation) as the baseline")
+    ap.add_argument("--partition", type=str, default="dp", choices=["dp", "region"],
+                    help="multi-rank strategy: data parallel (weak scaling) or "
+                         "region partition (activation grid sharded across ranks)")
     args = ap.parse_args()
 
     import torch.distributed as dist
@@ -89,6 +92,10 @@ def main():
     O_dyn_raw = torch.rand(7, N, N, device=device)
     D_dyn_raw = torch.rand(7, N, N, device=device)
 
+    region = args.partition == "region" and world > 1
+    if region:
+        from mpgcn_amd.parallel.region import mpgcn_forward_sharded, shard_dest
+
     def step(i: int):
         g = (torch.arange(B, device=device) * 7 + i) % (T_pool - T - 1)
         x = pool[g.unsqueeze(1) + torch.arange(T, device=device)]  # (B,T,N,N,1)
@@ -96,7 +103,12 @@ def main():
         key = (g + T) % 7
         G_o = build_supports(O_dyn_raw[key], kernel, K_order)
         G_d = build_supports(D_dyn_raw[key], kernel, K_order)
-        y_pred = model(x, [G_static, (G_o, G_d)])
+        if region:
+            xs = shard_dest(x, rank, world)
+            y = shard_dest(y, rank, world)
+            y_pred = mpgcn_forward_sharded(model, xs, [G_static, (G_o, G_d)])
+        else:
+            y_pred = model(x, [G_static, (G_o, G_d)])
         loss = criterion(y_pred, y)
         opt.zero_grad(set_to_none=True)
         loss.backward()
@@ -152,7 +164,7 @@ def main():
                 "supports_K": S,
                 "gcn_layers": 3,
                 "branches": 2,
-                "parallelism": f"dp{max(world, 1)}",
+                "parallelism": f"{args.partition}{max(world, 1)}",
                 "final_loss": round(float(loss.item()), 5),
             },
         }))

@@ -7,9 +7,9 @@ The BDGCN layer uses the factored algorithm (mpgcn_amd/ops/eager.py docstring):
 
 Graph supports carry no gradient (they are built from input data each step,
 reference Model_Trainer.py:82-84,106), so backward only produces dX, dW, dbias:
-    dY  = dH * 1[H > 0]                       (ReLU mask, elementwise)
+    dY, dbias = relu_bwd_colsum(dH, Y)        [fused mask + bias colsum kernel]
     dV  = mode2_bwd(dY, A2)                   [axis_gemm]
-    dW  = Uflat^T @ dVflat  (reordered)       [rocBLAS reduction GEMM]
+    dW  = (dV^T @ U reordered)                [red_gemm fused reduction kernel]
     dU  = dVflat @ Wre^T                      [row_gemm]
     dX  = mode1_bwd(dU, A3T)                  [axis_gemm]
 """
