@@ -20,7 +20,9 @@ template <typename T, int AKF, int ANF>
 __launch_bounds__(256) __global__ void red_gemm_kernel(RedGemmParams p) {
     using MT = MfmaTraits<T>;
     constexpr int CH = 16 / sizeof(T);
-    constexpr int RCH = sizeof(T) == 2 ? 128 : 64;  // rows per chunk (LDS budget)
+    // rows per chunk: smaller -> less LDS -> more resident blocks (the
+    // kernel is latency-bound; occupancy is the lever)
+    constexpr int RCH = 64;
     constexpr int PAD = MT::LDS_PAD;
     constexpr int KMAX = AKF * 32, NMAX = ANF * 32;
 
@@ -154,9 +156,8 @@ __launch_bounds__(256) __global__ void red_gemm_kernel(RedGemmParams p) {
 }
 
 extern "C" void red_gemm_launch(RedGemmParams p, int is_f32, hipStream_t s) {
-    const int rch = is_f32 ? 64 : 128;
-    long chunks = (p.R + rch - 1) / rch;
-    long blocks = chunks < 1216 ? chunks : 1216;
+    long chunks = (p.R + 63) / 64;
+    long blocks = chunks < 2432 ? chunks : 2432;
     dim3 grid((unsigned)blocks), block(256);
 #define DISPATCH(TT)                                                        \
     do {                                                                    \
