@@ -47,37 +47,23 @@ __launch_bounds__(256) __global__ void row_gemm_kernel(RowGemmParams p, int nf16
         const bool row_ok = row < p.R;
         f32x4 acc[8] = {};  // up to 8 column fragments (N <= 128)
 
-        if (p.x_vec && row_ok && kfrags <= 6 && kfrags * MT::MFMA_K <= p.K) {
-            // fast path: issue ALL of this row's K-fragments up front so the
-            // loads overlap (one global latency instead of kfrags serial ones)
-            typename MT::frag_t afs[6];
-#pragma unroll
-            for (int kf = 0; kf < 6; ++kf)
-                if (kf < kfrags)
-                    afs[kf] = *(const typename MT::frag_t*)
-                        &X[row * p.K + kf * MT::MFMA_K + kgrp * MT::FRAG_ELEMS];
-#pragma unroll
-            for (int kf = 0; kf < 6; ++kf)
-                for (int nf = 0; nf < nf16 && kf < kfrags; ++nf) {
-                    const typename MT::frag_t bf = *(const typename MT::frag_t*)
-                        &ldsWT[(nf * 16 + lrow) * KP + kf * MT::MFMA_K +
-                               kgrp * MT::FRAG_ELEMS];
-                    acc[nf] = MT::mfma(afs[kf], bf, acc[nf]);
-                }
-        } else {
-            for (int kf = 0; kf < kfrags; ++kf) {
-                const int k = kf * MT::MFMA_K + kgrp * MT::FRAG_ELEMS;
+        for (int kf = 0; kf < kfrags; ++kf) {
+            const int k = kf * MT::MFMA_K + kgrp * MT::FRAG_ELEMS;
+            typename MT::frag_t af;
+            if (row_ok && k + MT::FRAG_ELEMS <= p.K && p.x_vec) {
+                af = *(const typename MT::frag_t*)&X[row * p.K + k];
+            } else {
                 alignas(16) T tmp[MT::FRAG_ELEMS];
 #pragma unroll
                 for (int i = 0; i < MT::FRAG_ELEMS; ++i)
                     tmp[i] = (row_ok && k + i < p.K) ? X[row * p.K + k + i] : (T)0.f;
-                const typename MT::frag_t af = *(const typename MT::frag_t*)tmp;
-                for (int nf = 0; nf < nf16; ++nf) {
-                    const typename MT::frag_t bf = *(const typename MT::frag_t*)
-                        &ldsWT[(nf * 16 + lrow) * KP + kf * MT::MFMA_K +
-                               kgrp * MT::FRAG_ELEMS];
-                    acc[nf] = MT::mfma(af, bf, acc[nf]);
-                }
+                af = *(const typename MT::frag_t*)tmp;
+            }
+            for (int nf = 0; nf < nf16; ++nf) {
+                const typename MT::frag_t bf = *(const typename MT::frag_t*)
+                    &ldsWT[(nf * 16 + lrow) * KP + kf * MT::MFMA_K +
+                           kgrp * MT::FRAG_ELEMS];
+                acc[nf] = MT::mfma(af, bf, acc[nf]);
             }
         }
 
