@@ -326,3 +326,35 @@ def test_train_step_decreases_loss_on_gpu():
         opt.step()
         losses.append(loss.item())
     assert losses[-1] < losses[0] * 0.7, losses[:3] + losses[-3:]
+
+
+def test_full_model_dual_rwd_s5_gpu():
+    """S=5 (dual random-walk) exercises the chunked row_gemm path (S*H=160)
+    inside a real train step on GPU."""
+    from mpgcn_amd.graph import build_supports
+    from mpgcn_amd.models import MPGCN
+
+    torch.manual_seed(14)
+    B, N, H, T, order = 2, 32, 32, 7, 2
+    S = 2 * order + 1
+    model = MPGCN(M=2, K=S, input_dim=1, lstm_hidden_dim=H, lstm_num_layers=1,
+                  gcn_hidden_dim=H, gcn_num_layers=2, num_nodes=N,
+                  compute_dtype=torch.bfloat16).to(DEV)
+    model_cpu = MPGCN(M=2, K=S, input_dim=1, lstm_hidden_dim=H, lstm_num_layers=1,
+                      gcn_hidden_dim=H, gcn_num_layers=2, num_nodes=N)
+    model_cpu.load_state_dict(model.state_dict())
+
+    x = torch.rand(B, T, N, N, 1)
+    flow = torch.rand(B, N, N) + 0.05
+    Gs = build_supports(torch.rand(1, N, N) + 0.05, "dual_random_walk_diffusion", order)[0]
+    Go = build_supports(flow, "dual_random_walk_diffusion", order)
+    Gd = build_supports(flow.transpose(-2, -1), "dual_random_walk_diffusion", order)
+
+    ref = model_cpu(x, [Gs, (Go, Gd)])
+    out = model(x.to(DEV), [Gs.to(DEV), (Go.to(DEV), Gd.to(DEV))])
+    torch.testing.assert_close(out.cpu(), ref, atol=6e-2, rtol=6e-2)
+
+    loss = out.square().sum()
+    loss.backward()
+    for n, p in model.named_parameters():
+        assert p.grad is not None and torch.isfinite(p.grad).all(), n
