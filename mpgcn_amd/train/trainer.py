@@ -240,6 +240,9 @@ class ModelTrainer:
 
     # -- test loop: autoregressive rollout (Model_Trainer.py:145-185) --
     def test(self, data_loader: dict, modes: list):
+        """Evaluation runs the full (unsharded) model on every rank — the
+        region-partition mode only shards training steps; eval at the tested
+        scales fits a single GPU and rank 0 writes the scores file."""
         ckpt = torch.load(self._ckpt_path(), map_location=self.device, weights_only=False)
         self.model.load_state_dict(ckpt["state_dict"])
         self.model.eval()
