@@ -104,9 +104,8 @@ __launch_bounds__(WVM * WVN * 64) __global__ void axis_gemm_kernel(AxisGemmParam
                 if (qvec) {
                     // X is streamed once per block set: nontemporal keeps L2
                     // for the re-read graph operand
-                    typedef __attribute__((ext_vector_type(4))) int i32x4;
-                    *(i32x4*)tmp = __builtin_nontemporal_load(
-                        (const i32x4*)(X + rbase + xq));
+                    *(Chunk16*)tmp = __builtin_nontemporal_load(
+                        (const Chunk16*)(X + rbase + xq));
                 } else {
                     for (int i = 0; i < CH; ++i)
                         if (q0 + i < p.L) tmp[i] = X[rbase + xcol_off(p, q0 + i)];
