@@ -102,10 +102,7 @@ __launch_bounds__(WVM * WVN * 64) __global__ void axis_gemm_kernel(AxisGemmParam
             if (k < p.K) {
                 const long rbase = x_base + xrow_off(p, k);
                 if (qvec) {
-                    // X is streamed once per block set: nontemporal keeps L2
-                    // for the re-read graph operand
-                    *(Chunk16*)tmp = __builtin_nontemporal_load(
-                        (const Chunk16*)(X + rbase + xq));
+                    *(Chunk16*)tmp = *(const Chunk16*)(X + rbase + xq);
                 } else {
                     for (int i = 0; i < CH; ++i)
                         if (q0 + i < p.L) tmp[i] = X[rbase + xcol_off(p, q0 + i)];
