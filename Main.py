@@ -64,6 +64,13 @@ def build_parser() -> argparse.ArgumentParser:
                         choices=["dp", "region"], default="dp",
                         help="multi-GPU strategy: data parallel (default) or "
                              "region partition (shard the N x N activation grid)")
+    parser.add_argument("-M", "--perspectives", type=int, choices=[2, 3], default=2,
+                        help="graph perspectives: 2 (reference: adjacency + "
+                             "dynamic OD-correlation) or 3 (+ static OD-correlation)")
+    parser.add_argument("-fusion", "--fusion", type=str,
+                        choices=["mean", "attention"], default="mean",
+                        help="branch fusion: arithmetic mean (reference) or "
+                             "learned attention weights")
     parser.add_argument("-resume", "--resume", action="store_true",
                         help="resume training from the extended checkpoint "
                              "({model}_od.resume.pkl) if present")

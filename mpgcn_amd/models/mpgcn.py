@@ -62,11 +62,14 @@ class MPGCN(nn.Module):
     def __init__(self, M: int, K: int, input_dim: int, lstm_hidden_dim: int,
                  lstm_num_layers: int, gcn_hidden_dim: int, gcn_num_layers: int,
                  num_nodes: int, user_bias: bool = True, activation: str = "relu",
-                 compute_dtype: torch.dtype = torch.float32):
+                 compute_dtype: torch.dtype = torch.float32, fusion: str = "mean"):
         super().__init__()
         if lstm_num_layers != 1:
             raise ValueError("MPGCN uses a 1-layer LSTM (Model_Trainer.py:50)")
+        if fusion not in ("mean", "attention"):
+            raise ValueError("fusion must be 'mean' or 'attention'")
         self.M = M
+        self.fusion = fusion
         self.K = K
         self.num_nodes = num_nodes
         self.lstm_hidden_dim = lstm_hidden_dim
@@ -90,6 +93,12 @@ class MPGCN(nn.Module):
                 nn.Linear(gcn_hidden_dim, input_dim, bias=True), nn.ReLU()
             )
             self.branch_models.append(branch)
+        if fusion == "attention":
+            # learned softmax weights over perspectives (the multi-graph
+            # attention fusion of the MPGCN paper; the reference replication
+            # hardcodes the arithmetic mean, MPGCN.py:110 — 'mean' keeps its
+            # state_dict exactly, 'attention' adds only this parameter)
+            self.fusion_w = nn.Parameter(torch.zeros(M))
 
     def _graph_operators(self, G_list) -> list[GraphOperator]:
         """G_list entries: a static (K, N, N) tensor (origin == destination
@@ -163,5 +172,10 @@ class MPGCN(nn.Module):
                 branch_out[m].record_stream(main)
         else:
             branch_out = [run_branch(m) for m in range(self.M)]
-        ensemble = torch.mean(torch.stack(branch_out, dim=-1), dim=-1)
+        stacked = torch.stack(branch_out, dim=-1)
+        if self.fusion == "attention":
+            w = torch.softmax(self.fusion_w, dim=0).to(stacked.dtype)
+            ensemble = (stacked * w).sum(dim=-1)
+        else:
+            ensemble = torch.mean(stacked, dim=-1)
         return ensemble.float().unsqueeze(1)

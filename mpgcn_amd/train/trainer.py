@@ -45,6 +45,7 @@ class ModelTrainer:
         )
 
         self.partition = params.get("partition", "dp")
+        self._data_O_dyn = data["O_dyn_G"]
         self.K = get_support_K(params["kernel_type"], params["cheby_order"])
         self.G = self.preprocess_adj(data["adj"])
         self.model = self.get_model().to(self.device)
@@ -72,7 +73,7 @@ class ModelTrainer:
         if self.params["model"] != "MPGCN":
             raise NotImplementedError("Invalid model name.")
         return MPGCN(
-            M=2,
+            M=int(self.params.get("perspectives", 2)),
             K=self.K,
             input_dim=1,
             lstm_hidden_dim=self.params["hidden_dim"],
@@ -83,6 +84,7 @@ class ModelTrainer:
             user_bias=True,
             activation="relu",
             compute_dtype=self.compute_dtype,
+            fusion=self.params.get("fusion", "mean"),
         )
 
     def get_loss(self):
@@ -110,6 +112,21 @@ class ModelTrainer:
     def _scores_path(self) -> str:
         return self.params["output_dir"] + f"/{self.params['model']}_prediction_scores.txt"
 
+    def _graph_list(self, dyn):
+        """[static adjacency supports, dynamic OD-correlation tuple] — plus,
+        at perspectives=3, a static OD-correlation perspective built from the
+        day-of-week-averaged correlation graphs (BASELINE config #2)."""
+        G_list = [self.G, dyn]
+        if int(self.params.get("perspectives", 2)) == 3:
+            if not hasattr(self, "_G_corr"):
+                corr = self._data_O_dyn.mean(dim=-1).to(self.device)
+                self._G_corr = build_supports(
+                    corr.unsqueeze(0), self.params["kernel_type"],
+                    self.params["cheby_order"],
+                ).squeeze(0)
+            G_list.append(self._G_corr)
+        return G_list
+
     def _forward(self, x_seq, O_dyn_G, D_dyn_G):
         with trace_range("dyn_supports"):
             dyn = (
@@ -117,7 +134,7 @@ class ModelTrainer:
                 self.preprocess_dynamic_graph(D_dyn_G),
             )
         with trace_range("forward"):
-            return self.model(x_seq=x_seq, G_list=[self.G, dyn])
+            return self.model(x_seq=x_seq, G_list=self._graph_list(dyn))
 
     def _step_forward(self, x_seq, y_true, O_dyn_G, D_dyn_G):
         """Forward + matching target; region partition shards the destination
@@ -134,7 +151,7 @@ class ModelTrainer:
             xs = shard_dest(x_seq, self.ctx.rank, self.ctx.world_size)
             ys = shard_dest(y_true, self.ctx.rank, self.ctx.world_size)
             with trace_range("forward"):
-                return mpgcn_forward_sharded(self.model, xs, [self.G, dyn]), ys
+                return mpgcn_forward_sharded(self.model, xs, self._graph_list(dyn)), ys
         return self._forward(x_seq, O_dyn_G, D_dyn_G), y_true
 
     # -- training loop (Model_Trainer.py:87-142 equivalent) --
@@ -263,7 +280,7 @@ class ModelTrainer:
                     # horizons — kept for score compatibility with the reference
                     # (Model_Trainer.py:156-163); see docs for the quirk note.
                     for _ in range(self.params["pred_len"]):
-                        step_y = self.model(x_seq=cur_x_seq, G_list=[self.G, dyn])
+                        step_y = self.model(x_seq=cur_x_seq, G_list=self._graph_list(dyn))
                         cur_x_seq = torch.cat([cur_x_seq[:, 1:], step_y], dim=1)
                         y_pred.append(step_y)
                 forecast.append(torch.cat(y_pred, dim=1).cpu().numpy())

@@ -87,3 +87,29 @@ def test_backward_produces_grads():
     for name, p in model.named_parameters():
         assert p.grad is not None, name
         assert torch.isfinite(p.grad).all(), name
+
+
+def test_three_perspectives_attention_fusion():
+    """M=3 + learned attention fusion (BASELINE config #2 / north-star
+    'multi-graph attention fusion'); the default M=2 mean fusion keeps the
+    reference state_dict exactly (tested above)."""
+    torch.manual_seed(3)
+    N, K, H = 8, 3, 16
+    model = MPGCN(M=3, K=K, input_dim=1, lstm_hidden_dim=H, lstm_num_layers=1,
+                  gcn_hidden_dim=H, gcn_num_layers=2, num_nodes=N,
+                  fusion="attention")
+    assert "fusion_w" in model.state_dict()
+    x = torch.rand(2, 5, N, N, 1)
+    flow = torch.rand(2, N, N)
+    Gs = build_supports(torch.rand(1, N, N), "random_walk_diffusion", K - 1)[0]
+    Gc = build_supports(torch.rand(1, N, N), "random_walk_diffusion", K - 1)[0]
+    Go = build_supports(flow, "random_walk_diffusion", K - 1)
+    out = model(x, [Gs, (Go, Go), Gc])
+    assert out.shape == (2, 1, N, N, 1)
+    out.sum().backward()
+    assert model.fusion_w.grad is not None
+
+    # mean fusion (default) must not add parameters
+    m2 = MPGCN(M=2, K=K, input_dim=1, lstm_hidden_dim=H, lstm_num_layers=1,
+               gcn_hidden_dim=H, gcn_num_layers=3, num_nodes=N)
+    assert "fusion_w" not in m2.state_dict()
