@@ -37,6 +37,9 @@ def main():
     ap.add_argument("--impl", type=str, default="native", choices=["native", "eager"],
                     help="'eager' runs the reference-math fp32 transcription "
                          "(stock torch ops, K^2-pair formulation) as the baseline")
+    ap.add_argument("--branches", type=int, default=2, choices=[2, 3],
+                    help="graph perspectives (3 = BASELINE config #2)")
+    ap.add_argument("--fusion", type=str, default="mean", choices=["mean", "attention"])
     ap.add_argument("--partition", type=str, default="dp", choices=["dp", "region"],
                     help="multi-rank strategy: data parallel (weak scaling) or "
                          "region partition (activation grid sharded across ranks)")
@@ -76,9 +79,10 @@ def main():
         model = MPGCNReference(M=2, K=S, input_dim=1, hidden=H, gcn_layers=3,
                                num_nodes=N).to(device)
     else:
-        model = MPGCN(M=2, K=S, input_dim=1, lstm_hidden_dim=H, lstm_num_layers=1,
-                      gcn_hidden_dim=H, gcn_num_layers=3, num_nodes=N,
-                      compute_dtype=cdtype).to(device)
+        model = MPGCN(M=args.branches, K=S, input_dim=1, lstm_hidden_dim=H,
+                      lstm_num_layers=1, gcn_hidden_dim=H, gcn_num_layers=3,
+                      num_nodes=N, compute_dtype=cdtype,
+                      fusion=args.fusion).to(device)
     opt = torch.optim.Adam(model.parameters(), lr=1e-4)
     reducer = GradAllReducer(model, ctx)
     criterion = torch.nn.MSELoss()
@@ -88,6 +92,8 @@ def main():
     pool = torch.log1p(20.0 * torch.rand(T_pool, N, N, 1, device=device))
     adj = (torch.rand(N, N, device=device) < 0.1).float()
     G_static = build_supports(adj.unsqueeze(0), kernel, K_order).squeeze(0)
+    G_corr = build_supports(torch.rand(1, N, N, device=device), kernel,
+                            K_order).squeeze(0)  # third perspective (static)
     # raw day-of-week correlation graphs (support build runs per-step, timed)
     O_dyn_raw = torch.rand(7, N, N, device=device)
     D_dyn_raw = torch.rand(7, N, N, device=device)
@@ -103,12 +109,15 @@ def main():
         key = (g + T) % 7
         G_o = build_supports(O_dyn_raw[key], kernel, K_order)
         G_d = build_supports(D_dyn_raw[key], kernel, K_order)
+        G_list = [G_static, (G_o, G_d)]
+        if args.branches == 3 and args.impl == "native":
+            G_list.append(G_corr)
         if region:
             xs = shard_dest(x, rank, world)
             y = shard_dest(y, rank, world)
-            y_pred = mpgcn_forward_sharded(model, xs, [G_static, (G_o, G_d)])
+            y_pred = mpgcn_forward_sharded(model, xs, G_list)
         else:
-            y_pred = model(x, [G_static, (G_o, G_d)])
+            y_pred = model(x, G_list)
         loss = criterion(y_pred, y)
         opt.zero_grad(set_to_none=True)
         loss.backward()
@@ -163,7 +172,7 @@ def main():
                 "kernel": kernel,
                 "supports_K": S,
                 "gcn_layers": 3,
-                "branches": 2,
+                "branches": args.branches if args.impl == "native" else 2,
                 "parallelism": f"{args.partition}{max(world, 1)}",
                 "final_loss": round(float(loss.item()), 5),
             },
