@@ -15,7 +15,8 @@
 
 __launch_bounds__(256) __global__ void relu_bwd_colsum_kernel(ReluBwdParams p) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
-    float* lcol = (float*)smem;  // [H]
+    float* lcol = (float*)smem;           // [H]
+    float* lpart = (float*)smem + p.H;    // [256][8] (deterministic path)
     const int tid = threadIdx.x;
     for (int i = tid; i < p.H; i += 256) lcol[i] = 0.f;
     __syncthreads();
@@ -45,15 +46,37 @@ __launch_bounds__(256) __global__ void relu_bwd_colsum_kernel(ReluBwdParams p) {
         *(Chunk16*)&dY[e0] = ov;
     }
     // thread's column for slot j is fixed: (start*8 + j) % H
+    if (p.det) {
+        // deterministic block reduce: stage per-thread partials, then one
+        // thread per column sums contributors in fixed thread order
 #pragma unroll
-    for (int j = 0; j < 8; ++j)
-        atomicAdd(&lcol[(int)((start * 8 + j) % p.H)], part[j]);
-    __syncthreads();
-    for (int i = tid; i < p.H; i += 256) atomicAdd(&p.colsum[i], lcol[i]);
+        for (int j = 0; j < 8; ++j) lpart[tid * 8 + j] = part[j];
+        __syncthreads();
+        for (int i = tid; i < p.H; i += 256) {
+            float s = 0.f;
+            for (int t = 0; t < 256; ++t) {
+                const long base = ((long)blockIdx.x * 256 + t) * 8;
+                const int j = (int)(((long)i - base) % p.H + p.H) % p.H;
+                if (j < 8) s += lpart[t * 8 + j];
+            }
+            p.colsum[(long)blockIdx.x * p.H + i] = s;
+        }
+    } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+            atomicAdd(&lcol[(int)((start * 8 + j) % p.H)], part[j]);
+        __syncthreads();
+        for (int i = tid; i < p.H; i += 256) atomicAdd(&p.colsum[i], lcol[i]);
+    }
+}
+
+extern "C" long relu_bwd_nblocks(long total) {
+    long blocks = (total / 8 + 255) / 256;
+    return blocks > 2048 ? 2048 : blocks;
 }
 
 extern "C" void relu_bwd_colsum_launch(ReluBwdParams p, hipStream_t s) {
-    long blocks = (p.total / 8 + 255) / 256;
-    if (blocks > 2048) blocks = 2048;
-    relu_bwd_colsum_kernel<<<dim3((unsigned)blocks), dim3(256), p.H * 4, s>>>(p);
+    const long blocks = relu_bwd_nblocks(p.total);
+    const size_t smem = p.H * 4 + (p.det ? 256 * 8 * 4 : 0);
+    relu_bwd_colsum_kernel<<<dim3((unsigned)blocks), dim3(256), smem, s>>>(p);
 }

@@ -227,10 +227,17 @@ std::vector<torch::Tensor> red_gemm(torch::Tensor X, torch::Tensor Y,
     const long R = X.size(0), K = X.size(1), N = Y.size(1);
     TORCH_CHECK(Y.size(0) == R, "row mismatch");
     auto f32 = X.options().dtype(torch::kFloat);
-    auto out = torch::zeros({K, N}, f32);
-    auto colsum = want_colsum ? torch::zeros({K}, f32) : torch::Tensor();
     const bool has_xv = xvec.has_value() && xvec->defined();
-    auto xdot = has_xv ? torch::zeros({K}, f32) : torch::Tensor();
+    const bool det = at::globalContext().deterministicAlgorithms();
+    const long nb = det ? red_gemm_nblocks(R) : 1;
+    // deterministic mode: per-block workspace rows, order-independent sum
+    auto out = det ? torch::empty({nb, K, N}, f32) : torch::zeros({K, N}, f32);
+    // workspaces are zero-initialized: the kernel's NSPLIT row-partition
+    // factor is template-dependent, so some rows may stay unwritten
+    auto colsum = !want_colsum ? torch::Tensor()
+                  : det ? torch::zeros({2 * nb, K}, f32) : torch::zeros({K}, f32);
+    auto xdot = !has_xv ? torch::Tensor()
+                : det ? torch::zeros({2 * nb, K}, f32) : torch::zeros({K}, f32);
     RedGemmParams p{};
     p.X = X.data_ptr();
     p.Y = Y.data_ptr();
@@ -240,10 +247,15 @@ std::vector<torch::Tensor> red_gemm(torch::Tensor X, torch::Tensor Y,
     p.colsum = want_colsum ? colsum.data_ptr<float>() : nullptr;
     p.xdot = has_xv ? xdot.data_ptr<float>() : nullptr;
     p.R = R; p.K = (int)K; p.N = (int)N;
+    p.det = det ? 1 : 0;
     const int ch = chunk_elems(X);
     p.x_vec = (K % ch == 0);
     p.y_vec = (N % ch == 0);
     red_gemm_launch(p, is_f32(X), stream());
+    if (det) {
+        return {out.sum(0), want_colsum ? colsum.sum(0) : colsum,
+                has_xv ? xdot.sum(0) : xdot};
+    }
     return {out, colsum, xdot};
 }
 
@@ -368,7 +380,10 @@ std::vector<torch::Tensor> relu_bwd_colsum(torch::Tensor dH, torch::Tensor Y,
     TORCH_CHECK(H >= 8 && (H & (H - 1)) == 0 && H <= 2048,
                 "H must be a power of two in [8, 2048]");
     auto dY = torch::empty_like(dH);
-    auto colsum = torch::zeros({H}, dH.options().dtype(torch::kFloat));
+    const bool det = at::globalContext().deterministicAlgorithms();
+    const long nb = det ? relu_bwd_nblocks(dH.numel()) : 1;
+    auto colsum = det ? torch::empty({nb, H}, dH.options().dtype(torch::kFloat))
+                      : torch::zeros({H}, dH.options().dtype(torch::kFloat));
     ReluBwdParams p{};
     p.dH = dH.data_ptr();
     p.Y = mask ? Y.data_ptr() : dH.data_ptr();
@@ -377,8 +392,9 @@ std::vector<torch::Tensor> relu_bwd_colsum(torch::Tensor dH, torch::Tensor Y,
     p.total = dH.numel();
     p.H = (int)H;
     p.mask = mask ? 1 : 0;
+    p.det = det ? 1 : 0;
     relu_bwd_colsum_launch(p, stream());
-    return {dY, colsum};
+    return {dY, det ? colsum.sum(0) : colsum};
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {

@@ -73,6 +73,8 @@ struct RedGemmParams {
     long R;
     int K, N;
     int x_vec, y_vec;
+    int det;            // deterministic: out/colsum/xdot are per-block
+                        // workspaces (nblocks, ...) written with plain stores
 };
 
 struct LstmFusedParams {
@@ -95,10 +97,11 @@ struct ReluBwdParams {
     const void* dH;  // (R, H) bf16
     const void* Y;   // (R, H) bf16 (forward output; ignored when !mask)
     void* dY;        // (R, H) bf16 out
-    float* colsum;   // (H,) f32, zeroed by caller
+    float* colsum;   // (H,) f32 zeroed, or (nblocks, H) workspace when det
     long total;      // R * H (multiple of 8)
     int H;           // power of two dividing 2048
     int mask;
+    int det;
 };
 
 extern "C" {
@@ -111,4 +114,6 @@ void lstm_fused_fwd_launch(LstmFusedParams p, hipStream_t s);
 void lstm_fused_bwd_launch(LstmFusedParams p, hipStream_t s);
 int lstm_fused_bwd_blocks(long R);
 void relu_bwd_colsum_launch(ReluBwdParams p, hipStream_t s);
+long relu_bwd_nblocks(long total);
+long red_gemm_nblocks(long R);
 }

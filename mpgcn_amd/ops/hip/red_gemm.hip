@@ -131,7 +131,9 @@ __launch_bounds__(256) __global__ void red_gemm_kernel(RedGemmParams p) {
         __syncthreads();
     }
 
-    // one atomic accumulation per output element at the end
+    // epilogue: atomic accumulation (fast path) or per-block workspace rows
+    // (deterministic path — summed order-independently by the caller)
+    float* out_row = p.det ? p.out + (long)blockIdx.x * p.K * p.N : p.out;
 #pragma unroll
     for (int kf = 0; kf < AKF; ++kf) {
 #pragma unroll
@@ -140,24 +142,39 @@ __launch_bounds__(256) __global__ void red_gemm_kernel(RedGemmParams p) {
             for (int r = 0; r < 4; ++r) {
                 const int k = wk * AKF * 16 + kf * 16 + kgrp * 4 + r;
                 const int n = wn * ANF * 16 + nf * 16 + lrow;
-                if (k < p.K && n < p.N)
-                    atomicAdd(&p.out[(long)k * p.N + n], acc[kf][nf][r]);
+                if (k < p.K && n < p.N) {
+                    if (p.det) out_row[(long)k * p.N + n] = acc[kf][nf][r];
+                    else atomicAdd(&out_row[(long)k * p.N + n], acc[kf][nf][r]);
+                }
             }
         }
     }
     {
         constexpr int NSPLIT = (256 / KMAX >= 2) ? 2 : 1;
         const int k = tid % KMAX;
-        if (k < p.K && tid / KMAX < NSPLIT) {
-            if (p.colsum) atomicAdd(&p.colsum[k], cs);
-            if (XV) atomicAdd(&p.xdot[k], xd);
+        const int part = tid / KMAX;
+        if (k < p.K && part < NSPLIT) {
+            if (p.det) {
+                // two row-range partials per k land in separate workspace halves
+                if (p.colsum)
+                    p.colsum[((long)blockIdx.x * NSPLIT + part) * p.K + k] = cs;
+                if (XV)
+                    p.xdot[((long)blockIdx.x * NSPLIT + part) * p.K + k] = xd;
+            } else {
+                if (p.colsum) atomicAdd(&p.colsum[k], cs);
+                if (XV) atomicAdd(&p.xdot[k], xd);
+            }
         }
     }
 }
 
+extern "C" long red_gemm_nblocks(long R) {
+    long chunks = (R + 63) / 64;
+    return chunks < 2432 ? chunks : 2432;
+}
+
 extern "C" void red_gemm_launch(RedGemmParams p, int is_f32, hipStream_t s) {
-    long chunks = (p.R + 63) / 64;
-    long blocks = chunks < 2432 ? chunks : 2432;
+    long blocks = red_gemm_nblocks(p.R);
     dim3 grid((unsigned)blocks), block(256);
 #define DISPATCH(TT)                                                        \
     do {                                                                    \

@@ -358,3 +358,40 @@ def test_full_model_dual_rwd_s5_gpu():
     loss.backward()
     for n, p in model.named_parameters():
         assert p.grad is not None and torch.isfinite(p.grad).all(), n
+
+
+def test_deterministic_mode_bitwise_reproducible():
+    """torch.use_deterministic_algorithms(True) routes the atomic reductions
+    through per-block workspaces: identical seeds give bit-identical training
+    (SURVEY §5's reproducibility-as-race-detector obligation)."""
+    from mpgcn_amd.graph import build_supports
+    from mpgcn_amd.models import MPGCN
+
+    torch.use_deterministic_algorithms(True)
+    try:
+        results = []
+        for _ in range(2):
+            torch.manual_seed(21)
+            torch.cuda.manual_seed_all(21)
+            model = MPGCN(M=2, K=3, input_dim=1, lstm_hidden_dim=32,
+                          lstm_num_layers=1, gcn_hidden_dim=32, gcn_num_layers=3,
+                          num_nodes=48, compute_dtype=torch.bfloat16).to(DEV)
+            opt = torch.optim.Adam(model.parameters(), lr=1e-3)
+            x = torch.rand(4, 7, 48, 48, 1, device=DEV)
+            y = torch.rand(4, 1, 48, 48, 1, device=DEV)
+            flow = torch.rand(4, 48, 48, device=DEV)
+            Gs = build_supports(torch.rand(1, 48, 48, device=DEV),
+                                "random_walk_diffusion", 2)[0]
+            Go = build_supports(flow, "random_walk_diffusion", 2)
+            Gd = build_supports(flow.transpose(-2, -1), "random_walk_diffusion", 2)
+            losses = []
+            for _ in range(8):
+                loss = torch.nn.functional.mse_loss(model(x, [Gs, (Go, Gd)]), y)
+                opt.zero_grad()
+                loss.backward()
+                opt.step()
+                losses.append(loss.item())
+            results.append(losses)
+        assert results[0] == results[1], (results[0][-1], results[1][-1])
+    finally:
+        torch.use_deterministic_algorithms(False)
