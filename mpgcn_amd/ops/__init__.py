@@ -64,6 +64,8 @@ from mpgcn_amd.ops.functional import (  # noqa: E402,F401
     bdgcn_layer,
     fused_lstm_last,
     linear_act,
+    mode1_proj,
+    mode2_bias_act,
 )
 
 __all__ = [
@@ -74,4 +76,6 @@ __all__ = [
     "bdgcn_layer",
     "fused_lstm_last",
     "linear_act",
+    "mode1_proj",
+    "mode2_bias_act",
 ]

@@ -160,6 +160,102 @@ def bdgcn_layer(X, W, bias, gop: GraphOperator, relu: bool = True):
     return eager.bdgcn_layer_eager(X, gop.Go, gop.Gd, W, bias, "relu" if relu else "none")
 
 
+class _Mode1ProjFn(torch.autograd.Function):
+    """mode-1 + projection half of a BDGCN layer (region-partition path,
+    mpgcn_amd/parallel/region.py): the layer splits at the all-to-all seam, so
+    each half is its own autograd node. X may be destination-sharded
+    (B, N, Nd, C) with Nd = N/P; the axis kernels take rectangular shapes."""
+
+    @staticmethod
+    def forward(ctx, X, W, gop: GraphOperator):
+        ext = _ops.get_ext()
+        B, No, Nd, C = X.shape
+        S = gop.S
+        Hdim = W.shape[1]
+        U = ext.bdgcn_mode1(X, gop.GoT)  # (B, No, Nd, S, C)
+        Wre = eager.reorder_projection_weight(W, S, C).contiguous()
+        Vflat = _row_gemm_chunked(ext, U.reshape(B * No * Nd, S * C), Wre, None, False)
+        ctx.save_for_backward(U, Wre)
+        ctx.gop = gop
+        ctx.dims = (B, No, Nd, S, C, Hdim)
+        return Vflat.view(B, No, Nd, S * Hdim)
+
+    @staticmethod
+    def backward(ctx, dV):
+        ext = _ops.get_ext()
+        U, Wre = ctx.saved_tensors
+        B, No, Nd, S, C, Hdim = ctx.dims
+        R = B * No * Nd
+        dVflat = dV.reshape(R, S * Hdim).contiguous()
+        dWreT, _, _ = ext.red_gemm(dVflat, U.reshape(R, S * C), False, None, 0, 0)
+        dWre = dWreT.t().to(dV.dtype)
+        dW = dWre.reshape(S, C, S, Hdim).permute(0, 2, 1, 3).reshape(S * S * C, Hdim)
+        dU = _row_gemm_chunked(ext, dVflat, Wre.t().contiguous(), None, False)
+        dX = ext.bdgcn_mode1_bwd(dU.view(B, No, Nd, S, C), ctx.gop.A3T)
+        return dX, dW, None
+
+
+class _Mode2BiasActFn(torch.autograd.Function):
+    """mode-2 + bias + activation half of a BDGCN layer (region-partition
+    path). V is origin-sharded (B, Nm, N, S, H) with Nm = N/P."""
+
+    @staticmethod
+    def forward(ctx, V, bias, gop: GraphOperator, relu: bool):
+        ext = _ops.get_ext()
+        B, Nm, N, S, Hdim = V.shape
+        bias_f32 = bias.float().contiguous() if bias is not None else None
+        Y = ext.bdgcn_mode2(V.reshape(B, Nm, N * S, Hdim).contiguous(),
+                            gop.A2T, bias_f32, relu, N, S)
+        ctx.save_for_backward(Y)
+        ctx.gop = gop
+        ctx.relu = relu
+        ctx.has_bias = bias is not None
+        ctx.dims = (B, Nm, N, S, Hdim)
+        return Y  # (B, Nm, N, H)
+
+    @staticmethod
+    def backward(ctx, dH):
+        ext = _ops.get_ext()
+        (Y,) = ctx.saved_tensors
+        B, Nm, N, S, Hdim = ctx.dims
+        dH = dH.contiguous()
+        if dH.dtype == torch.bfloat16 and (Hdim & (Hdim - 1)) == 0 and Hdim >= 8:
+            dY, dbias = ext.relu_bwd_colsum(dH, Y, ctx.relu)
+            dY = dY.view_as(dH)
+            if not ctx.has_bias:
+                dbias = None
+        else:
+            dY = dH * (Y > 0).to(dH.dtype) if ctx.relu else dH
+            dbias = dY.sum(dim=(0, 1, 2)).to(torch.float32) if ctx.has_bias else None
+        dV = ext.bdgcn_mode2_bwd(dY, ctx.gop.A2, S)  # (B, Nm, N, S, H)
+        return dV, dbias if ctx.has_bias else None, None, None
+
+
+def mode1_proj(X, W, gop: GraphOperator):
+    """mode-1 contraction + projection: (B, N, Nd, C) -> (B, N, Nd, S*H)."""
+    if X.is_cuda:
+        return _Mode1ProjFn.apply(X, W, gop)
+    S = gop.S
+    C = X.shape[-1]
+    B, No, Nd = X.shape[:3]
+    U = eager.mode1_apply(X, gop.Go)
+    Wre = eager.reorder_projection_weight(W, S, C)
+    return (U.reshape(B * No * Nd, S * C) @ Wre).view(B, No, Nd, -1)
+
+
+def mode2_bias_act(V, bias, gop: GraphOperator, relu: bool = True):
+    """mode-2 contraction + bias + act: (B, Nm, N, S, H) -> (B, Nm, N, H)."""
+    if V.is_cuda:
+        return _Mode2BiasActFn.apply(V, bias, gop, relu)
+    if gop.Gd.dim() == 3:
+        Y = torch.einsum("scd,bmcsh->bmdh", gop.Gd.to(V.dtype), V)
+    else:
+        Y = torch.einsum("bscd,bmcsh->bmdh", gop.Gd.to(V.dtype), V)
+    if bias is not None:
+        Y = Y + bias.to(Y.dtype)
+    return torch.relu(Y) if relu else Y
+
+
 class _FusedLSTMLastFn(torch.autograd.Function):
     """Single-layer LSTM over R sequences, returning only the LAST hidden state
     (the only timestep MPGCN consumes, reference MPGCN.py:104). Input dim 1.

@@ -35,32 +35,34 @@ const float* bias_ptr(const c10::optional<torch::Tensor>& b) {
 // U[b,m,d,o,l] = sum_n GT[(b,)o,m,n] X[b,n,d,l].
 // X: (B, N, N, C); GT: (S, N, N) static or (B, S, N, N) dynamic, ALREADY
 // transposed per support (GT[..., m, n] = G[..., n, m]). Out: (B, N, N, S, C).
+// Rectangular: X may be destination-sharded, (B, No, Nd, C) with Nd != No
+// (the region-partition path, mpgcn_amd/parallel/region.py).
 torch::Tensor bdgcn_mode1(torch::Tensor X, torch::Tensor GT) {
     check_in(X, "X");
     check_in(GT, "GT");
     const bool dyn = GT.dim() == 4;
-    const long B = X.size(0), N = X.size(1), C = X.size(3);
+    const long B = X.size(0), No = X.size(1), Nd = X.size(2), C = X.size(3);
     const long S = dyn ? GT.size(1) : GT.size(0);
-    TORCH_CHECK(X.size(2) == N && GT.size(-1) == N && GT.size(-2) == N, "shape mismatch");
+    TORCH_CHECK(GT.size(-1) == No && GT.size(-2) == No, "shape mismatch");
     TORCH_CHECK(!dyn || GT.size(0) == B, "dynamic GT batch mismatch");
     TORCH_CHECK(B * S <= 65535, "too many instances");
-    auto U = torch::empty({B, N, N, S, C}, X.options());
+    auto U = torch::empty({B, No, Nd, S, C}, X.options());
 
     AxisGemmParams p{};
     p.AT = GT.data_ptr();
     p.X = X.data_ptr();
     p.OUT = U.data_ptr();
-    p.M = (int)N; p.K = (int)N; p.L = (int)(N * C);
-    p.a_div = (int)S; p.a_bs1 = dyn ? S * N * N : 0; p.a_bs2 = N * N;
-    p.x_div = (int)S; p.x_bs1 = N * N * C; p.x_bs2 = 0;
-    p.o_div = (int)S; p.o_bs1 = N * N * S * C; p.o_bs2 = C;
-    p.kdiv = 1; p.k_lo = N * C;
+    p.M = (int)No; p.K = (int)No; p.L = (int)(Nd * C);
+    p.a_div = (int)S; p.a_bs1 = dyn ? S * No * No : 0; p.a_bs2 = No * No;
+    p.x_div = (int)S; p.x_bs1 = No * Nd * C; p.x_bs2 = 0;
+    p.o_div = (int)S; p.o_bs1 = No * Nd * S * C; p.o_bs2 = C;
+    p.kdiv = 1; p.k_lo = Nd * C;
     p.qdiv = 0;
-    p.o_row = N * S * C;
+    p.o_row = Nd * S * C;
     p.ogdiv = (int)C; p.og_hi = S * C;
     const int ch = chunk_elems(X);
-    p.a_vec = (N % ch == 0);
-    p.x_vec = ((N * C) % ch == 0) && (C % ch == 0);
+    p.a_vec = (No % ch == 0);
+    p.x_vec = ((Nd * C) % ch == 0) && (C % ch == 0);
     axis_gemm_launch(p, (int)(B * S), is_f32(X), stream());
     return U;
 }
@@ -75,23 +77,24 @@ torch::Tensor bdgcn_mode2(torch::Tensor V, torch::Tensor A2T,
     check_in(A2T, "A2T");
     const bool dyn = A2T.dim() == 3;
     const long B = V.size(0);
+    const long Nm = V.size(1);  // origin rows held locally (== N unsharded)
     const long H = V.size(-1);
     TORCH_CHECK(A2T.size(-1) == N * S && A2T.size(-2) == N, "A2T shape mismatch");
-    TORCH_CHECK(B * N <= 65535, "too many instances");
-    auto Y = torch::empty({B, N, N, H}, V.options());
+    TORCH_CHECK(B <= 65535, "too many instances");
+    auto Y = torch::empty({B, Nm, N, H}, V.options());
 
     // One GEMM per batch element: OUT[d, (m,h)] = sum_cs A2T[d,cs] V[b,m,cs,h]
     // — folding the m axis into the L (column) dimension gives each staged
-    // A-tile (the graph) 256x more MFMA work than per-(b,m) instances.
+    // A-tile (the graph) full-tile MFMA work instead of per-(b,m) instances.
     AxisGemmParams p{};
     p.AT = A2T.data_ptr();
     p.X = V.data_ptr();
     p.OUT = Y.data_ptr();
     p.bias = bias_ptr(bias);
-    p.M = (int)N; p.K = (int)(N * S); p.L = (int)(N * H);
+    p.M = (int)N; p.K = (int)(N * S); p.L = (int)(Nm * H);
     p.a_div = 1; p.a_bs1 = dyn ? N * N * S : 0; p.a_bs2 = 0;
-    p.x_div = 1; p.x_bs1 = N * N * S * H; p.x_bs2 = 0;
-    p.o_div = 1; p.o_bs1 = N * N * H; p.o_bs2 = 0;
+    p.x_div = 1; p.x_bs1 = Nm * N * S * H; p.x_bs2 = 0;
+    p.o_div = 1; p.o_bs1 = Nm * N * H; p.o_bs2 = 0;
     p.kdiv = 1; p.k_lo = H;
     p.qdiv = (int)H; p.q_hi = N * S * H;   // q = (m, h)
     p.o_row = H;                           // out row = d
@@ -107,14 +110,17 @@ torch::Tensor bdgcn_mode2(torch::Tensor V, torch::Tensor A2T,
 
 // dV[b,m,cs,h] = sum_d A2[(b,)cs,d] dY[b,m,d,h].
 // dY: (B, N, N, H); A2: (N*S, N) or (B, N*S, N), A2[c*S+s, d] = Gd[s, c, d].
+// Rectangular: dY may be origin-sharded (B, Nm, N, H) with Nm != N
+// (region partition, mpgcn_amd/parallel/region.py); N comes from A2.
 torch::Tensor bdgcn_mode2_bwd(torch::Tensor dY, torch::Tensor A2, long S) {
     check_in(dY, "dY");
     check_in(A2, "A2");
     const bool dyn = A2.dim() == 3;
-    const long B = dY.size(0), N = dY.size(1), H = dY.size(3);
-    TORCH_CHECK(A2.size(-2) == N * S && A2.size(-1) == N, "A2 shape mismatch");
-    TORCH_CHECK(B * N <= 65535, "too many instances");
-    auto dV = torch::empty({B, N, N, S, H}, dY.options());
+    const long B = dY.size(0), Nm = dY.size(1), H = dY.size(3);
+    const long N = A2.size(-1);
+    TORCH_CHECK(A2.size(-2) == N * S && dY.size(2) == N, "A2 shape mismatch");
+    TORCH_CHECK(B <= 65535, "too many instances");
+    auto dV = torch::empty({B, Nm, N, S, H}, dY.options());
 
     // One GEMM per batch element (m folded into L, as in bdgcn_mode2):
     // dV[cs, (m,h)] = sum_d A2[cs,d] dY[b,m,d,h]
@@ -122,10 +128,10 @@ torch::Tensor bdgcn_mode2_bwd(torch::Tensor dY, torch::Tensor A2, long S) {
     p.AT = A2.data_ptr();
     p.X = dY.data_ptr();
     p.OUT = dV.data_ptr();
-    p.M = (int)(N * S); p.K = (int)N; p.L = (int)(N * H);
+    p.M = (int)(N * S); p.K = (int)N; p.L = (int)(Nm * H);
     p.a_div = 1; p.a_bs1 = dyn ? N * S * N : 0; p.a_bs2 = 0;
-    p.x_div = 1; p.x_bs1 = N * N * H; p.x_bs2 = 0;
-    p.o_div = 1; p.o_bs1 = N * N * S * H; p.o_bs2 = 0;
+    p.x_div = 1; p.x_bs1 = Nm * N * H; p.x_bs2 = 0;
+    p.o_div = 1; p.o_bs1 = Nm * N * S * H; p.o_bs2 = 0;
     p.kdiv = 1; p.k_lo = H;
     p.qdiv = (int)H; p.q_hi = N * H;       // q = (m, h)
     p.o_row = H;                           // out row = cs
@@ -139,29 +145,31 @@ torch::Tensor bdgcn_mode2_bwd(torch::Tensor dY, torch::Tensor A2, long S) {
 
 // dX[b,n,d,l] = sum_{om} A3T[(b,)n,o*N+m] dU[b,m,d,o,l].
 // dU: (B, N, N, S, C); A3T: (N, S*N) or (B, N, S*N), A3T[n, o*N+m] = G[o,n,m].
+// Rectangular: dU may be destination-sharded (B, No, Nd, S, C), Nd != No.
 torch::Tensor bdgcn_mode1_bwd(torch::Tensor dU, torch::Tensor A3T) {
     check_in(dU, "dU");
     check_in(A3T, "A3T");
     const bool dyn = A3T.dim() == 3;
-    const long B = dU.size(0), N = dU.size(1), S = dU.size(3), C = dU.size(4);
-    TORCH_CHECK(A3T.size(-2) == N && A3T.size(-1) == S * N, "A3T shape mismatch");
+    const long B = dU.size(0), No = dU.size(1), Nd = dU.size(2);
+    const long S = dU.size(3), C = dU.size(4);
+    TORCH_CHECK(A3T.size(-2) == No && A3T.size(-1) == S * No, "A3T shape mismatch");
     TORCH_CHECK(B <= 65535, "too many instances");
-    auto dX = torch::empty({B, N, N, C}, dU.options());
+    auto dX = torch::empty({B, No, Nd, C}, dU.options());
 
     AxisGemmParams p{};
     p.AT = A3T.data_ptr();
     p.X = dU.data_ptr();
     p.OUT = dX.data_ptr();
-    p.M = (int)N; p.K = (int)(S * N); p.L = (int)(N * C);
-    p.a_div = 1; p.a_bs1 = dyn ? N * S * N : 0; p.a_bs2 = 0;
-    p.x_div = 1; p.x_bs1 = N * N * S * C; p.x_bs2 = 0;
-    p.o_div = 1; p.o_bs1 = N * N * C; p.o_bs2 = 0;
-    p.kdiv = (int)N; p.k_hi = C; p.k_lo = N * S * C;  // k = o*N + m
-    p.qdiv = (int)C; p.q_hi = S * C;                  // q = d*C + l
-    p.o_row = N * C;
+    p.M = (int)No; p.K = (int)(S * No); p.L = (int)(Nd * C);
+    p.a_div = 1; p.a_bs1 = dyn ? No * S * No : 0; p.a_bs2 = 0;
+    p.x_div = 1; p.x_bs1 = No * Nd * S * C; p.x_bs2 = 0;
+    p.o_div = 1; p.o_bs1 = No * Nd * C; p.o_bs2 = 0;
+    p.kdiv = (int)No; p.k_hi = C; p.k_lo = Nd * S * C;  // k = o*No + m
+    p.qdiv = (int)C; p.q_hi = S * C;                    // q = d*C + l
+    p.o_row = Nd * C;
     p.ogdiv = 0;
     const int ch = chunk_elems(dU);
-    p.a_vec = ((S * N) % ch == 0);
+    p.a_vec = ((S * No) % ch == 0);
     p.x_vec = (C % ch == 0);
     axis_gemm_launch(p, (int)B, is_f32(dU), stream());
     return dX;

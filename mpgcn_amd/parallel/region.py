@@ -29,7 +29,14 @@ from __future__ import annotations
 import torch
 import torch.distributed as dist
 
-from mpgcn_amd.ops import eager, fused_lstm_last, linear_act
+from mpgcn_amd.ops import (
+    GraphOperator,
+    eager,
+    fused_lstm_last,
+    linear_act,
+    mode1_proj,
+    mode2_bias_act,
+)
 
 
 class _AllToAllShard(torch.autograd.Function):
@@ -93,40 +100,29 @@ def shard_dest(x_full: torch.Tensor, rank: int, P: int) -> torch.Tensor:
     return x_full[..., rank * Nl:(rank + 1) * Nl, :].contiguous()
 
 
-def bdgcn_layer_sharded(Xd, Go, Gd, W, bias, group=None, relu=True):
+def bdgcn_layer_sharded(Xd, gop: GraphOperator, W, bias, group=None, relu=True):
     """One BDGCN layer on a destination-sharded input.
 
-    Xd: (B, N, N/P, C) destination-sharded; Go/Gd: full (S, N, N) or
-    (B, S, N, N) (graphs are O(S*N^2), replicated); W: (C*S*S, H).
+    Xd: (B, N, N/P, C) destination-sharded; gop: GraphOperator over the full
+    (replicated) supports — graphs are O(S*N^2); W: (C*S*S, H).
     Returns the next layer's destination-sharded input (B, N, N/P, H).
 
-    Math identical to mpgcn_amd.ops.eager.bdgcn_layer_eager (verified by
-    tests/test_region.py against the unsharded computation).
+    On GPU the two local halves run through the HIP axis kernels (the
+    bindings accept rectangular origin/destination extents); on CPU the same
+    math runs as einsums. Math identical to
+    mpgcn_amd.ops.eager.bdgcn_layer_eager (verified by tests/test_region.py
+    against the unsharded computation).
     """
-    S = Go.shape[-3]
+    S = gop.S
     B, N, Nl, C = Xd.shape
     Hdim = W.shape[1]
 
-    # mode-1: contract the (full, local) origin axis
-    if Go.dim() == 3:
-        U = torch.einsum("onm,bndl->bmdol", Go, Xd)  # (B, N, Nl, S, C)
-    else:
-        U = torch.einsum("bonm,bndl->bmdol", Go, Xd)
-    # projection (row-local)
-    Wre = eager.reorder_projection_weight(W, S, C)
-    V = (U.reshape(B * N * Nl, S * C) @ Wre).view(B, N, Nl, S * Hdim)
+    # mode-1 (contract the full, local origin axis) + row-local projection
+    V = mode1_proj(Xd, W, gop)  # (B, N, Nl, S*H)
     # re-shard: destination-sharded -> origin-sharded (full dest axis)
     Vo = dest_to_origin(V, group)  # (B, Nl, N, S*H)
-    Vo = Vo.view(B, Nl, N, S, Hdim)
-    # mode-2: contract the (full, local) destination axis
-    if Gd.dim() == 3:
-        Y = torch.einsum("scd,bmcsh->bmdh", Gd, Vo)  # (B, Nl, N, H)
-    else:
-        Y = torch.einsum("bscd,bmcsh->bmdh", Gd, Vo)
-    if bias is not None:
-        Y = Y + bias.to(Y.dtype)
-    if relu:
-        Y = torch.relu(Y)
+    # mode-2: contract the (full, local) destination axis, + bias + act
+    Y = mode2_bias_act(Vo.view(B, Nl, N, S, Hdim), bias, gop, relu)
     # re-shard back for the next layer's mode-1
     return origin_to_dest(Y, group)  # (B, N, Nl, H)
 
@@ -159,7 +155,7 @@ def mpgcn_forward_sharded(model, x_seq_shard, G_list, group=None):
         gop = gops[m]
         for layer in branch["spatial"]:
             X = bdgcn_layer_sharded(
-                X, gop.Go, gop.Gd, layer.W.to(cd), layer.b, group, relu=layer.relu
+                X, gop, layer.W.to(cd), layer.b, group, relu=layer.relu
             )
         fc = branch["fc"][0]
         out = linear_act(X.reshape(B * N * Nl, -1), fc.weight.to(cd), fc.bias, True)
