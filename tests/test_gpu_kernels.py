@@ -440,3 +440,56 @@ def test_rectangular_sharded_halves_match_eager():
     assert relerr(W.grad, W32.grad) < 4e-2
     assert relerr(bias.grad, bias32.grad) < 4e-2
     torch.cuda.synchronize()
+
+
+@pytest.mark.timeout(600)
+def test_capacity_4096_regions_train_step():
+    """Config #5 scale (BASELINE.json: 4096-region OD) on one GPU: a full
+    train step at N=4096, batch 1 — exercises the >1e9-element axis-kernel
+    index ranges (offsets are 64-bit throughout) and the 288 GB HBM sizing.
+    Numerics at this scale are pinned by a layer comparison at N=2048."""
+    from mpgcn_amd.graph import build_supports
+    from mpgcn_amd.models import MPGCN
+    from mpgcn_amd.ops import GraphOperator, bdgcn_layer, eager
+
+    # layer numerics at N=2048 (same code path, 8x less eager work)
+    torch.manual_seed(12)
+    Nn, S, C, Hd = 2048, 3, 32, 32
+    X32 = torch.randn(1, Nn, Nn, C, device=DEV)
+    Go = torch.randn(S, Nn, Nn, device=DEV) / Nn**0.5
+    Gd = torch.randn(S, Nn, Nn, device=DEV) / Nn**0.5
+    W32 = torch.randn(C * S * S, Hd, device=DEV) / (C * S * S) ** 0.5
+    b32 = torch.randn(Hd, device=DEV)
+    ref = eager.bdgcn_layer_eager(X32, Go, Gd, W32, b32, "relu")
+    gop = GraphOperator(Go.bfloat16(), Gd.bfloat16())
+    out = bdgcn_layer(X32.bfloat16(), W32.bfloat16(), b32, gop, True)
+    rel = (out.float() - ref).norm() / ref.norm()
+    assert rel < 3e-2, rel.item()
+    del X32, Go, Gd, ref, out, gop
+    torch.cuda.empty_cache()
+
+    # full model step at N=4096
+    N4 = 4096
+    model = MPGCN(M=2, K=3, input_dim=1, lstm_hidden_dim=32, lstm_num_layers=1,
+                  gcn_hidden_dim=32, gcn_num_layers=3, num_nodes=N4,
+                  compute_dtype=torch.bfloat16).to(DEV)
+    opt = torch.optim.Adam(model.parameters(), lr=1e-3)
+    x = torch.rand(1, 7, N4, N4, 1, device=DEV)
+    y = torch.rand(1, 1, N4, N4, 1, device=DEV)
+    flow = torch.rand(1, N4, N4, device=DEV)
+    Gs = build_supports(torch.rand(1, N4, N4, device=DEV),
+                        "random_walk_diffusion", 2)[0]
+    Go4 = build_supports(flow, "random_walk_diffusion", 2)
+    Gd4 = build_supports(flow.transpose(-2, -1), "random_walk_diffusion", 2)
+    losses = []
+    for _ in range(3):
+        loss = torch.nn.functional.mse_loss(model(x, [Gs, (Go4, Gd4)]), y)
+        opt.zero_grad()
+        loss.backward()
+        opt.step()
+        losses.append(loss.item())
+    torch.cuda.synchronize()
+    assert all(torch.isfinite(torch.tensor(losses))), losses
+    assert losses[-1] < losses[0], losses
+    peak = torch.cuda.max_memory_allocated() / 2**30
+    assert peak < 200, f"peak {peak:.1f} GiB"
