@@ -80,3 +80,24 @@ extern "C" void relu_bwd_colsum_launch(ReluBwdParams p, hipStream_t s) {
     const size_t smem = p.H * 4 + (p.det ? 256 * 8 * 4 : 0);
     relu_bwd_colsum_kernel<<<dim3((unsigned)blocks), dim3(256), smem, s>>>(p);
 }
+
+// Column sum of a per-block f32 workspace: out[e] = sum_b ws[b*E + e], fixed
+// ascending-b order (deterministic). Consecutive threads read consecutive e,
+// so every iteration is a fully coalesced row segment. Replaces three
+// aten::sum launches per fused-LSTM backward (each small reduction was
+// wall-expensive when co-scheduled with the other branch's axis kernels).
+__launch_bounds__(256) __global__ void slab_colsum_kernel(
+    const float* __restrict__ ws, float* __restrict__ out, long nb, long E) {
+    const long e = (long)blockIdx.x * 256 + threadIdx.x;
+    if (e >= E) return;
+    float s = 0.f;
+    long off = e;
+    for (long b = 0; b < nb; ++b, off += E) s += ws[off];
+    out[e] = s;
+}
+
+extern "C" void slab_colsum_launch(const float* ws, float* out, long nb, long E,
+                                   hipStream_t s) {
+    slab_colsum_kernel<<<dim3((unsigned)((E + 255) / 256)), dim3(256), 0, s>>>(
+        ws, out, nb, E);
+}

@@ -376,7 +376,18 @@ std::vector<torch::Tensor> lstm_fused_bwd(torch::Tensor x, long T_logical,
     p.dx = need_dx ? dx.data_ptr() : nullptr;
     p.R = R; p.T = T;
     lstm_fused_bwd_launch(p, stream());
-    return {ws_dw.sum(0), ws_db.sum(0), ws_dwih.sum(0), dx};
+    // in-kernel fixed-order workspace reduction (deterministic; three
+    // aten::sum launches were wall-expensive under branch-stream overlap)
+    auto dwhh = torch::empty({128, 32}, f32);
+    auto dbias = torch::empty({128}, f32);
+    auto dwih = torch::empty({128}, f32);
+    slab_colsum_launch(ws_dw.data_ptr<float>(), dwhh.data_ptr<float>(), nb,
+                       128 * 32, stream());
+    slab_colsum_launch(ws_db.data_ptr<float>(), dbias.data_ptr<float>(), nb,
+                       128, stream());
+    slab_colsum_launch(ws_dwih.data_ptr<float>(), dwih.data_ptr<float>(), nb,
+                       128, stream());
+    return {dwhh, dbias, dwih, dx};
 }
 
 // Fused ReLU backward + bias-grad column sum: dY = dH * 1[Y>0], dbias=colsum.
@@ -402,7 +413,13 @@ std::vector<torch::Tensor> relu_bwd_colsum(torch::Tensor dH, torch::Tensor Y,
     p.mask = mask ? 1 : 0;
     p.det = det ? 1 : 0;
     relu_bwd_colsum_launch(p, stream());
-    return {dY, det ? colsum.sum(0) : colsum};
+    if (det) {
+        auto cs = torch::empty({H}, dH.options().dtype(torch::kFloat));
+        slab_colsum_launch(colsum.data_ptr<float>(), cs.data_ptr<float>(), nb,
+                           H, stream());
+        return {dY, cs};
+    }
+    return {dY, colsum};
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {

@@ -394,20 +394,32 @@ __launch_bounds__(256) __global__ void lstm_fused_bwd_kernel(LstmFusedParams p) 
     for (int i = tid; i < LF_G4 * LF_H; i += 256) ws[i] = red[i];
 
     // dbias / dwih: lane covers n = nf*16 + lrow for its 8 nf; rows are summed
-    // already; reduce over kgrp lanes (same n) then across waves via LDS
-    float* redb = (float*)&ldsHT[0][0];  // 2 * 4H floats
-    for (int i = tid; i < 2 * LF_G4; i += 256) redb[i] = 0.f;
+    // already. Reduce over the 4 kgrp lanes sharing n with a fixed-order
+    // shuffle tree, stage per-wave rows in LDS, then sum waves in fixed order
+    // — bitwise-reproducible (an LDS float atomicAdd here made the whole
+    // backward nondeterministic across runs).
+    float* redb = (float*)&ldsHT[0][0];  // 4 waves * 2 * 4H floats
     __syncthreads();
-    // every lane LDS-atomically adds its partial into the shared row
 #pragma unroll
     for (int nf = 0; nf < 8; ++nf) {
-        atomicAdd(&redb[nf * 16 + lrow], db_acc[nf]);
-        atomicAdd(&redb[LF_G4 + nf * 16 + lrow], dwih_acc[nf]);
+        float vb = db_acc[nf], vw = dwih_acc[nf];
+        vb += __shfl_down(vb, 32, 64); vb += __shfl_down(vb, 16, 64);
+        vw += __shfl_down(vw, 32, 64); vw += __shfl_down(vw, 16, 64);
+        if (lane < 16) {
+            redb[(w * 2 + 0) * LF_G4 + nf * 16 + lane] = vb;
+            redb[(w * 2 + 1) * LF_G4 + nf * 16 + lane] = vw;
+        }
     }
     __syncthreads();
     for (int i = tid; i < LF_G4; i += 256) {
-        p.ws_db[(long)blockIdx.x * LF_G4 + i] = redb[i];
-        p.ws_dwih[(long)blockIdx.x * LF_G4 + i] = redb[LF_G4 + i];
+        float sb = 0.f, sw = 0.f;
+#pragma unroll
+        for (int wv = 0; wv < 4; ++wv) {
+            sb += redb[(wv * 2 + 0) * LF_G4 + i];
+            sw += redb[(wv * 2 + 1) * LF_G4 + i];
+        }
+        p.ws_db[(long)blockIdx.x * LF_G4 + i] = sb;
+        p.ws_dwih[(long)blockIdx.x * LF_G4 + i] = sw;
     }
 }
 
