@@ -43,6 +43,10 @@ def main():
     ap.add_argument("--partition", type=str, default="dp", choices=["dp", "region"],
                     help="multi-rank strategy: data parallel (weak scaling) or "
                          "region partition (activation grid sharded across ranks)")
+    ap.add_argument("--graph", action="store_true",
+                    help="capture the whole train step in a hipGraph and replay "
+                         "it (single-rank; removes host launch gaps — matters "
+                         "in the small-batch regime)")
     args = ap.parse_args()
 
     import torch.distributed as dist
@@ -83,7 +87,8 @@ def main():
                       lstm_num_layers=1, gcn_hidden_dim=H, gcn_num_layers=3,
                       num_nodes=N, compute_dtype=cdtype,
                       fusion=args.fusion).to(device)
-    opt = torch.optim.Adam(model.parameters(), lr=1e-4)
+    use_graph = args.graph and is_cuda and world == 1 and args.impl == "native"
+    opt = torch.optim.Adam(model.parameters(), lr=1e-4, capturable=use_graph)
     reducer = GradAllReducer(model, ctx)
     criterion = torch.nn.MSELoss()
 
@@ -102,8 +107,10 @@ def main():
     if region:
         from mpgcn_amd.parallel.region import mpgcn_forward_sharded, shard_dest
 
-    def step(i: int):
-        g = (torch.arange(B, device=device) * 7 + i) % (T_pool - T - 1)
+    i_buf = torch.zeros((), dtype=torch.long, device=device)
+
+    def step_body():
+        g = (torch.arange(B, device=device) * 7 + i_buf) % (T_pool - T - 1)
         x = pool[g.unsqueeze(1) + torch.arange(T, device=device)]  # (B,T,N,N,1)
         y = pool[(g + T).unsqueeze(1) + torch.arange(1, device=device)]
         key = (g + T) % 7
@@ -119,11 +126,35 @@ def main():
         else:
             y_pred = model(x, G_list)
         loss = criterion(y_pred, y)
-        opt.zero_grad(set_to_none=True)
+        # graph mode needs stable grad buffers across replays
+        opt.zero_grad(set_to_none=not use_graph)
         loss.backward()
         reducer.finalize()
         opt.step()
         return loss
+
+    if use_graph:
+        # warm up on a side stream (allocator + lazy state), then capture one
+        # full step — dynamic-support build, forward, backward, Adam — into a
+        # hipGraph; replay re-reads i_buf so the data window still advances
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(3):
+                step_body()
+        torch.cuda.current_stream().wait_stream(side)
+        torch.cuda.synchronize()
+        hip_graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(hip_graph):
+            graph_loss = step_body()
+
+    def step(i: int):
+        if use_graph:
+            i_buf.fill_(i)
+            hip_graph.replay()
+            return graph_loss
+        i_buf.fill_(i)
+        return step_body()
 
     def barrier_sync():
         if ctx.enabled:

@@ -156,20 +156,26 @@ class MPGCN(nn.Module):
             # overlap branch 0's — autograd replays each op's backward on the
             # stream it was recorded on, so the overlap holds in backward too.
             main = torch.cuda.current_stream()
+            # under hipGraph capture the graph pool owns block lifetimes and
+            # record_stream is not permitted; the wait_stream edges still
+            # capture as graph dependencies, so the overlap is preserved
+            capturing = torch.cuda.is_current_stream_capturing()
             while len(self._streams) < self.M - 1:
                 self._streams.append(torch.cuda.Stream())
             branch_out = [None] * self.M
             for m in range(1, self.M):
                 s = self._streams[m - 1]
                 s.wait_stream(main)
-                lstm_in.record_stream(s)  # allocated on main, read on s
+                if not capturing:
+                    lstm_in.record_stream(s)  # allocated on main, read on s
                 with torch.cuda.stream(s):
                     branch_out[m] = run_branch(m)
             branch_out[0] = run_branch(0)
             for m in range(1, self.M):
                 main.wait_stream(self._streams[m - 1])
-                # keep the side-stream allocations alive for the main stream
-                branch_out[m].record_stream(main)
+                if not capturing:
+                    # keep the side-stream allocations alive for the main stream
+                    branch_out[m].record_stream(main)
         else:
             branch_out = [run_branch(m) for m in range(self.M)]
         stacked = torch.stack(branch_out, dim=-1)
