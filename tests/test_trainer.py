@@ -4,6 +4,8 @@ format match the reference (Model_Trainer.py:88,129,180)."""
 
 import re
 
+import pytest
+
 import torch
 
 from mpgcn_amd.data import DataGenerator, DataInput
@@ -92,3 +94,29 @@ def test_loss_variants(tmp_path):
     for loss in ("MAE", "Huber"):
         params, trainer, loaders = _setup(tmp_path, loss=loss, num_epochs=1)
         trainer.train(loaders, ["train", "validate"])
+
+
+@pytest.mark.timeout(300)
+def test_main_cli_end_to_end(tmp_path):
+    """Main.py contract (reference Main.py:7-67): train mode trains, writes
+    the checkpoint, then immediately evaluates and appends both score lines."""
+    import subprocess
+    import sys
+    from pathlib import Path
+
+    repo = Path(__file__).resolve().parent.parent
+    out = tmp_path / "cli_out"
+    r = subprocess.run(
+        [sys.executable, str(repo / "Main.py"), "-GPU", "cpu",
+         "-synthetic-nodes", "16", "-synthetic-days", "60", "-epoch", "1",
+         "-out", str(out)],
+        capture_output=True, text=True, timeout=240, cwd=str(repo),
+    )
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert (out / "MPGCN_od.pkl").exists()
+    scores = (out / "MPGCN_prediction_scores.txt").read_text().splitlines()
+    assert scores[0].startswith("train, MSE, RMSE, MAE, MAPE, ")
+    assert scores[1].startswith("test, MSE, RMSE, MAE, MAPE, ")
+    ckpt = torch.load(out / "MPGCN_od.pkl", map_location="cpu",
+                      weights_only=True)
+    assert set(ckpt) == {"epoch", "state_dict"}
