@@ -109,8 +109,11 @@ def main():
 
     i_buf = torch.zeros((), dtype=torch.long, device=device)
 
-    def step_body():
-        g = (torch.arange(B, device=device) * 7 + i_buf) % (T_pool - T - 1)
+    def step_body(i=None):
+        # graph mode reads the device scalar (re-read at every replay);
+        # eager mode keeps the host int to avoid extra device round-trips
+        g = (torch.arange(B, device=device) * 7 + (i_buf if i is None else i)) \
+            % (T_pool - T - 1)
         x = pool[g.unsqueeze(1) + torch.arange(T, device=device)]  # (B,T,N,N,1)
         y = pool[(g + T).unsqueeze(1) + torch.arange(1, device=device)]
         key = (g + T) % 7
@@ -153,8 +156,7 @@ def main():
             i_buf.fill_(i)
             hip_graph.replay()
             return graph_loss
-        i_buf.fill_(i)
-        return step_body()
+        return step_body(i)
 
     def barrier_sync():
         if ctx.enabled:

@@ -90,10 +90,17 @@ __launch_bounds__(256) __global__ void slab_colsum_kernel(
     const float* __restrict__ ws, float* __restrict__ out, long nb, long E) {
     const long e = (long)blockIdx.x * 256 + threadIdx.x;
     if (e >= E) return;
-    float s = 0.f;
+    // 8 independent accumulator chains (fixed combine order, so still
+    // bitwise-reproducible); a single serial chain was latency-bound at
+    // nb=512 rows for the narrow E=128 slabs
+    float s[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+    long b = 0;
     long off = e;
-    for (long b = 0; b < nb; ++b, off += E) s += ws[off];
-    out[e] = s;
+    for (; b + 8 <= nb; b += 8, off += 8 * E)
+#pragma unroll
+        for (int j = 0; j < 8; ++j) s[j] += ws[off + j * E];
+    for (int j = 0; b < nb; ++b, ++j, off += E) s[j] += ws[off];
+    out[e] = ((s[0] + s[1]) + (s[2] + s[3])) + ((s[4] + s[5]) + (s[6] + s[7]));
 }
 
 extern "C" void slab_colsum_launch(const float* ws, float* out, long nb, long E,
