@@ -10,7 +10,6 @@ from __future__ import annotations
 
 import importlib.machinery
 import importlib.util
-import os
 
 _ext = None
 _ext_err: Exception | None = None

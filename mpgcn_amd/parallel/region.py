@@ -31,7 +31,6 @@ import torch.distributed as dist
 
 from mpgcn_amd.ops import (
     GraphOperator,
-    eager,
     fused_lstm_last,
     linear_act,
     mode1_proj,
