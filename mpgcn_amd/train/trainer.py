@@ -174,7 +174,6 @@ class ModelTrainer:
         log("\n", datetime.now().strftime("%Y/%m/%d %H:%M:%S"))
         log(f'     {self.params["model"]} model training begins:')
         for epoch in range(start_epoch, 1 + self.params["num_epochs"]):
-            running_loss = {mode: 0.0 for mode in modes}
             epoch_samples = 0
             t0 = time.time()
             for mode in modes:
@@ -182,6 +181,10 @@ class ModelTrainer:
                 if hasattr(data_loader[mode], "set_epoch"):
                     data_loader[mode].set_epoch(epoch)
                 step = 0
+                # device-side accumulator: a per-step loss.item() would force
+                # a host sync every step (the reference's running-loss pattern,
+                # Model_Trainer.py:117, keeps the host from running ahead)
+                loss_sum = None
                 for x_seq, y_true, O_dyn_G, D_dyn_G in data_loader[mode]:
                     with torch.set_grad_enabled(mode == "train"):
                         y_pred, y_tgt = self._step_forward(
@@ -197,7 +200,8 @@ class ModelTrainer:
                             with trace_range("optimizer"):
                                 self.optimizer.step()
                     bs = y_true.shape[0]
-                    running_loss[mode] += loss.item() * bs
+                    contrib = loss.detach() * bs
+                    loss_sum = contrib if loss_sum is None else loss_sum + contrib
                     step += bs
                     if mode == "train":
                         epoch_samples += bs
@@ -207,14 +211,15 @@ class ModelTrainer:
                     if self.ctx.enabled:
                         # average validation loss across ranks for a consistent
                         # early-stopping decision
-                        t = torch.tensor(
-                            [running_loss[mode], float(step)], device=self.device
-                            if self.device.type == "cuda" else "cpu"
+                        t = torch.stack(
+                            [loss_sum.float(),
+                             torch.tensor(float(step), device=loss_sum.device)]
                         )
                         torch.distributed.all_reduce(t)
                         epoch_val_loss = (t[0] / t[1]).item()
                     else:
-                        epoch_val_loss = running_loss[mode] / max(step, 1)
+                        epoch_val_loss = (loss_sum.item() if loss_sum is not None
+                                          else 0.0) / max(step, 1)
                     dt = time.time() - t0
                     sps = epoch_samples * self.ctx.world_size / max(dt, 1e-9)
                     if epoch_val_loss <= val_loss:
@@ -264,9 +269,10 @@ class ModelTrainer:
         self.model.load_state_dict(ckpt["state_dict"])
         self.model.eval()
 
+        log = print if self.ctx.is_main else (lambda *a, **k: None)
         for mode in modes:
-            print("\n", datetime.now().strftime("%Y/%m/%d %H:%M:%S"))
-            print(f'     {self.params["model"]} model testing on {mode} data begins:')
+            log("\n", datetime.now().strftime("%Y/%m/%d %H:%M:%S"))
+            log(f'     {self.params["model"]} model testing on {mode} data begins:')
             forecast, ground_truth = [], []
             for x_seq, y_true, O_dyn_G, D_dyn_G in data_loader[mode]:
                 dyn = (
@@ -296,5 +302,5 @@ class ModelTrainer:
                         % (mode, MSE, RMSE, MAE, MAPE)
                     )
 
-        print("\n", datetime.now().strftime("%Y/%m/%d %H:%M:%S"))
-        print(f'     {self.params["model"]} model testing ends.')
+        log("\n", datetime.now().strftime("%Y/%m/%d %H:%M:%S"))
+        log(f'     {self.params["model"]} model testing ends.')
