@@ -116,7 +116,9 @@ class MPGCN(nn.Module):
                 if cached is None or cached[0] is not G:
                     Gc = G.to(self.compute_dtype)
                     cached = (G, GraphOperator(Gc, Gc))
-                    self._gop_cache = {key: cached}  # keep one static entry
+                    if len(self._gop_cache) >= 8:  # bound: one per static graph
+                        self._gop_cache.clear()
+                    self._gop_cache[key] = cached
                 gops.append(cached[1])
             else:
                 Go, Gd = G
