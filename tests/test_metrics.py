@@ -36,3 +36,24 @@ def test_torch_matches_numpy():
     assert np.isclose(mae, M.MAE(p, t), rtol=1e-5)
     assert np.isclose(mape, M.MAPE(p, t), rtol=1e-5)
     assert np.isclose(pcc, M.PCC(p, t), rtol=1e-4)
+
+
+def test_evaluate_torch_matches_numpy():
+    """Device-side metric variants agree with the reference-contract numpy
+    definitions (Metrics.py:5-26) on random data."""
+    import numpy as np
+    import torch
+
+    from mpgcn_amd.train import metrics as m
+
+    rng = np.random.default_rng(0)
+    yp = rng.random((50, 7, 9, 9, 1)).astype(np.float32) * 3
+    yt = rng.random((50, 7, 9, 9, 1)).astype(np.float32) * 3
+    mse, rmse, mae, mape = m.evaluate(yp, yt)
+    tmse, trmse, tmae, tmape, tpcc = m.evaluate_torch(
+        torch.from_numpy(yp), torch.from_numpy(yt))
+    assert abs(mse - tmse) < 1e-5
+    assert abs(rmse - trmse) < 1e-5
+    assert abs(mae - tmae) < 1e-6
+    assert abs(mape - tmape) < 1e-6
+    assert abs(m.PCC(yp, yt) - tpcc) < 1e-5
