@@ -174,6 +174,19 @@ __launch_bounds__(WVM * WVN * 64) __global__ void axis_gemm_kernel(AxisGemmParam
     }
 }
 
+// fp8 probe path (measurement only, docs/ROADMAP.md): same engine at
+// BK=128 — equal LDS footprint to the bf16 BK=64 tile but half the staged
+// bytes per K element, i.e. half the stage count at the same per-stage cost.
+extern "C" void axis_gemm_fp8_launch(AxisGemmParams p, int instances,
+                                     hipStream_t stream) {
+    constexpr int BN = 256, BM = 256;
+    const int tiles_m = (p.M + BM - 1) / BM;
+    p.tiles_l = (p.L + BN - 1) / BN;
+    dim3 grid(tiles_m * p.tiles_l, instances);
+    axis_gemm_kernel<unsigned char, 256, 256, 128, 4, 2, 1>
+        <<<grid, dim3(512), 0, stream>>>(p);
+}
+
 extern "C" void axis_gemm_launch(AxisGemmParams p, int instances, int is_f32,
                                  hipStream_t stream) {
     constexpr int BK = 64;

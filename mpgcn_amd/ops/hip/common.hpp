@@ -44,12 +44,32 @@ struct MfmaTraits<float> {
     }
 };
 
+// fp8 (OCP e4m3, gfx950 native): same 16x16x32 shape as bf16 — the MFMA rate
+// is identical, but operands are 1 byte, so a K-stage of equal LDS footprint
+// covers 2x the K depth (the probe instantiates BK=128). Raw storage type is
+// unsigned char; torch::kFloat8_e4m3fn bytes are bit-compatible.
+template <>
+struct MfmaTraits<unsigned char> {
+    static constexpr int MFMA_K = 32;
+    static constexpr int FRAG_ELEMS = 8;   // 8 fp8 bytes = one i64 operand
+    static constexpr int LDS_PAD = 16;     // 16 B
+    using frag_t = long;
+    __device__ static inline f32x4 mfma(frag_t a, frag_t b, f32x4 c) {
+        return __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a, b, c, 0, 0, 0);
+    }
+};
+
 template <typename T>
 __device__ __forceinline__ T from_f32(float v);
 template <>
 __device__ __forceinline__ __bf16 from_f32<__bf16>(float v) { return (__bf16)v; }
 template <>
 __device__ __forceinline__ float from_f32<float>(float v) { return v; }
+template <>
+__device__ __forceinline__ unsigned char from_f32<unsigned char>(float v) {
+    // v_cvt_pk_fp8_f32 (OCP e4m3 on gfx950); low byte of the packed pair
+    return (unsigned char)(__builtin_amdgcn_cvt_pk_fp8_f32(v, 0.f, 0, false) & 0xff);
+}
 
 __device__ __forceinline__ float to_f32(__bf16 v) { return (float)v; }
 __device__ __forceinline__ float to_f32(float v) { return v; }

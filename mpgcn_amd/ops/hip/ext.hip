@@ -390,6 +390,44 @@ std::vector<torch::Tensor> lstm_fused_bwd(torch::Tensor x, long T_logical,
     return {dwhh, dbias, dwih, dx};
 }
 
+// fp8 probe of the mode-2 contraction (docs/ROADMAP.md byte-reduction
+// lever; measurement-only — the training path stays bf16). V8/A2T8 are
+// torch float8_e4m3fn (bit-compatible with gfx950's OCP e4m3); returns a
+// float8_e4m3fn Y of the usual (B, Nm, N, H) mode-2 shape.
+torch::Tensor bdgcn_mode2_fp8(torch::Tensor V8, torch::Tensor A2T8,
+                              c10::optional<torch::Tensor> bias, bool relu,
+                              long N, long S) {
+    TORCH_CHECK(V8.is_cuda() && V8.is_contiguous() &&
+                V8.scalar_type() == torch::kFloat8_e4m3fn, "V8 must be fp8");
+    TORCH_CHECK(A2T8.is_cuda() && A2T8.is_contiguous() &&
+                A2T8.scalar_type() == torch::kFloat8_e4m3fn, "A2T8 must be fp8");
+    const long B = V8.size(0), Nm = V8.size(1), H = V8.size(-1);
+    TORCH_CHECK(A2T8.size(-1) == N * S && A2T8.size(-2) == N, "A2T8 shape");
+    TORCH_CHECK(N % 256 == 0 && (N * S) % 128 == 0 && (Nm * H) % 256 == 0,
+                "fp8 probe requires full tiles");
+    auto Y = torch::empty({B, Nm, N, H}, V8.options());
+
+    AxisGemmParams p{};
+    p.AT = A2T8.data_ptr();
+    p.X = V8.data_ptr();
+    p.OUT = Y.data_ptr();
+    p.bias = bias_ptr(bias);
+    p.M = (int)N; p.K = (int)(N * S); p.L = (int)(Nm * H);
+    p.a_div = 1; p.a_bs1 = 0; p.a_bs2 = 0;
+    p.x_div = 1; p.x_bs1 = Nm * N * S * H; p.x_bs2 = 0;
+    p.o_div = 1; p.o_bs1 = Nm * N * H; p.o_bs2 = 0;
+    p.kdiv = 1; p.k_lo = H;
+    p.qdiv = (int)H; p.q_hi = N * S * H;
+    p.o_row = H;
+    p.ogdiv = (int)H; p.og_hi = N * H;
+    p.relu = relu ? 1 : 0;
+    p.bias_mod = (int)H;
+    p.a_vec = ((N * S) % 16 == 0);
+    p.x_vec = (H % 16 == 0);
+    axis_gemm_fp8_launch(p, (int)B, stream());
+    return Y;
+}
+
 // Fused ReLU backward + bias-grad column sum: dY = dH * 1[Y>0], dbias=colsum.
 std::vector<torch::Tensor> relu_bwd_colsum(torch::Tensor dH, torch::Tensor Y,
                                            bool mask) {
@@ -430,6 +468,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("row_gemm", &row_gemm, "fused row GEMM + bias + act (K3/K6)");
     m.def("red_gemm", &red_gemm, "fused reduction GEMM X^T@Y + colsum + xdot");
     m.def("relu_bwd_colsum", &relu_bwd_colsum, "fused ReLU bwd mask + bias colsum");
+    m.def("bdgcn_mode2_fp8", &bdgcn_mode2_fp8, "fp8 e4m3 mode-2 probe (measurement only)");
     m.def("row_gemm_out", &row_gemm_out, "row GEMM into strided output slice");
     m.def("lstm_step_fwd", &lstm_step_fwd, "fused LSTM cell forward step (K4)");
     m.def("lstm_fused_fwd", &lstm_fused_fwd, "register-resident fused LSTM forward");

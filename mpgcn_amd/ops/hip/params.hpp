@@ -107,6 +107,7 @@ struct ReluBwdParams {
 extern "C" {
 void slab_colsum_launch(const float* ws, float* out, long nb, long E, hipStream_t s);
 void axis_gemm_launch(AxisGemmParams p, int instances, int is_f32, hipStream_t s);
+void axis_gemm_fp8_launch(AxisGemmParams p, int instances, hipStream_t s);
 void red_gemm_launch(RedGemmParams p, int is_f32, hipStream_t s);
 void row_gemm_launch(RowGemmParams p, int is_f32, hipStream_t s);
 void lstm_step_fwd_launch(LstmStepParams p, int is_f32, hipStream_t s);
