@@ -31,7 +31,8 @@ __device__ __forceinline__ long ocol_off(const AxisGemmParams& p, int q) {
     return p.ogdiv ? (long)(q / p.ogdiv) * p.og_hi + (long)(q % p.ogdiv) : (long)q;
 }
 
-template <typename T, int BM, int BN, int BK, int WVM, int WVN, int BUFS = 2>
+template <typename T, int BM, int BN, int BK, int WVM, int WVN, int BUFS = 2,
+          bool VEC_ONLY = false>
 __launch_bounds__(WVM * WVN * 64) __global__ void axis_gemm_kernel(AxisGemmParams p) {
     using MT = MfmaTraits<T>;
     constexpr int CH = 16 / sizeof(T);
@@ -101,9 +102,12 @@ __launch_bounds__(WVM * WVN * 64) __global__ void axis_gemm_kernel(AxisGemmParam
             for (int i = 0; i < CH; ++i) tmp[i] = (T)0.f;
             if (k < p.K) {
                 const long rbase = x_base + xrow_off(p, k);
-                if (qvec) {
+                if (VEC_ONLY || qvec) {
+                    // VEC_ONLY instantiations (fp8 probe) compile the vector
+                    // path alone: the CH-wide scalar fallback's address set
+                    // spills at CH = 16 bytes/lane
                     *(Chunk16*)tmp = *(const Chunk16*)(X + rbase + xq);
-                } else {
+                } else if constexpr (!VEC_ONLY) {
                     for (int i = 0; i < CH; ++i)
                         if (q0 + i < p.L) tmp[i] = X[rbase + xcol_off(p, q0 + i)];
                 }
@@ -183,7 +187,7 @@ extern "C" void axis_gemm_fp8_launch(AxisGemmParams p, int instances,
     const int tiles_m = (p.M + BM - 1) / BM;
     p.tiles_l = (p.L + BN - 1) / BN;
     dim3 grid(tiles_m * p.tiles_l, instances);
-    axis_gemm_kernel<unsigned char, 256, 256, 128, 4, 2, 1>
+    axis_gemm_kernel<unsigned char, 256, 256, 128, 4, 2, 1, true>
         <<<grid, dim3(512), 0, stream>>>(p);
 }
 
