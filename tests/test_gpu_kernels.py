@@ -493,3 +493,19 @@ def test_capacity_4096_regions_train_step():
     assert losses[-1] < losses[0], losses
     peak = torch.cuda.max_memory_allocated() / 2**30
     assert peak < 200, f"peak {peak:.1f} GiB"
+
+
+def test_fp8_mode2_probe_numerics():
+    """fp8 e4m3 probe path (measurement-only, docs/ROADMAP.md): same axis
+    engine at BK=128 on OCP fp8 operands; numerics within e4m3 quantization
+    error of the fp32 reference."""
+    ext = _ext()
+    torch.manual_seed(5)
+    B, Nn, S, Hd = 4, 256, 3, 32
+    V32 = torch.randn(B, Nn, Nn * S, Hd, device=DEV) / (Nn * S) ** 0.25
+    A32 = torch.randn(Nn, Nn * S, device=DEV) / (Nn * S) ** 0.25
+    ref = torch.einsum("dk,bmkh->bmdh", A32, V32)
+    y8 = ext.bdgcn_mode2_fp8(V32.to(torch.float8_e4m3fn),
+                             A32.to(torch.float8_e4m3fn), None, False, Nn, S)
+    rel = (y8.float() - ref).norm() / ref.norm()
+    assert rel < 8e-2, rel.item()
