@@ -509,3 +509,36 @@ def test_fp8_mode2_probe_numerics():
                              A32.to(torch.float8_e4m3fn), None, False, Nn, S)
     rel = (y8.float() - ref).norm() / ref.norm()
     assert rel < 8e-2, rel.item()
+
+
+@pytest.mark.timeout(300)
+def test_rectangular_sharded_halves_1024_p8():
+    """Region partition at the config #5 aspect ratio (8-way shards of a
+    large grid): N=1024, P=8 — Nl=128-wide rectangular extents through the
+    HIP kernels, forward vs the fp32 eager layer."""
+    from mpgcn_amd.ops import GraphOperator, eager, mode1_proj, mode2_bias_act
+
+    torch.manual_seed(13)
+    Nn, S, C, Hd, B, P = 1024, 3, 32, 32, 1, 8
+    Nl = Nn // P
+    dt = torch.bfloat16
+    X32 = torch.randn(B, Nn, Nn, C, device=DEV)
+    Go = torch.randn(S, Nn, Nn, device=DEV) / Nn**0.5
+    Gd = torch.randn(S, Nn, Nn, device=DEV) / Nn**0.5
+    W32 = torch.randn(C * S * S, Hd, device=DEV) / (C * S * S) ** 0.5
+    b32 = torch.randn(Hd, device=DEV)
+    with torch.no_grad():
+        ref = eager.bdgcn_layer_eager(X32, Go, Gd, W32, b32, "relu")
+        gop = GraphOperator(Go.to(dt), Gd.to(dt))
+        X, W = X32.to(dt), W32.to(dt)
+        Vs = [mode1_proj(X[:, :, p * Nl:(p + 1) * Nl, :].contiguous(), W, gop)
+              for p in range(P)]
+        Vfull = torch.cat(Vs, dim=2)
+        Ys = [mode2_bias_act(
+                  Vfull[:, p * Nl:(p + 1) * Nl].reshape(B, Nl, Nn, S, Hd)
+                  .contiguous(), b32, gop, True)
+              for p in range(P)]
+        out = torch.cat(Ys, dim=1)
+    rel = (out.float() - ref).norm() / ref.norm()
+    assert rel < 3e-2, rel.item()
+    torch.cuda.synchronize()
