@@ -1,6 +1,7 @@
 // Fused elementwise + column-sum kernels (gfx950).
 //
 // relu_bwd_colsum: one pass producing both pieces the BDGCN backward needs
+// (the backward of the reference's relu + bias at MPGCN.py:47-49)
 // from the upstream gradient:   dY = dH * 1[Y > 0]   and   dbias = colsum(dY).
 // Replaces a torch elementwise (3 tensor passes) plus a slow non-contiguous
 // torch reduction (together ~350 us per layer at the flagship config) with a

@@ -35,6 +35,8 @@ const float* bias_ptr(const c10::optional<torch::Tensor>& b) {
 // U[b,m,d,o,l] = sum_n GT[(b,)o,m,n] X[b,n,d,l].
 // X: (B, N, N, C); GT: (S, N, N) static or (B, S, N, N) dynamic, ALREADY
 // transposed per support (GT[..., m, n] = G[..., n, m]). Out: (B, N, N, S, C).
+// Mode-1 (origin-axis) product: the reference's einsum('bncl,nm->bmcl') /
+// ('bncl,bnm->bmcl') pair (reference MPGCN.py:30,38), batched over supports.
 // Rectangular: X may be destination-sharded, (B, No, Nd, C) with Nd != No
 // (the region-partition path, mpgcn_amd/parallel/region.py).
 torch::Tensor bdgcn_mode1(torch::Tensor X, torch::Tensor GT) {
