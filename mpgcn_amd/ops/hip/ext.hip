@@ -430,6 +430,36 @@ torch::Tensor bdgcn_mode2_fp8(torch::Tensor V8, torch::Tensor A2T8,
     return Y;
 }
 
+// fp8 probe of the mode-1 contraction (same measurement-only contract as
+// bdgcn_mode2_fp8; square static graphs, full tiles).
+torch::Tensor bdgcn_mode1_fp8(torch::Tensor X8, torch::Tensor GT8) {
+    TORCH_CHECK(X8.is_cuda() && X8.is_contiguous() &&
+                X8.scalar_type() == torch::kFloat8_e4m3fn, "X8 must be fp8");
+    TORCH_CHECK(GT8.is_cuda() && GT8.is_contiguous() &&
+                GT8.scalar_type() == torch::kFloat8_e4m3fn, "GT8 must be fp8");
+    const long B = X8.size(0), N = X8.size(1), C = X8.size(3);
+    const long S = GT8.size(0);
+    TORCH_CHECK(GT8.dim() == 3 && GT8.size(1) == N && GT8.size(2) == N, "GT8 shape");
+    TORCH_CHECK(N % 128 == 0 && (N * C) % 256 == 0, "fp8 probe requires full tiles");
+    auto U = torch::empty({B, N, N, S, C}, X8.options());
+    AxisGemmParams p{};
+    p.AT = GT8.data_ptr();
+    p.X = X8.data_ptr();
+    p.OUT = U.data_ptr();
+    p.M = (int)N; p.K = (int)N; p.L = (int)(N * C);
+    p.a_div = (int)S; p.a_bs1 = 0; p.a_bs2 = N * N;
+    p.x_div = (int)S; p.x_bs1 = N * N * C; p.x_bs2 = 0;
+    p.o_div = (int)S; p.o_bs1 = N * N * S * C; p.o_bs2 = C;
+    p.kdiv = 1; p.k_lo = N * C;
+    p.qdiv = 0;
+    p.o_row = N * S * C;
+    p.ogdiv = (int)C; p.og_hi = S * C;
+    p.a_vec = (N % 16 == 0);
+    p.x_vec = ((N * C) % 16 == 0) && (C % 16 == 0);
+    axis_gemm_fp8_launch(p, (int)(B * S), stream());
+    return U;
+}
+
 // Fused ReLU backward + bias-grad column sum: dY = dH * 1[Y>0], dbias=colsum.
 std::vector<torch::Tensor> relu_bwd_colsum(torch::Tensor dH, torch::Tensor Y,
                                            bool mask) {
@@ -471,6 +501,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("red_gemm", &red_gemm, "fused reduction GEMM X^T@Y + colsum + xdot");
     m.def("relu_bwd_colsum", &relu_bwd_colsum, "fused ReLU bwd mask + bias colsum");
     m.def("bdgcn_mode2_fp8", &bdgcn_mode2_fp8, "fp8 e4m3 mode-2 probe (measurement only)");
+    m.def("bdgcn_mode1_fp8", &bdgcn_mode1_fp8, "fp8 e4m3 mode-1 probe (measurement only)");
     m.def("row_gemm_out", &row_gemm_out, "row GEMM into strided output slice");
     m.def("lstm_step_fwd", &lstm_step_fwd, "fused LSTM cell forward step (K4)");
     m.def("lstm_fused_fwd", &lstm_fused_fwd, "register-resident fused LSTM forward");
