@@ -542,3 +542,18 @@ def test_rectangular_sharded_halves_1024_p8():
     rel = (out.float() - ref).norm() / ref.norm()
     assert rel < 3e-2, rel.item()
     torch.cuda.synchronize()
+
+
+def test_fp8_mode1_probe_numerics():
+    """fp8 e4m3 mode-1 probe (short-K shape; measurement-only path)."""
+    ext = _ext()
+    torch.manual_seed(6)
+    B, Nn, S, C = 4, 256, 3, 32
+    X32 = torch.randn(B, Nn, Nn, C, device=DEV) / Nn**0.25
+    G32 = torch.randn(S, Nn, Nn, device=DEV) / Nn**0.25
+    ref = torch.einsum("onm,bndl->bmdol", G32, X32)
+    GT = G32.transpose(-2, -1).contiguous()
+    u8 = ext.bdgcn_mode1_fp8(X32.to(torch.float8_e4m3fn),
+                             GT.to(torch.float8_e4m3fn))
+    rel = (u8.float() - ref).norm() / ref.norm()
+    assert rel < 8e-2, rel.item()
