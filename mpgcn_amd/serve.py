@@ -1,0 +1,139 @@
+"""Inference serving: load a reference-format checkpoint, forecast over HTTP.
+
+The reference stops at the offline test loop (Model_Trainer.py:145-185); this
+adds the deployment surface a production user needs: a FastAPI app that holds
+the model and the day-of-week correlation graphs resident on one GPU and
+answers forecast requests with the same autoregressive rollout protocol as
+trainer.test() (dynamic graphs held at the first step's day-of-week, kept for
+parity with the reference's scoring path).
+
+    uvicorn "mpgcn_amd.serve:create_app" --factory ...
+    # or: python -m mpgcn_amd.serve -ckpt out/MPGCN_od.pkl -synthetic-nodes 64
+
+POST /predict  {"x_seq": [[..]], "dow": 3, "horizon": 7}
+    x_seq: (T_obs, N, N) nested lists (or with trailing singleton channel);
+    dow:   day-of-week index of the first forecast step (graph key, 0-6);
+    horizon: number of autoregressive steps (default 1).
+    -> {"forecast": (horizon, N, N) nested lists}
+GET  /healthz -> {"status": "ok", "regions": N, "device": "..."}
+"""
+
+from __future__ import annotations
+
+import argparse
+
+import torch
+
+from mpgcn_amd.graph import build_supports, get_support_K
+from mpgcn_amd.models import MPGCN
+
+
+class Forecaster:
+    """Model + resident graphs; the serving core (framework API, no HTTP)."""
+
+    def __init__(self, params: dict, data: dict):
+        self.params = params
+        self.device = torch.device(params.get("device", "cpu"))
+        cd = str(params.get("compute_dtype", "float32"))
+        compute_dtype = (torch.bfloat16 if cd in ("bf16", "bfloat16")
+                        else torch.float32)
+        K = get_support_K(params["kernel_type"], params["cheby_order"])
+        N = data["adj"].shape[-1]
+        self.N = N
+        self.model = MPGCN(
+            M=int(params.get("perspectives", 2)), K=K, input_dim=1,
+            lstm_hidden_dim=params["hidden_dim"], lstm_num_layers=1,
+            gcn_hidden_dim=params["hidden_dim"], gcn_num_layers=3,
+            num_nodes=N, compute_dtype=compute_dtype,
+            fusion=params.get("fusion", "mean"),
+        ).to(self.device)
+        ckpt = torch.load(params["checkpoint"], map_location=self.device,
+                          weights_only=False)
+        self.model.load_state_dict(ckpt["state_dict"])
+        self.model.eval()
+
+        adj = data["adj"].float().to(self.device)
+        self.G_static = build_supports(
+            adj.unsqueeze(0), params["kernel_type"], params["cheby_order"]
+        ).squeeze(0)
+        # (N, N, 7) -> per-dow support stacks, built once at startup
+        O_dyn = data["O_dyn_G"].permute(2, 0, 1).float().to(self.device)
+        D_dyn = data["D_dyn_G"].permute(2, 0, 1).float().to(self.device)
+        self.G_o = build_supports(O_dyn, params["kernel_type"], params["cheby_order"])
+        self.G_d = build_supports(D_dyn, params["kernel_type"], params["cheby_order"])
+
+    @torch.no_grad()
+    def forecast(self, x_seq: torch.Tensor, dow: int, horizon: int = 1) -> torch.Tensor:
+        """x_seq: (T, N, N) or (T, N, N, 1) -> (horizon, N, N)."""
+        if x_seq.dim() == 3:
+            x_seq = x_seq.unsqueeze(-1)
+        cur = x_seq.unsqueeze(0).float().to(self.device)  # (1, T, N, N, 1)
+        g_list = [self.G_static,
+                  (self.G_o[dow % 7:dow % 7 + 1], self.G_d[dow % 7:dow % 7 + 1])]
+        if int(self.params.get("perspectives", 2)) == 3:
+            g_list.append(self.G_static)
+        preds = []
+        for _ in range(horizon):
+            step = self.model(x_seq=cur, G_list=g_list)  # (1, 1, N, N, 1)
+            cur = torch.cat([cur[:, 1:], step], dim=1)
+            preds.append(step)
+        return torch.cat(preds, dim=1)[0, :, :, :, 0].cpu()
+
+
+def create_app(params: dict | None = None, data: dict | None = None):
+    """FastAPI app factory; params/data as in ModelTrainer (tests inject both)."""
+    from fastapi import FastAPI, HTTPException
+
+    if params is None:
+        params, data = _params_from_cli()
+    fc = Forecaster(params, data)
+    app = FastAPI(title="mpgcn-amd forecast service")
+
+    @app.get("/healthz")
+    def healthz():
+        return {"status": "ok", "regions": fc.N, "device": str(fc.device)}
+
+    @app.post("/predict")
+    def predict(payload: dict):
+        try:
+            x = torch.tensor(payload["x_seq"], dtype=torch.float32)
+            dow = int(payload.get("dow", 0))
+            horizon = int(payload.get("horizon", 1))
+        except (KeyError, ValueError, TypeError) as e:
+            raise HTTPException(status_code=422, detail=str(e))
+        if x.dim() not in (3, 4) or x.shape[-2] != fc.N or horizon < 1:
+            raise HTTPException(status_code=422, detail="bad x_seq shape/horizon")
+        out = fc.forecast(x, dow, horizon)
+        return {"forecast": out.tolist()}
+
+    return app
+
+
+def _params_from_cli():
+    from mpgcn_amd.data import DataInput
+
+    ap = argparse.ArgumentParser()
+    ap.add_argument("-ckpt", "--checkpoint", required=True)
+    ap.add_argument("-in", "--input_dir", default="../data")
+    ap.add_argument("-synthetic-nodes", "--synthetic_nodes", type=int, default=0)
+    ap.add_argument("-synthetic-days", "--synthetic_days", type=int, default=425)
+    ap.add_argument("-device", "--device", default="cuda:0")
+    ap.add_argument("-hidden", "--hidden_dim", type=int, default=32)
+    ap.add_argument("-kernel", "--kernel_type", default="random_walk_diffusion")
+    ap.add_argument("-K", "--cheby_order", type=int, default=2)
+    ap.add_argument("-dtype", "--compute_dtype", default="bf16")
+    ap.add_argument("-M", "--perspectives", type=int, default=2)
+    ap.add_argument("-split", "--split_ratio", type=float, nargs="+",
+                    default=[7, 1.5, 1.5])
+    ap.add_argument("-port", "--port", type=int, default=8321)
+    args = ap.parse_args()
+    params = args.__dict__
+    data = DataInput(params=params).load_data()
+    return params, data
+
+
+if __name__ == "__main__":
+    import uvicorn
+
+    params, data = _params_from_cli()
+    uvicorn.run(create_app(params, data), host="127.0.0.1", port=params["port"])

@@ -1,0 +1,54 @@
+"""Serving layer: checkpoint load + HTTP forecast contract (CPU, TestClient)."""
+
+import pytest
+import torch
+
+from mpgcn_amd.data import DataInput
+from mpgcn_amd.models import MPGCN
+
+fastapi = pytest.importorskip("fastapi")
+from fastapi.testclient import TestClient  # noqa: E402
+
+
+@pytest.fixture(scope="module")
+def app_and_n(tmp_path_factory):
+    out = tmp_path_factory.mktemp("serve")
+    params = {
+        "synthetic_nodes": 12, "synthetic_days": 60, "seed": 0,
+        "split_ratio": [7, 1.5, 1.5], "norm": "none",
+        "hidden_dim": 16, "kernel_type": "random_walk_diffusion",
+        "cheby_order": 2, "device": "cpu", "compute_dtype": "float32",
+        "checkpoint": str(out / "MPGCN_od.pkl"),
+    }
+    data = DataInput(params=params).load_data()
+    model = MPGCN(M=2, K=3, input_dim=1, lstm_hidden_dim=16, lstm_num_layers=1,
+                  gcn_hidden_dim=16, gcn_num_layers=3, num_nodes=12)
+    torch.save({"epoch": 1, "state_dict": model.state_dict()},
+               params["checkpoint"])
+    from mpgcn_amd.serve import create_app
+
+    return create_app(params, data), 12
+
+
+def test_healthz(app_and_n):
+    app, N = app_and_n
+    r = TestClient(app).get("/healthz")
+    assert r.status_code == 200
+    assert r.json()["regions"] == N
+
+
+def test_predict_contract(app_and_n):
+    app, N = app_and_n
+    x = torch.rand(7, N, N).tolist()
+    r = TestClient(app).post("/predict",
+                             json={"x_seq": x, "dow": 2, "horizon": 3})
+    assert r.status_code == 200, r.text
+    out = torch.tensor(r.json()["forecast"])
+    assert out.shape == (3, N, N)
+    assert torch.isfinite(out).all()
+
+
+def test_predict_rejects_bad_shape(app_and_n):
+    app, _ = app_and_n
+    r = TestClient(app).post("/predict", json={"x_seq": [[1.0, 2.0]]})
+    assert r.status_code == 422
