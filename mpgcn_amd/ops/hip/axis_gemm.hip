@@ -10,9 +10,12 @@
 //   bwd dV     : dV[b,m,cs,h]  = sum_d  A2[cs,d]    dY[b,m,d,h]
 //   bwd dX     : dX[b,n,d,l]   = sum_om A3T[n,om]   dU[b,m,d,o,l]
 //
-// Structure: double-buffered LDS tiles with next-tile staging issued BEFORE
-// the current tile's MFMAs and ONE barrier per K-tile (the 2-phase schedule of
-// the CDNA GEMM playbook). The graph operand AT is (M, K) row-major (the
+// Structure: LDS-tiled K loop. bf16 instantiations run SINGLE-buffered —
+// these contractions are global-latency-bound (PMC: 67% WAIT_ANY), so the
+// halved LDS footprint (2x resident blocks/CU) beats the double-buffered
+// 2-phase pipeline (A/B-measured; profiles/SUMMARY.md). The f32 test-oracle
+// path keeps the classic double-buffered prefetch schedule.
+// The graph operand AT is (M, K) row-major (the
 // Python layer pre-permutes the tiny graph tensors), so the A-tile stages as a
 // straight vectorized copy; the X-tile transposes into an [n][k]-major image
 // with lane-rotated element order (spreads write banks over 4*{0..7}). LDS
