@@ -13,8 +13,9 @@
 //             sequence), then walks t = T-1..0 computing the gate gradients,
 //             the dh chain (MFMA vs W_hh), and the dW_hh / dbias / dw_ih
 //             partials in-register; partials land in a per-block f32
-//             workspace reduced by one torch sum (no atomics, no dgates
-//             materialization).
+//             workspace reduced by the fixed-order slab_colsum kernel
+//             (elemwise.hip) — no atomics anywhere, so gradients are
+//             bitwise-reproducible; no dgates materialization.
 //
 // Weight-grad MFMAs pair two timesteps per K=32 contraction (16 rows each).
 // Per-wave LDS images: dg_img [2*16][4H] (dh A-operand, vector reads; dW
