@@ -557,3 +557,25 @@ def test_fp8_mode1_probe_numerics():
                              GT.to(torch.float8_e4m3fn))
     rel = (u8.float() - ref).norm() / ref.norm()
     assert rel < 8e-2, rel.item()
+
+
+def test_serving_forecaster_gpu():
+    """Serving path on GPU: Forecaster rollout through the HIP forward."""
+    from mpgcn_amd.data import DataInput
+    from mpgcn_amd.models import MPGCN
+    from mpgcn_amd.serve import Forecaster
+
+    params = {"synthetic_nodes": 64, "synthetic_days": 60, "seed": 0,
+              "split_ratio": [7, 1.5, 1.5], "norm": "none", "hidden_dim": 32,
+              "kernel_type": "random_walk_diffusion", "cheby_order": 2,
+              "device": DEV, "compute_dtype": "bf16",
+              "checkpoint": "/tmp/_serve_ck.pkl"}
+    data = DataInput(params=params).load_data()
+    model = MPGCN(M=2, K=3, input_dim=1, lstm_hidden_dim=32, lstm_num_layers=1,
+                  gcn_hidden_dim=32, gcn_num_layers=3, num_nodes=64)
+    torch.save({"epoch": 1, "state_dict": model.state_dict()},
+               params["checkpoint"])
+    fc = Forecaster(params, data)
+    out = fc.forecast(torch.rand(7, 64, 64), dow=3, horizon=4)
+    assert out.shape == (4, 64, 64)
+    assert torch.isfinite(out).all()

@@ -109,7 +109,7 @@ def test_main_cli_end_to_end(tmp_path):
     r = subprocess.run(
         [sys.executable, str(repo / "Main.py"), "-GPU", "cpu",
          "-synthetic-nodes", "16", "-synthetic-days", "60", "-epoch", "1",
-         "-out", str(out)],
+         "-norm", "minmax", "-out", str(out)],
         capture_output=True, text=True, timeout=240, cwd=str(repo),
     )
     assert r.returncode == 0, r.stderr[-2000:]
