@@ -135,6 +135,11 @@ def mpgcn_forward_sharded(model, x_seq_shard, G_list, group=None):
     Returns the destination-sharded prediction (B, 1, N, N/P, 1).
     """
     B, T, N, Nl, _ = x_seq_shard.shape
+    P = dist.get_world_size(group)
+    if N % P != 0:
+        raise ValueError(
+            f"region partition needs the region count ({N}) divisible by the "
+            f"world size ({P}); pad the grid or change the rank count")
     gops = model._graph_operators(G_list)
     cd = model.compute_dtype
     lstm_in = (

@@ -84,3 +84,38 @@ def test_training_is_seed_deterministic(tmp_path):
         states.append({k: v.clone() for k, v in trainer.model.state_dict().items()})
     for k in states[0]:
         assert torch.equal(states[0][k], states[1][k]), k
+
+
+@settings(max_examples=15, deadline=None)
+@given(
+    n=st.sampled_from([4, 6, 8, 12]),
+    p=st.sampled_from([1, 2, 4]),
+    c=st.integers(1, 6),
+    h=st.integers(2, 8),
+    s=st.integers(1, 3),
+    seed=st.integers(0, 2**16),
+)
+def test_region_halves_property(n, p, c, h, s, seed):
+    """mode1_proj | manual re-shard | mode2_bias_act over P rectangular shards
+    equals the fused eager layer for random shapes (the region-partition
+    algebra, mpgcn_amd/parallel/region.py)."""
+    if n % p != 0:
+        return
+    from mpgcn_amd.ops import GraphOperator, mode1_proj, mode2_bias_act
+
+    g = torch.Generator().manual_seed(seed)
+    nl = n // p
+    X = torch.randn(2, n, n, c, generator=g)
+    Go = torch.randn(s, n, n, generator=g)
+    Gd = torch.randn(s, n, n, generator=g)
+    W = torch.randn(c * s * s, h, generator=g)
+    bias = torch.randn(h, generator=g)
+    ref = eager.bdgcn_layer_eager(X, Go, Gd, W, bias, "relu")
+    gop = GraphOperator(Go, Gd)
+    V = torch.cat([mode1_proj(X[:, :, q * nl:(q + 1) * nl, :], W, gop)
+                   for q in range(p)], dim=2)
+    out = torch.cat([
+        mode2_bias_act(V[:, q * nl:(q + 1) * nl].reshape(2, nl, n, s, h),
+                       bias, gop, True)
+        for q in range(p)], dim=1)
+    assert torch.allclose(out, ref, atol=1e-4, rtol=1e-4)

@@ -189,3 +189,39 @@ def test_region_trainer_end_to_end(tmp_path):
         p_.join(timeout=240)
         assert p_.exitcode == 0
     assert os.path.exists(os.path.join(out_dir, "MPGCN_od.pkl"))
+
+
+def test_region_world_size_divisibility_error():
+    import torch.distributed as dist
+
+    from mpgcn_amd.parallel.region import mpgcn_forward_sharded
+
+    store = "/tmp/_pgdiv"
+    import os as _os
+    if _os.path.exists(store):
+        _os.remove(store)
+    dist.init_process_group("gloo", init_method=f"file://{store}",
+                            rank=0, world_size=1)
+    try:
+        model = _model()
+        x = torch.rand(1, T, 7, 7, 1)  # 7 regions, any P>... N=7 with P=1 ok;
+        # use a fake group world via monkeypatched N % P: P=1 divides, so
+        # instead check the error path directly
+        import pytest as _pytest
+
+        from mpgcn_amd.parallel import region as reg
+
+        class _FakeDist:
+            @staticmethod
+            def get_world_size(group=None):
+                return 2
+
+        orig = reg.dist
+        reg.dist = _FakeDist
+        try:
+            with _pytest.raises(ValueError, match="divisible"):
+                mpgcn_forward_sharded(model, x, None)
+        finally:
+            reg.dist = orig
+    finally:
+        dist.destroy_process_group()
