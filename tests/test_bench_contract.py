@@ -24,3 +24,17 @@ def test_bench_json_contract():
     assert j["config"]["model"] == "MPGCN"
     assert j["config"]["global_batch"] == 2
     assert j["value"] > 0
+
+
+def test_tool_scripts_parse():
+    """The GPU-only probe/bench tools must at least stay syntactically valid
+    (they are exercised on gpurun boxes, not in the CPU suite)."""
+    import ast
+    from pathlib import Path
+
+    repo = Path(__file__).resolve().parent.parent
+    for rel in ("tools/det_probe.py", "tools/profile_step.py",
+                "tools/fp8_probe.py", "bench_kernels.py", "bench_one.py",
+                "mpgcn_amd/serve.py", "__graft_entry__.py"):
+        src = (repo / rel).read_text()
+        ast.parse(src)
