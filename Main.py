@@ -122,10 +122,6 @@ def main():
     with rank_watchdog(ctx):
         if params["mode"] == "train":
             trainer.train(data_loader=data_loader, modes=["train", "validate"])
-            # the reference evaluates right after training with the same
-            # invocation as test mode (reference Main.py:66-67)
-            if ctx.rank == 0:
-                trainer.test(data_loader=data_loader, modes=["train", "test"])
         else:
             trainer.test(data_loader=data_loader, modes=["train", "test"])
 

@@ -124,7 +124,7 @@ def _params_from_cli():
     ap.add_argument("-dtype", "--compute_dtype", default="bf16")
     ap.add_argument("-M", "--perspectives", type=int, default=2)
     ap.add_argument("-split", "--split_ratio", type=float, nargs="+",
-                    default=[7, 1.5, 1.5])
+                    default=[6.4, 1.6, 2])
     ap.add_argument("-port", "--port", type=int, default=8321)
     args = ap.parse_args()
     params = args.__dict__

@@ -98,25 +98,29 @@ def test_loss_variants(tmp_path):
 
 @pytest.mark.timeout(300)
 def test_main_cli_end_to_end(tmp_path):
-    """Main.py contract (reference Main.py:7-67): train mode trains, writes
-    the checkpoint, then immediately evaluates and appends both score lines."""
+    """Main.py contract: the reference workflow is train mode (checkpoint)
+    followed by a separate test-mode invocation (scores file) — reference
+    Main.py:63-67 runs strictly one or the other."""
     import subprocess
     import sys
     from pathlib import Path
 
     repo = Path(__file__).resolve().parent.parent
     out = tmp_path / "cli_out"
-    r = subprocess.run(
-        [sys.executable, str(repo / "Main.py"), "-GPU", "cpu",
-         "-synthetic-nodes", "16", "-synthetic-days", "60", "-epoch", "1",
-         "-norm", "minmax", "-out", str(out)],
-        capture_output=True, text=True, timeout=240, cwd=str(repo),
-    )
+    common = [sys.executable, str(repo / "Main.py"), "-GPU", "cpu",
+              "-synthetic-nodes", "16", "-synthetic-days", "60",
+              "-norm", "minmax", "-out", str(out)]
+    r = subprocess.run(common + ["-epoch", "1"], capture_output=True,
+                       text=True, timeout=240, cwd=str(repo))
     assert r.returncode == 0, r.stderr[-2000:]
     assert (out / "MPGCN_od.pkl").exists()
-    scores = (out / "MPGCN_prediction_scores.txt").read_text().splitlines()
-    assert scores[0].startswith("train, MSE, RMSE, MAE, MAPE, ")
-    assert scores[1].startswith("test, MSE, RMSE, MAE, MAPE, ")
     ckpt = torch.load(out / "MPGCN_od.pkl", map_location="cpu",
                       weights_only=True)
     assert set(ckpt) == {"epoch", "state_dict"}
+    r = subprocess.run(common + ["-mode", "test", "-pred", "2"],
+                       capture_output=True, text=True, timeout=240,
+                       cwd=str(repo))
+    assert r.returncode == 0, r.stderr[-2000:]
+    scores = (out / "MPGCN_prediction_scores.txt").read_text().splitlines()
+    assert scores[0].startswith("train, MSE, RMSE, MAE, MAPE, ")
+    assert scores[1].startswith("test, MSE, RMSE, MAE, MAPE, ")
