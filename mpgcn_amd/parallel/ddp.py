@@ -106,6 +106,7 @@ class GradAllReducer:
         self._pending: list[tuple[torch.distributed.Work, list[torch.Tensor], torch.Tensor]] = []
         self._bucket: list[torch.Tensor] = []
         self._bucket_sz = 0
+        self._bucket_streams: set = set()  # streams that produced this bucket's grads
         self._hooks = []
         if not ctx.enabled:
             return
@@ -123,6 +124,12 @@ class GradAllReducer:
             return
         self._bucket.append(g)
         self._bucket_sz += g.numel() * g.element_size()
+        if g.is_cuda:
+            # autograd runs each AccumulateGrad on the stream its producing op
+            # was recorded on (branch overlap records branches 1..M-1 on side
+            # streams, models/mpgcn.py:155-180) — remember it so a mid-backward
+            # flush can synchronize against every producer before reducing
+            self._bucket_streams.add(torch.cuda.current_stream(g.device))
         if self._bucket_sz >= self.bucket_bytes:
             self._flush()
 
@@ -130,9 +137,27 @@ class GradAllReducer:
         if not self._bucket:
             return
         grads = self._bucket
+        streams = self._bucket_streams
         self._bucket = []
         self._bucket_sz = 0
+        self._bucket_streams = set()
+        if grads[0].is_cuda:
+            # A bucket can mix gradients produced on different streams (branch
+            # overlap). The flatten copies and the collective run on the
+            # CURRENT stream, so order the current stream after every producing
+            # stream first — without this, a bucket that fills mid-backward
+            # launches all_reduce racing the side-stream grad writes.
+            cur = torch.cuda.current_stream(grads[0].device)
+            for s in streams:
+                if s != cur:
+                    cur.wait_stream(s)
         flat = torch._utils._flatten_dense_tensors(grads)
+        if grads[0].is_cuda:
+            # the flat buffer is consumed (all_reduce + unflatten-copy) on the
+            # current stream; grads are later overwritten on it too — keep the
+            # side-stream allocator from reusing their blocks early
+            for g in grads:
+                g.record_stream(torch.cuda.current_stream(grads[0].device))
         work = dist.all_reduce(flat, op=dist.ReduceOp.SUM, async_op=True)
         self._pending.append((work, grads, flat))
 

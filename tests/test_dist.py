@@ -41,13 +41,13 @@ def _single_process_grads():
     return {n: p.grad.clone() for n, p in model.named_parameters()}
 
 
-def _rank_worker(rank, world, file_name, out_file):
+def _rank_worker(rank, world, file_name, out_file, bucket_bytes=16 << 20):
     os.environ.update(RANK=str(rank), WORLD_SIZE=str(world), LOCAL_RANK=str(rank))
     dist.init_process_group("gloo", init_method=f"file://{file_name}",
                             rank=rank, world_size=world)
     ctx = DistContext(rank=rank, world_size=world, local_rank=rank, backend="gloo")
     model = _make_model(seed=100 + rank)  # divergent init: broadcast must fix it
-    reducer = GradAllReducer(model, ctx)
+    reducer = GradAllReducer(model, ctx, bucket_bytes=bucket_bytes)
 
     x, y, Gs, Go, Gd = _make_inputs()
     half = B // world
@@ -85,6 +85,33 @@ def test_two_rank_dp_grads_match_full_batch(tmp_path):
     for n, p in model.named_parameters():
         # DP averages the two half-batch means; MSE over equal halves averages
         # to the full-batch mean, so grads must match to fp tolerance
+        assert torch.allclose(dp_grads[n], p.grad, atol=1e-5), n
+
+
+@pytest.mark.timeout(300)
+def test_two_rank_dp_small_buckets_mid_backward_flush(tmp_path):
+    """bucket_bytes small enough that every parameter fills a bucket -> every
+    all_reduce launches MID-backward from the post-accumulate-grad hook (the
+    tail-flush path never carries the reduction). Grad equivalence must still
+    hold — this exercises the flush ordering/synchronization path that the
+    default 16 MB bucket never reaches at MPGCN's ~300 KB gradient volume."""
+    file_name = str(tmp_path / "pg_init_sb")
+    out_file = str(tmp_path / "rank0_grads_sb.pt")
+    ctxm = mp.get_context("spawn")
+    procs = [ctxm.Process(target=_rank_worker, args=(r, 2, file_name, out_file, 64))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=240)
+        assert p.exitcode == 0
+    dp_grads = torch.load(out_file, weights_only=True)
+
+    model = _make_model(seed=100)
+    x, y, Gs, Go, Gd = _make_inputs()
+    out = model(x, [Gs, (Go, Gd)])
+    torch.nn.functional.mse_loss(out, y).backward()
+    for n, p in model.named_parameters():
         assert torch.allclose(dp_grads[n], p.grad, atol=1e-5), n
 
 
