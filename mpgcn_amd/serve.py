@@ -61,6 +61,15 @@ class Forecaster:
         D_dyn = data["D_dyn_G"].permute(2, 0, 1).float().to(self.device)
         self.G_o = build_supports(O_dyn, params["kernel_type"], params["cheby_order"])
         self.G_d = build_supports(D_dyn, params["kernel_type"], params["cheby_order"])
+        if int(params.get("perspectives", 2)) == 3:
+            # third perspective: supports from the day-of-week-AVERAGED OD
+            # correlation graph — the same construction the trainer uses
+            # (train/trainer.py _graph_list), so a 3-perspective checkpoint is
+            # served with the graphs it was trained on
+            corr = data["O_dyn_G"].float().mean(dim=-1).to(self.device)
+            self.G_corr = build_supports(
+                corr.unsqueeze(0), params["kernel_type"], params["cheby_order"]
+            ).squeeze(0)
 
     @torch.no_grad()
     def forecast(self, x_seq: torch.Tensor, dow: int, horizon: int = 1) -> torch.Tensor:
@@ -71,7 +80,7 @@ class Forecaster:
         g_list = [self.G_static,
                   (self.G_o[dow % 7:dow % 7 + 1], self.G_d[dow % 7:dow % 7 + 1])]
         if int(self.params.get("perspectives", 2)) == 3:
-            g_list.append(self.G_static)
+            g_list.append(self.G_corr)
         preds = []
         for _ in range(horizon):
             step = self.model(x_seq=cur, G_list=g_list)  # (1, 1, N, N, 1)

@@ -52,3 +52,45 @@ def test_predict_rejects_bad_shape(app_and_n):
     app, _ = app_and_n
     r = TestClient(app).post("/predict", json={"x_seq": [[1.0, 2.0]]})
     assert r.status_code == 422
+
+
+def test_serve_matches_trainer_at_three_perspectives(tmp_path):
+    """A 3-perspective checkpoint served through Forecaster must see the SAME
+    third-perspective graph the trainer trained it on: supports built from the
+    day-of-week-averaged OD-correlation graph (train/trainer.py _graph_list),
+    not the static adjacency supports."""
+    from mpgcn_amd.serve import Forecaster
+    from mpgcn_amd.train.trainer import ModelTrainer
+
+    N = 10
+    params = {
+        "model": "MPGCN", "synthetic_nodes": N, "synthetic_days": 60, "seed": 3,
+        "split_ratio": [7, 1.5, 1.5], "norm": "none", "N": N,
+        "hidden_dim": 16, "kernel_type": "random_walk_diffusion",
+        "cheby_order": 2, "device": "cpu", "compute_dtype": "float32",
+        "perspectives": 3, "learn_rate": 1e-4, "output_dir": str(tmp_path),
+        "checkpoint": str(tmp_path / "MPGCN_od.pkl"),
+    }
+    data = DataInput(params=params).load_data()
+    trainer = ModelTrainer(params, data)
+    torch.save({"epoch": 1, "state_dict": trainer.model.state_dict()},
+               params["checkpoint"])
+
+    dow = 4
+    x = torch.rand(7, N, N, 1)
+    fc = Forecaster(params, data)
+    served = fc.forecast(x, dow=dow, horizon=1)  # (1, N, N)
+
+    # trainer-side forward on the same input, dynamic graphs at the same dow
+    O_raw = data["O_dyn_G"][:, :, dow].unsqueeze(0).float()
+    D_raw = data["D_dyn_G"][:, :, dow].unsqueeze(0).float()
+    dyn = (trainer.preprocess_dynamic_graph(O_raw),
+           trainer.preprocess_dynamic_graph(D_raw))
+    trainer.model.eval()
+    with torch.no_grad():
+        expected = trainer.model(
+            x_seq=x.unsqueeze(0), G_list=trainer._graph_list(dyn)
+        )[0, :, :, :, 0]
+    assert torch.allclose(served, expected, atol=1e-6), (
+        (served - expected).abs().max()
+    )
