@@ -164,5 +164,13 @@ def mpgcn_forward_sharded(model, x_seq_shard, G_list, group=None):
         fc = branch["fc"][0]
         out = linear_act(X.reshape(B * N * Nl, -1), fc.weight.to(cd), fc.bias, True)
         outs.append(out.view(B, N, Nl, 1))
-    ens = torch.mean(torch.stack(outs, dim=-1), dim=-1)
+    stacked = torch.stack(outs, dim=-1)
+    if model.fusion == "attention":
+        # learned softmax fusion is pointwise over (origin, destination) pairs,
+        # so it shards trivially; fusion_w is replicated and its gradient
+        # all-reduces through the usual GradAllReducer like every other weight
+        w = torch.softmax(model.fusion_w, dim=0).to(stacked.dtype)
+        ens = (stacked * w).sum(dim=-1)
+    else:
+        ens = torch.mean(stacked, dim=-1)
     return ens.float().unsqueeze(1)

@@ -143,6 +143,34 @@ def test_split_halves_match_full_layer_rectangular():
     torch.testing.assert_close(bias.grad, gb, atol=1e-5, rtol=1e-4)
 
 
+def test_sharded_attention_fusion_matches_unsharded(tmp_path):
+    """-fusion attention + -partition region: the sharded forward applies the
+    learned softmax fusion (it is pointwise, so shard-compatible) instead of
+    silently degrading to mean; P=1 sharded output must equal the full model."""
+    from mpgcn_amd.parallel.region import mpgcn_forward_sharded
+
+    store_file = str(tmp_path / "pg_att")
+    dist.init_process_group("gloo", init_method=f"file://{store_file}",
+                            rank=0, world_size=1)
+    try:
+        torch.manual_seed(2)
+        model = MPGCN(M=2, K=K, input_dim=1, lstm_hidden_dim=H,
+                      lstm_num_layers=1, gcn_hidden_dim=H, gcn_num_layers=2,
+                      num_nodes=N, fusion="attention")
+        with torch.no_grad():
+            model.fusion_w.copy_(torch.tensor([0.7, -0.3]))  # non-uniform
+        x, y, Gs, Go, Gd = _inputs()
+        out_sharded = mpgcn_forward_sharded(model, x, [Gs, (Go, Gd)])
+        ref = model(x, [Gs, (Go, Gd)])
+        torch.testing.assert_close(out_sharded, ref, atol=1e-6, rtol=1e-5)
+        # fusion_w must receive gradient through the sharded path
+        out_sharded.square().sum().backward()
+        assert model.fusion_w.grad is not None
+        assert model.fusion_w.grad.abs().sum() > 0
+    finally:
+        dist.destroy_process_group()
+
+
 def _trainer_worker(rank, file_name, out_dir):
     os.environ.update(RANK=str(rank), WORLD_SIZE=str(P), LOCAL_RANK=str(rank))
     dist.init_process_group("gloo", init_method=f"file://{file_name}",
