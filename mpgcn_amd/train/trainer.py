@@ -261,40 +261,80 @@ class ModelTrainer:
             torch.save(checkpoint, self._ckpt_path())
 
     # -- test loop: autoregressive rollout (Model_Trainer.py:145-185) --
+    @torch.no_grad()
+    def _rollout(self, x_seq, dyn, region: bool):
+        """pred_len-step autoregressive rollout. Dynamic graphs held at the
+        first step's day-of-week across horizons — kept for score
+        compatibility with the reference (Model_Trainer.py:156-163); see docs
+        for the quirk note. Region mode: x_seq is destination-sharded and the
+        feedback loop stays shard-local (the step output shards the same way)."""
+        y_pred = []
+        cur_x_seq = x_seq
+        for _ in range(self.params["pred_len"]):
+            if region:
+                from mpgcn_amd.parallel.region import mpgcn_forward_sharded
+
+                step_y = mpgcn_forward_sharded(
+                    self.model, cur_x_seq, self._graph_list(dyn)
+                )
+            else:
+                step_y = self.model(x_seq=cur_x_seq, G_list=self._graph_list(dyn))
+            cur_x_seq = torch.cat([cur_x_seq[:, 1:], step_y], dim=1)
+            y_pred.append(step_y)
+        return torch.cat(y_pred, dim=1)
+
     def test(self, data_loader: dict, modes: list):
-        """Evaluation runs the full (unsharded) model on every rank — the
-        region-partition mode only shards training steps; eval at the tested
-        scales fits a single GPU and rank 0 writes the scores file."""
+        """Evaluation shards across ranks: region mode runs the sharded
+        forward (per-rank memory O(N^2/P), same envelope as training); DP mode
+        round-robins batches over ranks. Metric sufficient statistics
+        all-reduce and rank 0 writes the scores file. Single-process keeps the
+        reference's numpy path (byte-compatible scores)."""
+        if self.ctx.enabled:
+            # rank 0 writes the checkpoint at train end; don't read it early
+            torch.distributed.barrier()
         ckpt = torch.load(self._ckpt_path(), map_location=self.device, weights_only=False)
         self.model.load_state_dict(ckpt["state_dict"])
         self.model.eval()
+        region = self.partition == "region" and self.ctx.enabled
 
         log = print if self.ctx.is_main else (lambda *a, **k: None)
         for mode in modes:
             log("\n", datetime.now().strftime("%Y/%m/%d %H:%M:%S"))
             log(f'     {self.params["model"]} model testing on {mode} data begins:')
-            forecast, ground_truth = [], []
-            for x_seq, y_true, O_dyn_G, D_dyn_G in data_loader[mode]:
-                dyn = (
-                    self.preprocess_dynamic_graph(O_dyn_G),
-                    self.preprocess_dynamic_graph(D_dyn_G),
-                )
-                y_pred = []
-                cur_x_seq = x_seq
-                with torch.no_grad():
-                    # dynamic graphs held at the first step's day-of-week across
-                    # horizons — kept for score compatibility with the reference
-                    # (Model_Trainer.py:156-163); see docs for the quirk note.
-                    for _ in range(self.params["pred_len"]):
-                        step_y = self.model(x_seq=cur_x_seq, G_list=self._graph_list(dyn))
-                        cur_x_seq = torch.cat([cur_x_seq[:, 1:], step_y], dim=1)
-                        y_pred.append(step_y)
-                forecast.append(torch.cat(y_pred, dim=1).cpu().numpy())
-                ground_truth.append(y_true.cpu().numpy())
+            if self.ctx.enabled:
+                acc = metrics_mod.MetricAccumulator(device=self.device)
+                for i, (x_seq, y_true, O_dyn_G, D_dyn_G) in enumerate(data_loader[mode]):
+                    if not region and i % self.ctx.world_size != self.ctx.rank:
+                        continue  # DP: round-robin batches over ranks
+                    dyn = (
+                        self.preprocess_dynamic_graph(O_dyn_G),
+                        self.preprocess_dynamic_graph(D_dyn_G),
+                    )
+                    if region:
+                        from mpgcn_amd.parallel.region import shard_dest
 
-            forecast = np.concatenate(forecast, axis=0)
-            ground_truth = np.concatenate(ground_truth, axis=0)
-            MSE, RMSE, MAE, MAPE = metrics_mod.evaluate(forecast, ground_truth)
+                        x_seq = shard_dest(x_seq, self.ctx.rank, self.ctx.world_size)
+                        y_true = shard_dest(y_true, self.ctx.rank, self.ctx.world_size)
+                    acc.update(self._rollout(x_seq, dyn, region), y_true)
+                acc.all_reduce()
+                MSE, RMSE, MAE, MAPE, PCC = acc.finalize()
+                log("MSE:", round(MSE, 4))
+                log("RMSE:", round(RMSE, 4))
+                log("MAE:", round(MAE, 4))
+                log("MAPE:", round(MAPE * 100, 4), "%")
+                log("PCC:", round(PCC, 4))
+            else:
+                forecast, ground_truth = [], []
+                for x_seq, y_true, O_dyn_G, D_dyn_G in data_loader[mode]:
+                    dyn = (
+                        self.preprocess_dynamic_graph(O_dyn_G),
+                        self.preprocess_dynamic_graph(D_dyn_G),
+                    )
+                    forecast.append(self._rollout(x_seq, dyn, False).cpu().numpy())
+                    ground_truth.append(y_true.cpu().numpy())
+                forecast = np.concatenate(forecast, axis=0)
+                ground_truth = np.concatenate(ground_truth, axis=0)
+                MSE, RMSE, MAE, MAPE = metrics_mod.evaluate(forecast, ground_truth)
             if self.ctx.is_main:
                 with open(self._scores_path(), "a") as f:
                     f.write(

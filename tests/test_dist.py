@@ -115,6 +115,98 @@ def test_two_rank_dp_small_buckets_mid_backward_flush(tmp_path):
         assert torch.allclose(dp_grads[n], p.grad, atol=1e-5), n
 
 
+def _eval_params(out_dir):
+    return {
+        "model": "MPGCN", "device": "cpu",
+        "synthetic_nodes": 12, "synthetic_days": 80, "norm": "none",
+        "split_ratio": [6.4, 1.6, 2], "batch_size": 4,
+        "obs_len": 5, "pred_len": 2, "hidden_dim": 16,
+        "kernel_type": "random_walk_diffusion", "cheby_order": 2,
+        "loss": "MSE", "optimizer": "Adam", "learn_rate": 1e-3,
+        "decay_rate": 0, "num_epochs": 1, "seed": 0, "N": 12,
+        "output_dir": out_dir,
+    }
+
+
+def _eval_worker(rank, world, file_name, out_dir):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world), LOCAL_RANK=str(rank))
+    dist.init_process_group("gloo", init_method=f"file://{file_name}",
+                            rank=rank, world_size=world)
+    from mpgcn_amd.data import DataGenerator, DataInput
+    from mpgcn_amd.train import ModelTrainer
+
+    params = _eval_params(out_dir)
+    data = DataInput(params).load_data()
+    gen = DataGenerator(params["obs_len"], params["pred_len"], params["split_ratio"])
+    loaders = gen.get_data_loader(data, params)
+    ctx = DistContext(rank=rank, world_size=world, local_rank=rank, backend="gloo")
+    trainer = ModelTrainer(params, data, dist_ctx=ctx)
+    trainer.test(loaders, ["test"])
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_dp_sharded_eval_matches_single_process(tmp_path):
+    """trainer.test() in DP mode round-robins batches over ranks and
+    all-reduces metric statistics — the scores line must match the
+    single-process numpy path to float64-accumulation tolerance (and no rank
+    recomputes the full test set)."""
+    from mpgcn_amd.data import DataGenerator, DataInput
+    from mpgcn_amd.train import ModelTrainer
+
+    # one shared checkpoint, written before any worker starts
+    sp_dir = tmp_path / "sp"
+    dp_dir = tmp_path / "dp"
+    sp_dir.mkdir(); dp_dir.mkdir()
+    params = _eval_params(str(sp_dir))
+    data = DataInput(params).load_data()
+    torch.manual_seed(7)
+    trainer = ModelTrainer(params, data)
+    ckpt = {"epoch": 1, "state_dict": trainer.model.state_dict()}
+    torch.save(ckpt, str(sp_dir / "MPGCN_od.pkl"))
+    torch.save(ckpt, str(dp_dir / "MPGCN_od.pkl"))
+
+    gen = DataGenerator(params["obs_len"], params["pred_len"], params["split_ratio"])
+    loaders = gen.get_data_loader(data, params)
+    trainer.test(loaders, ["test"])
+    sp_line = open(sp_dir / "MPGCN_prediction_scores.txt").read().splitlines()[0]
+
+    file_name = str(tmp_path / "pg_eval")
+    ctxm = mp.get_context("spawn")
+    procs = [ctxm.Process(target=_eval_worker, args=(r, 2, file_name, str(dp_dir)))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=240)
+        assert p.exitcode == 0
+    dp_line = open(dp_dir / "MPGCN_prediction_scores.txt").read().splitlines()[0]
+
+    sp_vals = [float(v) for v in sp_line.split(", ")[5:]]
+    dp_vals = [float(v) for v in dp_line.split(", ")[5:]]
+    for a, b in zip(sp_vals, dp_vals):
+        # numpy's float32 pairwise mean vs exact f64 sums: ~1e-7 relative
+        assert abs(a - b) <= 1e-5 * max(abs(a), 1.0), (sp_line, dp_line)
+
+
+def test_metric_accumulator_matches_numpy():
+    from mpgcn_amd.train import metrics as mm
+
+    torch.manual_seed(0)
+    p = torch.rand(3, 50) * 4
+    t = torch.rand(3, 50) * 4
+    acc = mm.MetricAccumulator()
+    for i in range(3):  # batched updates must compose
+        acc.update(p[i], t[i])
+    mse, rmse, mae, mape, pcc = acc.finalize()
+    pn, tn = p.double().numpy(), t.double().numpy()  # f64 oracle
+    assert abs(mse - mm.MSE(pn, tn)) < 1e-12
+    assert abs(rmse - mm.RMSE(pn, tn)) < 1e-12
+    assert abs(mae - mm.MAE(pn, tn)) < 1e-12
+    assert abs(mape - mm.MAPE(pn, tn)) < 1e-12
+    assert abs(pcc - mm.PCC(pn, tn)) < 1e-9
+
+
 def test_noop_context_without_env():
     ctx = DistContext()
     model = _make_model()

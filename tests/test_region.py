@@ -198,6 +198,9 @@ def _trainer_worker(rank, file_name, out_dir):
     trainer = ModelTrainer(params=params, data=data, data_container=di,
                            dist_ctx=ctx)
     trainer.train(data_loader=loaders, modes=["train", "validate"])
+    # region-sharded evaluation: each rank rolls out its destination shard,
+    # metric statistics all-reduce, rank 0 writes the scores line
+    trainer.test(data_loader=loaders, modes=["test"])
     dist.destroy_process_group()
 
 
@@ -217,6 +220,10 @@ def test_region_trainer_end_to_end(tmp_path):
         p_.join(timeout=240)
         assert p_.exitcode == 0
     assert os.path.exists(os.path.join(out_dir, "MPGCN_od.pkl"))
+    scores = open(os.path.join(out_dir, "MPGCN_prediction_scores.txt")).read()
+    line = scores.splitlines()[0]
+    assert line.startswith("test, MSE, RMSE, MAE, MAPE, ")
+    assert all(v == v for v in map(float, line.split(", ")[5:]))  # finite
 
 
 def test_region_world_size_divisibility_error():
