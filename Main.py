@@ -71,6 +71,10 @@ def build_parser() -> argparse.ArgumentParser:
                         choices=["mean", "attention"], default="mean",
                         help="branch fusion: arithmetic mean (reference) or "
                              "learned attention weights")
+    parser.add_argument("-ref-quirks", "--ref_quirks", action="store_true",
+                        help="reproduce the reference's dynamic D-graph "
+                             "computation exactly (Data_Container_OD.py:53-56 "
+                             "column/row mixing) for bit-parity validation")
     parser.add_argument("-resume", "--resume", action="store_true",
                         help="resume training from the extended checkpoint "
                              "({model}_od.resume.pkl) if present")

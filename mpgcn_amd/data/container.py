@@ -65,7 +65,10 @@ class DataInput:
         train_len = int(raw.shape[0] * split[0] / sum(split))
         period = 7
         whole = (train_len // period) * period
-        O_dyn, D_dyn = construct_dynamic_graphs(raw[:whole], period=period)
+        O_dyn, D_dyn = construct_dynamic_graphs(
+            raw[:whole], period=period,
+            ref_quirks=bool(p.get("ref_quirks", False)),
+        )
 
         return {"OD": OD, "adj": adj, "O_dyn_G": O_dyn, "D_dyn_G": D_dyn}
 

@@ -38,9 +38,22 @@ def _cosine_distance_gram(rows: torch.Tensor, eps: float = 1e-12) -> torch.Tenso
     return dist
 
 
+def _ref_quirk_distance(cols: torch.Tensor, rows: torch.Tensor,
+                        eps: float = 1e-12) -> torch.Tensor:
+    """Reference-exact D-graph: D[i, j] = cosine_distance(col_i, ROW_j) —
+    the column/row mixing of Data_Container_OD.py:56, reproduced verbatim for
+    bit-parity validation runs (no diagonal forcing: cosine(col_i, row_i) is
+    generally nonzero). cols/rows: (P, N, F) with cols[p, i] = OD_avg[:, i]
+    and rows[p, j] = OD_avg[j, :]."""
+    cu = cols / cols.norm(dim=-1, keepdim=True).clamp(min=eps)
+    ru = rows / rows.norm(dim=-1, keepdim=True).clamp(min=eps)
+    return 1.0 - torch.bmm(cu, ru.transpose(-2, -1))
+
+
 def construct_dynamic_graphs(
     OD_history: torch.Tensor,
     period: int = 7,
+    ref_quirks: bool = False,
 ) -> tuple[torch.Tensor, torch.Tensor]:
     """Period-phase averaged OD -> (O_dyn_G, D_dyn_G), each (N, N, period).
 
@@ -49,6 +62,11 @@ def construct_dynamic_graphs(
     Data_Container_OD.py:40-42 which dumps the remainder weeks).
     Output layout matches the reference's (N, N, period) stacking
     (Data_Container_OD.py:59) so downstream day-of-week indexing is identical.
+
+    ref_quirks=True reproduces the reference's D-graph computation EXACTLY
+    (Data_Container_OD.py:53-56): column i against ROW j, no symmetric
+    correction and no diagonal forcing — for users validating bit-parity
+    against the upstream framework on the same real dataset.
     """
     if OD_history.dim() == 4:
         OD_history = OD_history.squeeze(-1)
@@ -61,8 +79,15 @@ def construct_dynamic_graphs(
     # (T, N, N) -> (weeks, period, N, N) -> phase means (period, N, N)
     phase_avg = OD_history.reshape(T // period, period, N, N).mean(dim=0)
 
-    O_dyn = _cosine_distance_gram(phase_avg)  # rows = origin profiles
-    D_dyn = _cosine_distance_gram(phase_avg.transpose(-2, -1))  # rows = dest profiles
+    if ref_quirks:
+        # diagonal NOT forced to zero in quirk mode (scipy's cosine leaves
+        # ~1e-16 roundoff there; the reference keeps whatever scipy returns)
+        ru = phase_avg / phase_avg.norm(dim=-1, keepdim=True).clamp(min=1e-12)
+        O_dyn = 1.0 - torch.bmm(ru, ru.transpose(-2, -1))
+        D_dyn = _ref_quirk_distance(phase_avg.transpose(-2, -1), phase_avg)
+    else:
+        O_dyn = _cosine_distance_gram(phase_avg)  # rows = origin profiles
+        D_dyn = _cosine_distance_gram(phase_avg.transpose(-2, -1))  # rows = dest profiles
 
     # (period, N, N) -> (N, N, period), the reference's stacking layout
     return O_dyn.permute(1, 2, 0).contiguous(), D_dyn.permute(1, 2, 0).contiguous()

@@ -186,4 +186,7 @@ class MPGCN(nn.Module):
             ensemble = (stacked * w).sum(dim=-1)
         else:
             ensemble = torch.mean(stacked, dim=-1)
-        return ensemble.float().unsqueeze(1)
+        # upcast the bf16 compute result; keep f32/f64 (parity runs) as-is
+        if ensemble.dtype == torch.bfloat16:
+            ensemble = ensemble.float()
+        return ensemble.unsqueeze(1)

@@ -68,8 +68,10 @@ class MPGCNReference(nn.Module):
         outs = []
         for m in range(self.M):
             branch = self.branch_models[m]
-            h0 = torch.zeros(1, B * N * N, self.H, device=x_seq.device)
-            c0 = torch.zeros(1, B * N * N, self.H, device=x_seq.device)
+            h0 = torch.zeros(1, B * N * N, self.H, device=x_seq.device,
+                             dtype=x_seq.dtype)
+            c0 = torch.zeros(1, B * N * N, self.H, device=x_seq.device,
+                             dtype=x_seq.dtype)
             lstm_out, _ = branch["temporal"](lstm_in, (h0, c0))
             X = lstm_out[:, -1, :].reshape(B, N, N, self.H)
             G = G_list[m]
