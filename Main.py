@@ -99,8 +99,8 @@ def main():
         print("[mpgcn] no GPU visible, falling back to cpu")
         device = "cpu"
     ctx = init_distributed(device)
-    if ctx.enabled and device.startswith("cuda"):
-        device = f"cuda:{ctx.local_rank}"
+    if ctx.enabled and device.startswith("cuda") and torch.cuda.is_available():
+        device = f"cuda:{ctx.local_rank % torch.cuda.device_count()}"
     params["device"] = device
 
     data_input = DataInput(params=params)

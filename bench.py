@@ -61,7 +61,10 @@ def main():
     if args.device:
         device = args.device
     elif torch.cuda.is_available():
-        device = f"cuda:{int(os.environ.get('LOCAL_RANK', 0))}"
+        # ranks share devices round-robin when oversubscribed (2-rank RCCL
+        # bring-up on a 1-GPU box)
+        local = int(os.environ.get("LOCAL_RANK", 0)) % torch.cuda.device_count()
+        device = f"cuda:{local}"
     else:
         device = "cpu"
     ctx = init_distributed(device)

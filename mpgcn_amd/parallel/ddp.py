@@ -59,7 +59,9 @@ def init_distributed(device: str = "cuda", backend: str | None = None,
         backend = "nccl" if (device.startswith("cuda") and torch.cuda.is_available()) else "gloo"
     if not dist.is_initialized():
         if backend == "nccl":
-            torch.cuda.set_device(local)
+            # more ranks than GPUs (e.g. 2-rank RCCL bring-up on a 1-GPU box)
+            # share devices round-robin
+            torch.cuda.set_device(local % torch.cuda.device_count())
         dist.init_process_group(backend=backend, timeout=timedelta(seconds=timeout_s))
     return DistContext(rank=rank, world_size=world, local_rank=local, backend=backend)
 
