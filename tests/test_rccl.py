@@ -1,9 +1,18 @@
-"""RCCL bring-up on real hardware: 2 ranks sharing one MI355X over the nccl
-(=RCCL) backend — the round-1 verdict's single biggest risk was that the RCCL
-code path had never executed anywhere. These tests run process-group
-bootstrap, GradAllReducer (broadcast + bucketed async all-reduce, including
-forced MID-backward flushes against the branch side streams), and the region
-partition's all_to_all_single, all on the nccl backend.
+"""RCCL on real hardware — the round-1 verdict's single biggest risk was
+that the RCCL code path had never executed anywhere.
+
+Measured constraint (round 2): NCCL/RCCL 2.26 hard-rejects two ranks on one
+device ("Duplicate GPU detected"), and MI355X CPX compute partitioning is
+blocked in this container (rocm-smi set to CPX does not take), so cross-rank
+RCCL transport requires >= 2 physical GPUs. Therefore:
+
+  * the single-rank tests below run on the 1-GPU box and execute the real
+    RCCL communicator bring-up + collective launch path (init, all_reduce,
+    broadcast, all_to_all_single on CUDA tensors, stream semantics);
+  * the 2-rank tests skip unless >= 2 devices are visible, and run the full
+    DP/region transport (GradAllReducer incl. forced mid-backward flushes,
+    region all-to-all) on any multi-GPU node — e.g. the driver's 8-GPU
+    scaling run.
 
 Requires HSA_ENABLE_IPC_MODE_LEGACY=0 (dmabuf IPC; exported by the image) so
 cross-process CUDA tensor/RCCL transport works.
@@ -21,22 +30,56 @@ pytestmark = pytest.mark.gpu
 N, K, H, B, T = 16, 3, 32, 4, 5
 P = 2
 
+multi_gpu = pytest.mark.skipif(
+    not torch.cuda.is_available() or torch.cuda.device_count() < P,
+    reason="RCCL rejects 2 ranks on one device (Duplicate GPU); needs >= 2 GPUs",
+)
+
+
+@pytest.mark.timeout(300)
+def test_nccl_single_rank_bringup_collectives(tmp_path):
+    """World-1 RCCL communicator on the real GPU: init_process_group('nccl'),
+    all_reduce / broadcast / all_to_all_single launch through the RCCL
+    library and complete with correct results under stream semantics."""
+    os.environ.update(RANK="0", WORLD_SIZE="1", LOCAL_RANK="0",
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT="29809")
+    torch.cuda.set_device(0)
+    dist.init_process_group("nccl", rank=0, world_size=1)
+    try:
+        t = torch.full((1 << 20,), 3.0, device="cuda:0")
+        dist.all_reduce(t)
+        assert torch.all(t == 3.0).item()
+        b = torch.randn(64, 64, device="cuda:0")
+        ref = b.clone()
+        dist.broadcast(b, src=0)
+        torch.testing.assert_close(b, ref)
+        x = torch.arange(4096.0, device="cuda:0")
+        out = torch.empty_like(x)
+        dist.all_to_all_single(out, x)
+        torch.testing.assert_close(out, x)
+        g = torch.randn(1024, device="cuda:0", dtype=torch.bfloat16)
+        dist.all_reduce(g)  # bf16 reduction path (gradient dtype)
+        torch.cuda.synchronize()
+    finally:
+        dist.destroy_process_group()
+
 
 def _init(rank, port):
     os.environ.update(
         RANK=str(rank), WORLD_SIZE=str(P), LOCAL_RANK=str(rank),
         MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
     )
-    torch.cuda.set_device(0)  # both ranks share the single GPU
+    torch.cuda.set_device(rank % torch.cuda.device_count())
     dist.init_process_group("nccl", rank=rank, world_size=P)
 
 
 def _allreduce_worker(rank, port, out_file):
     _init(rank, port)
-    t = torch.full((1024,), float(rank + 1), device="cuda:0")
+    dev = f"cuda:{rank % torch.cuda.device_count()}"
+    t = torch.full((1024,), float(rank + 1), device=dev)
     dist.all_reduce(t)
     ok = bool(torch.all(t == 3.0).item())  # 1 + 2
-    x = torch.randn(8, 16, device="cuda:0") if rank == 0 else torch.empty(8, 16, device="cuda:0")
+    x = torch.randn(8, 16, device=dev) if rank == 0 else torch.empty(8, 16, device=dev)
     dist.broadcast(x, src=0)
     if rank == 0:
         torch.save({"allreduce_ok": ok, "bcast": x.cpu()}, out_file)
@@ -57,8 +100,9 @@ def _spawn(target, port, *args):
         assert p.exitcode == 0, [q.exitcode for q in procs]
 
 
+@multi_gpu
 @pytest.mark.timeout(300)
-def test_nccl_two_ranks_one_gpu_allreduce_broadcast(tmp_path):
+def test_nccl_two_rank_allreduce_broadcast(tmp_path):
     out = str(tmp_path / "ar.pt")
     _spawn(_allreduce_worker, 29811, out)
     got = torch.load(out, weights_only=True)
@@ -73,18 +117,19 @@ def _ddp_worker(rank, port, out_file, bucket_bytes):
     from mpgcn_amd.parallel import DistContext, GradAllReducer
 
     _init(rank, port)
-    ctx = DistContext(rank=rank, world_size=P, local_rank=0, backend="nccl")
+    dev = f"cuda:{rank % torch.cuda.device_count()}"
+    ctx = DistContext(rank=rank, world_size=P, local_rank=rank, backend="nccl")
     torch.manual_seed(100 + rank)  # divergent init: broadcast must fix it
     model = MPGCN(M=2, K=K, input_dim=1, lstm_hidden_dim=H, lstm_num_layers=1,
                   gcn_hidden_dim=H, gcn_num_layers=2, num_nodes=N,
-                  compute_dtype=torch.bfloat16).to("cuda:0")
+                  compute_dtype=torch.bfloat16).to(dev)
     reducer = GradAllReducer(model, ctx, bucket_bytes=bucket_bytes)
 
     torch.manual_seed(0)
-    x = torch.rand(B, T, N, N, 1, device="cuda:0")
-    y = torch.rand(B, 1, N, N, 1, device="cuda:0")
-    flow = torch.rand(B, N, N, device="cuda:0")
-    Gs = build_supports(torch.rand(1, N, N, device="cuda:0"),
+    x = torch.rand(B, T, N, N, 1, device=dev)
+    y = torch.rand(B, 1, N, N, 1, device=dev)
+    flow = torch.rand(B, N, N, device=dev)
+    Gs = build_supports(torch.rand(1, N, N, device=dev),
                         "random_walk_diffusion", K - 1)[0]
     Go = build_supports(flow, "random_walk_diffusion", K - 1)
     Gd = build_supports(flow.transpose(-2, -1), "random_walk_diffusion", K - 1)
@@ -102,6 +147,7 @@ def _ddp_worker(rank, port, out_file, bucket_bytes):
     dist.destroy_process_group()
 
 
+@multi_gpu
 @pytest.mark.timeout(300)
 @pytest.mark.parametrize("bucket_bytes", [16 << 20, 64])
 def test_nccl_ddp_grad_equivalence(tmp_path, bucket_bytes):
@@ -145,14 +191,15 @@ def _region_worker(rank, port, out_file):
     from mpgcn_amd.parallel.region import mpgcn_forward_sharded, shard_dest
 
     _init(rank, port)
+    dev = f"cuda:{rank % torch.cuda.device_count()}"
     torch.manual_seed(1)
     model = MPGCN(M=2, K=K, input_dim=1, lstm_hidden_dim=H, lstm_num_layers=1,
                   gcn_hidden_dim=H, gcn_num_layers=2, num_nodes=N,
-                  compute_dtype=torch.bfloat16).to("cuda:0")
+                  compute_dtype=torch.bfloat16).to(dev)
     torch.manual_seed(0)
-    x = torch.rand(B, T, N, N, 1, device="cuda:0")
-    flow = torch.rand(B, N, N, device="cuda:0")
-    Gs = build_supports(torch.rand(1, N, N, device="cuda:0"),
+    x = torch.rand(B, T, N, N, 1, device=dev)
+    flow = torch.rand(B, N, N, device=dev)
+    Gs = build_supports(torch.rand(1, N, N, device=dev),
                         "random_walk_diffusion", K - 1)[0]
     Go = build_supports(flow, "random_walk_diffusion", K - 1)
     Gd = build_supports(flow.transpose(-2, -1), "random_walk_diffusion", K - 1)
@@ -171,6 +218,7 @@ def _region_worker(rank, port, out_file):
     dist.destroy_process_group()
 
 
+@multi_gpu
 @pytest.mark.timeout(300)
 def test_nccl_region_all_to_all_matches_unsharded(tmp_path):
     """Region partition over RCCL all_to_all_single on hardware: rank 0's
