@@ -55,8 +55,9 @@ def build_parser() -> argparse.ArgumentParser:
                              "loading npz files from --input_dir")
     parser.add_argument("-synthetic-days", "--synthetic_days", type=int, default=425)
     parser.add_argument("-dtype", "--compute_dtype", type=str,
-                        choices=["float32", "bf16"], default="float32",
-                        help="Compute dtype on GPU (fp32 master weights either way)")
+                        choices=["float32", "bf16", "fp8"], default="float32",
+                        help="Compute dtype on GPU (fp32 master weights either "
+                             "way); fp8 = fp8-forward/bf16-backward mode")
     parser.add_argument("-shuffle", "--shuffle", action="store_true",
                         help="Shuffle training batches (reference default: off)")
     parser.add_argument("-seed", "--seed", type=int, default=0)

@@ -31,7 +31,10 @@ def main():
     ap.add_argument("--batch", type=int, default=32, help="per-GPU batch size")
     ap.add_argument("--hidden", type=int, default=32)
     ap.add_argument("--obs-len", type=int, default=7)
-    ap.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "float32"])
+    ap.add_argument("--dtype", type=str, default="bf16",
+                    choices=["bf16", "float32", "fp8"],
+                    help="fp8 = opt-in fp8-forward/bf16-backward mode (the "
+                         "contract headline stays bf16)")
     ap.add_argument("--device", type=str, default=None,
                     help="override device (cpu for debug)")
     ap.add_argument("--impl", type=str, default="native", choices=["native", "eager"],
@@ -77,7 +80,9 @@ def main():
     K_order = 2
     kernel = "random_walk_diffusion"
     S = K_order + 1
-    cdtype = torch.bfloat16 if (args.dtype == "bf16" and is_cuda) else torch.float32
+    fp8 = args.dtype == "fp8" and is_cuda
+    cdtype = (torch.bfloat16 if (args.dtype in ("bf16", "fp8") and is_cuda)
+              else torch.float32)
 
     if args.impl == "eager":
         from mpgcn_amd.models.reference_eager import MPGCNReference
@@ -89,7 +94,7 @@ def main():
         model = MPGCN(M=args.branches, K=S, input_dim=1, lstm_hidden_dim=H,
                       lstm_num_layers=1, gcn_hidden_dim=H, gcn_num_layers=3,
                       num_nodes=N, compute_dtype=cdtype,
-                      fusion=args.fusion).to(device)
+                      fusion=args.fusion, fp8_forward=fp8).to(device)
     use_graph = args.graph and is_cuda and world == 1 and args.impl == "native"
     opt = torch.optim.Adam(model.parameters(), lr=1e-4, capturable=use_graph)
     reducer = GradAllReducer(model, ctx)
@@ -197,7 +202,8 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16" if cdtype == torch.bfloat16 else "float32",
+            "dtype": ("fp8" if fp8 else
+                      "bf16" if cdtype == torch.bfloat16 else "float32"),
             "data": "synthetic",
             "config": {
                 "model": "MPGCN" if args.impl == "native" else "MPGCN-reference-eager",

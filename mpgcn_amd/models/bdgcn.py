@@ -12,7 +12,7 @@ from __future__ import annotations
 import torch
 from torch import nn
 
-from mpgcn_amd.ops import GraphOperator, bdgcn_layer
+from mpgcn_amd.ops import GraphOperator, bdgcn_layer, bdgcn_layer_fp8
 
 
 class BDGCN(nn.Module):
@@ -33,9 +33,16 @@ class BDGCN(nn.Module):
         else:
             self.register_parameter("b", None)
 
-    def forward(self, X: torch.Tensor, gop: GraphOperator) -> torch.Tensor:
-        """X: (B, N, N, input_dim) -> (B, N, N, hidden_dim)."""
-        W = self.W.to(X.dtype)
+    def forward(self, X: torch.Tensor, gop: GraphOperator, fp8: bool = False,
+                X8: torch.Tensor | None = None):
+        """X: (B, N, N, input_dim) -> (B, N, N, hidden_dim).
+
+        fp8=True runs the fp8-forward/bf16-backward path and returns
+        (Y, Y8_twin) so the caller can thread the fp8 twin into the next
+        layer without a quantize pass (ops/functional.py)."""
+        W = self.W.to(X.dtype if X.dtype != torch.float8_e4m3fn else torch.bfloat16)
+        if fp8:
+            return bdgcn_layer_fp8(X, W, self.b, gop, relu=self.relu, X8=X8)
         return bdgcn_layer(X, W, self.b, gop, relu=self.relu)
 
     def extra_repr(self) -> str:

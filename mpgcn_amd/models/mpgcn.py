@@ -62,12 +62,27 @@ class MPGCN(nn.Module):
     def __init__(self, M: int, K: int, input_dim: int, lstm_hidden_dim: int,
                  lstm_num_layers: int, gcn_hidden_dim: int, gcn_num_layers: int,
                  num_nodes: int, user_bias: bool = True, activation: str = "relu",
-                 compute_dtype: torch.dtype = torch.float32, fusion: str = "mean"):
+                 compute_dtype: torch.dtype = torch.float32, fusion: str = "mean",
+                 fp8_forward: bool = False):
         super().__init__()
         if lstm_num_layers != 1:
             raise ValueError("MPGCN uses a 1-layer LSTM (Model_Trainer.py:50)")
         if fusion not in ("mean", "attention"):
             raise ValueError("fusion must be 'mean' or 'attention'")
+        if fp8_forward:
+            from mpgcn_amd.ops import fp8_forward_compatible
+
+            if compute_dtype != torch.bfloat16:
+                raise ValueError("fp8_forward requires compute_dtype=bf16 "
+                                 "(fp8 forward / bf16 backward)")
+            if not fp8_forward_compatible(num_nodes, lstm_hidden_dim,
+                                          gcn_hidden_dim, K):
+                raise ValueError(
+                    f"fp8_forward shape gate failed for N={num_nodes}, "
+                    f"C={lstm_hidden_dim}, H={gcn_hidden_dim}, S={K}: needs "
+                    "(N*C)%256==0, C%16==0, (N*H)%256==0, H%16==0, S*H<=128 "
+                    "(ops/functional.py fp8_forward_compatible)")
+        self.fp8_forward = fp8_forward
         self.M = M
         self.fusion = fusion
         self.K = K
@@ -140,12 +155,21 @@ class MPGCN(nn.Module):
             .contiguous()
         )
 
+        fp8 = self.fp8_forward and x_seq.is_cuda
+
         def run_branch(m: int) -> torch.Tensor:
             branch = self.branch_models[m]
             h_last = branch["temporal"](lstm_in)  # (B*N*N, H)
             X = h_last.reshape(B, N, N, self.lstm_hidden_dim)
-            for layer in branch["spatial"]:
-                X = layer(X, gops[m])
+            if fp8:
+                # fp8 twins chain layer-to-layer through the dual-write
+                # epilogues; only the LSTM output needs a standalone cast
+                X8 = None
+                for layer in branch["spatial"]:
+                    X, X8 = layer(X, gops[m], fp8=True, X8=X8)
+            else:
+                for layer in branch["spatial"]:
+                    X = layer(X, gops[m])
             fc = branch["fc"][0]
             out = linear_act(
                 X.reshape(B * N * N, -1), fc.weight.to(X.dtype), fc.bias, relu=True

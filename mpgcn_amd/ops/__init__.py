@@ -61,6 +61,8 @@ from mpgcn_amd.ops import eager  # noqa: E402,F401
 from mpgcn_amd.ops.functional import (  # noqa: E402,F401
     GraphOperator,
     bdgcn_layer,
+    bdgcn_layer_fp8,
+    fp8_forward_compatible,
     fused_lstm_last,
     linear_act,
     mode1_proj,
@@ -73,6 +75,8 @@ __all__ = [
     "eager",
     "GraphOperator",
     "bdgcn_layer",
+    "bdgcn_layer_fp8",
+    "fp8_forward_compatible",
     "fused_lstm_last",
     "linear_act",
     "mode1_proj",

@@ -35,7 +35,7 @@ __device__ __forceinline__ long ocol_off(const AxisGemmParams& p, int q) {
 }
 
 template <typename T, int BM, int BN, int BK, int WVM, int WVN, int BUFS = 2,
-          bool VEC_ONLY = false>
+          bool VEC_ONLY = false, typename OT = T, typename OT2 = OT>
 __launch_bounds__(WVM * WVN * 64) __global__ void axis_gemm_kernel(AxisGemmParams p) {
     using MT = MfmaTraits<T>;
     constexpr int CH = 16 / sizeof(T);
@@ -50,7 +50,8 @@ __launch_bounds__(WVM * WVN * 64) __global__ void axis_gemm_kernel(AxisGemmParam
 
     const T* __restrict__ A = (const T*)p.AT;
     const T* __restrict__ X = (const T*)p.X;
-    T* __restrict__ O = (T*)p.OUT;
+    OT* __restrict__ O = (OT*)p.OUT;
+    OT2* __restrict__ O2 = (OT2*)p.OUT2;  // optional second copy (fp8 twins)
 
     const int inst = blockIdx.y;
     const long a_base = (long)(inst / p.a_div) * p.a_bs1 + (long)(inst % p.a_div) * p.a_bs2;
@@ -174,24 +175,40 @@ __launch_bounds__(WVM * WVN * 64) __global__ void axis_gemm_kernel(AxisGemmParam
                 if (m < p.M) {
                     float v = acc[mf][nf][r] + bv;
                     if (p.relu) v = fmaxf(v, 0.f);
-                    O[oc + (long)m * p.o_row] = from_f32<T>(v);
+                    O[oc + (long)m * p.o_row] = from_f32<OT>(v);
+                    if (p.OUT2) O2[oc + (long)m * p.o_row] = from_f32<OT2>(v);
                 }
             }
         }
     }
 }
 
-// fp8 probe path (measurement only, docs/ROADMAP.md): same engine at
-// BK=128 — equal LDS footprint to the bf16 BK=64 tile but half the staged
-// bytes per K element, i.e. half the stage count at the same per-stage cost.
+// fp8 engine: same schedule at BK=128 — equal LDS footprint to the bf16
+// BK=64 tile but half the staged bytes per K element, i.e. half the stage
+// count at the same per-stage cost (measured 1.43x mode-2 / 1.30x mode-1,
+// profiles/SUMMARY.md "fp8 probe"). out_kind selects the training epilogues:
+//   0: fp8 out only (the round-1 probe contract)
+//   1: fp8 out + bf16 twin  (mode-1: U8 feeds the fp8 projection, U_bf16 is
+//      the saved backward operand — forward stays fp8, backward stays bf16)
+//   2: bf16 out + fp8 twin  (mode-2: Y_bf16 is the autograd output/ReLU mask,
+//      Y8 feeds the next layer's mode-1 without a separate quantize pass)
 extern "C" void axis_gemm_fp8_launch(AxisGemmParams p, int instances,
-                                     hipStream_t stream) {
+                                     int out_kind, hipStream_t stream) {
     constexpr int BN = 256, BM = 256;
     const int tiles_m = (p.M + BM - 1) / BM;
     p.tiles_l = (p.L + BN - 1) / BN;
     dim3 grid(tiles_m * p.tiles_l, instances);
-    axis_gemm_kernel<unsigned char, 256, 256, 128, 4, 2, 1, true>
-        <<<grid, dim3(512), 0, stream>>>(p);
+    if (out_kind == 1)
+        axis_gemm_kernel<unsigned char, 256, 256, 128, 4, 2, 1, true,
+                         unsigned char, __bf16>
+            <<<grid, dim3(512), 0, stream>>>(p);
+    else if (out_kind == 2)
+        axis_gemm_kernel<unsigned char, 256, 256, 128, 4, 2, 1, true,
+                         __bf16, unsigned char>
+            <<<grid, dim3(512), 0, stream>>>(p);
+    else
+        axis_gemm_kernel<unsigned char, 256, 256, 128, 4, 2, 1, true>
+            <<<grid, dim3(512), 0, stream>>>(p);
 }
 
 extern "C" void axis_gemm_launch(AxisGemmParams p, int instances, int is_f32,

@@ -426,7 +426,7 @@ torch::Tensor bdgcn_mode2_fp8(torch::Tensor V8, torch::Tensor A2T8,
     p.bias_mod = (int)H;
     p.a_vec = ((N * S) % 16 == 0);
     p.x_vec = (H % 16 == 0);
-    axis_gemm_fp8_launch(p, (int)B, stream());
+    axis_gemm_fp8_launch(p, (int)B, 0, stream());
     return Y;
 }
 
@@ -456,8 +456,123 @@ torch::Tensor bdgcn_mode1_fp8(torch::Tensor X8, torch::Tensor GT8) {
     p.ogdiv = (int)C; p.og_hi = S * C;
     p.a_vec = (N % 16 == 0);
     p.x_vec = ((N * C) % 16 == 0) && (C % 16 == 0);
-    axis_gemm_fp8_launch(p, (int)(B * S), stream());
+    axis_gemm_fp8_launch(p, (int)(B * S), 0, stream());
     return U;
+}
+
+namespace {
+void check_fp8(const torch::Tensor& t, const char* name) {
+    TORCH_CHECK(t.is_cuda() && t.is_contiguous() &&
+                t.scalar_type() == torch::kFloat8_e4m3fn,
+                name, " must be contiguous CUDA float8_e4m3fn");
+}
+}  // namespace
+
+// fp8-forward training mode-1: U = mode1(X8, GT8) with DUAL outputs —
+// U8 (fp8, feeds the fp8 projection GEMM) and U_bf16 (the saved backward
+// operand for the dW reduction). Forward math runs entirely on fp8 operands
+// (half the staged bytes through the byte-bound load path — the measured
+// 1.30x kernel lever); backward stays bf16. Shape contract: the fp8 tile is
+// vector-only, so the column extent Nd*C must be a multiple of 256 and C a
+// multiple of 16 (flagship and large-N configs satisfy this; the Python
+// layer falls back to bf16 otherwise).
+std::vector<torch::Tensor> bdgcn_mode1_fp8_train(torch::Tensor X8,
+                                                 torch::Tensor GT8) {
+    check_fp8(X8, "X8");
+    check_fp8(GT8, "GT8");
+    const bool dyn = GT8.dim() == 4;
+    const long B = X8.size(0), No = X8.size(1), Nd = X8.size(2), C = X8.size(3);
+    const long S = dyn ? GT8.size(1) : GT8.size(0);
+    TORCH_CHECK(GT8.size(-1) == No && GT8.size(-2) == No, "shape mismatch");
+    TORCH_CHECK(!dyn || GT8.size(0) == B, "dynamic GT8 batch mismatch");
+    TORCH_CHECK(B * S <= 65535, "too many instances");
+    TORCH_CHECK((Nd * C) % 256 == 0 && C % 16 == 0,
+                "fp8 mode-1 needs (Nd*C) % 256 == 0 and C % 16 == 0");
+    auto U8 = torch::empty({B, No, Nd, S, C}, X8.options());
+    auto Ubf = torch::empty({B, No, Nd, S, C},
+                            X8.options().dtype(torch::kBFloat16));
+
+    AxisGemmParams p{};
+    p.AT = GT8.data_ptr();
+    p.X = X8.data_ptr();
+    p.OUT = U8.data_ptr();
+    p.OUT2 = Ubf.data_ptr();
+    p.M = (int)No; p.K = (int)No; p.L = (int)(Nd * C);
+    p.a_div = (int)S; p.a_bs1 = dyn ? S * No * No : 0; p.a_bs2 = No * No;
+    p.x_div = (int)S; p.x_bs1 = No * Nd * C; p.x_bs2 = 0;
+    p.o_div = (int)S; p.o_bs1 = No * Nd * S * C; p.o_bs2 = C;
+    p.kdiv = 1; p.k_lo = Nd * C;
+    p.qdiv = 0;
+    p.o_row = Nd * S * C;
+    p.ogdiv = (int)C; p.og_hi = S * C;
+    p.a_vec = (No % 16 == 0);
+    p.x_vec = 1;
+    axis_gemm_fp8_launch(p, (int)(B * S), 1, stream());
+    return {U8, Ubf};
+}
+
+// fp8-forward training mode-2: Y = mode2(V8, A2T8) + bias + act with DUAL
+// outputs — Y_bf16 (the autograd output / ReLU mask operand) and Y8 (fp8
+// twin that feeds the NEXT layer's mode-1 without a separate quantize pass).
+std::vector<torch::Tensor> bdgcn_mode2_fp8_train(torch::Tensor V8,
+                                                 torch::Tensor A2T8,
+                                                 c10::optional<torch::Tensor> bias,
+                                                 bool relu, long N, long S) {
+    check_fp8(V8, "V8");
+    check_fp8(A2T8, "A2T8");
+    const bool dyn = A2T8.dim() == 3;
+    const long B = V8.size(0), Nm = V8.size(1), H = V8.size(-1);
+    TORCH_CHECK(A2T8.size(-1) == N * S && A2T8.size(-2) == N, "A2T8 shape");
+    TORCH_CHECK(B <= 65535, "too many instances");
+    TORCH_CHECK((Nm * H) % 256 == 0 && H % 16 == 0,
+                "fp8 mode-2 needs (Nm*H) % 256 == 0 and H % 16 == 0");
+    auto Y = torch::empty({B, Nm, N, H}, V8.options().dtype(torch::kBFloat16));
+    auto Y8 = torch::empty({B, Nm, N, H}, V8.options());
+
+    AxisGemmParams p{};
+    p.AT = A2T8.data_ptr();
+    p.X = V8.data_ptr();
+    p.OUT = Y.data_ptr();
+    p.OUT2 = Y8.data_ptr();
+    p.bias = bias_ptr(bias);
+    p.M = (int)N; p.K = (int)(N * S); p.L = (int)(Nm * H);
+    p.a_div = 1; p.a_bs1 = dyn ? N * N * S : 0; p.a_bs2 = 0;
+    p.x_div = 1; p.x_bs1 = Nm * N * S * H; p.x_bs2 = 0;
+    p.o_div = 1; p.o_bs1 = Nm * N * H; p.o_bs2 = 0;
+    p.kdiv = 1; p.k_lo = H;
+    p.qdiv = (int)H; p.q_hi = N * S * H;
+    p.o_row = H;
+    p.ogdiv = (int)H; p.og_hi = N * H;
+    p.relu = relu ? 1 : 0;
+    p.bias_mod = (int)H;
+    p.a_vec = ((N * S) % 16 == 0);
+    p.x_vec = 1;
+    axis_gemm_fp8_launch(p, (int)B, 2, stream());
+    return {Y, Y8};
+}
+
+// fp8 projection GEMM: OUT8[r, n] = X8[r, :] @ W8 — the fp8-forward mode's
+// V = U @ Wre (output feeds fp8 mode-2; V is not needed by backward, so no
+// bf16 twin). K must be a multiple of 16 (vectorized fp8 row reads).
+torch::Tensor row_gemm_fp8(torch::Tensor X8, torch::Tensor W8) {
+    check_fp8(X8, "X8");
+    check_fp8(W8, "W8");
+    const long R = X8.size(0), K = X8.size(1), N = W8.size(1);
+    TORCH_CHECK(W8.size(0) == K, "W8 shape mismatch");
+    TORCH_CHECK(N <= 128, "row_gemm_fp8: N must be <= 128");
+    TORCH_CHECK(K % 16 == 0 && K <= 2048, "row_gemm_fp8: K % 16 != 0 or too large");
+    auto OUT = torch::empty({R, N}, X8.options());
+    RowGemmParams p{};
+    p.X = X8.data_ptr();
+    p.W = W8.data_ptr();
+    p.OUT = OUT.data_ptr();
+    p.bias = nullptr;
+    p.R = R; p.K = (int)K; p.N = (int)N;
+    p.o_row = N; p.o_off = 0;
+    p.relu = 0;
+    p.x_vec = 1;
+    row_gemm_fp8_launch(p, stream());
+    return OUT;
 }
 
 // Fused ReLU backward + bias-grad column sum: dY = dH * 1[Y>0], dbias=colsum.
@@ -502,6 +617,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("relu_bwd_colsum", &relu_bwd_colsum, "fused ReLU bwd mask + bias colsum");
     m.def("bdgcn_mode2_fp8", &bdgcn_mode2_fp8, "fp8 e4m3 mode-2 probe (measurement only)");
     m.def("bdgcn_mode1_fp8", &bdgcn_mode1_fp8, "fp8 e4m3 mode-1 probe (measurement only)");
+    m.def("bdgcn_mode1_fp8_train", &bdgcn_mode1_fp8_train,
+          "fp8-forward mode-1 with bf16 backward twin");
+    m.def("bdgcn_mode2_fp8_train", &bdgcn_mode2_fp8_train,
+          "fp8-forward mode-2 + bias + act, bf16 out + fp8 twin");
+    m.def("row_gemm_fp8", &row_gemm_fp8, "fp8 projection GEMM");
     m.def("row_gemm_out", &row_gemm_out, "row GEMM into strided output slice");
     m.def("lstm_step_fwd", &lstm_step_fwd, "fused LSTM cell forward step (K4)");
     m.def("lstm_fused_fwd", &lstm_fused_fwd, "register-resident fused LSTM forward");

@@ -8,6 +8,7 @@ struct AxisGemmParams {
     const void* AT;
     const void* X;
     void* OUT;
+    void* OUT2;  // optional second epilogue copy (fp8/bf16 twin); nullable
     const float* bias;
     int M, K, L;
     int a_div; long a_bs1, a_bs2;
@@ -107,9 +108,10 @@ struct ReluBwdParams {
 extern "C" {
 void slab_colsum_launch(const float* ws, float* out, long nb, long E, hipStream_t s);
 void axis_gemm_launch(AxisGemmParams p, int instances, int is_f32, hipStream_t s);
-void axis_gemm_fp8_launch(AxisGemmParams p, int instances, hipStream_t s);
+void axis_gemm_fp8_launch(AxisGemmParams p, int instances, int out_kind, hipStream_t s);
 void red_gemm_launch(RedGemmParams p, int is_f32, hipStream_t s);
 void row_gemm_launch(RowGemmParams p, int is_f32, hipStream_t s);
+void row_gemm_fp8_launch(RowGemmParams p, hipStream_t s);
 void lstm_step_fwd_launch(LstmStepParams p, int is_f32, hipStream_t s);
 void lstm_step_bwd_launch(LstmBwdParams p, int is_f32, hipStream_t s);
 void lstm_fused_fwd_launch(LstmFusedParams p, hipStream_t s);

@@ -96,3 +96,16 @@ extern "C" void row_gemm_launch(RowGemmParams p, int is_f32, hipStream_t stream)
     else
         row_gemm_kernel<float><<<grid, block, smem, stream>>>(p, nf16);
 }
+
+// fp8 e4m3 path (fp8-forward training mode): X rows and W both fp8, fp8 out.
+// Same schedule; the OCP fp8 16x16x32 MFMA runs at the bf16 rate and the row
+// stream is half the bytes — this GEMM is memory-bound on the X rows
+// (profiles/SUMMARY.md: 2.2 TB/s bf16), so byte halving is the lever.
+extern "C" void row_gemm_fp8_launch(RowGemmParams p, hipStream_t stream) {
+    const int nf16 = (p.N + 15) / 16;
+    const size_t smem = (size_t)nf16 * 16 * (p.K + 16) * 1 + 64;
+    long tiles = (p.R + 63) / 64;
+    if (tiles > 16384) tiles = 16384;
+    dim3 grid((unsigned)tiles), block(256);
+    row_gemm_kernel<unsigned char><<<grid, block, smem, stream>>>(p, nf16);
+}

@@ -38,9 +38,10 @@ class ModelTrainer:
         self.data_container = data_container
         self.ctx = dist_ctx or DistContext()
         self.device = torch.device(params.get("device", params.get("GPU", "cpu")))
-        cd = params.get("compute_dtype", "float32")
+        cd = str(params.get("compute_dtype", "float32"))
+        self.fp8_forward = cd == "fp8"  # fp8 forward / bf16 backward
         self.compute_dtype = (
-            torch.bfloat16 if str(cd) in ("bf16", "bfloat16", "torch.bfloat16")
+            torch.bfloat16 if cd in ("bf16", "bfloat16", "torch.bfloat16", "fp8")
             else torch.float32
         )
 
@@ -85,6 +86,7 @@ class ModelTrainer:
             activation="relu",
             compute_dtype=self.compute_dtype,
             fusion=self.params.get("fusion", "mean"),
+            fp8_forward=self.fp8_forward,
         )
 
     def get_loss(self):

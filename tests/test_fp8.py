@@ -1,0 +1,125 @@
+"""fp8-forward / bf16-backward training mode (GPU).
+
+Numerics: e4m3 quantization carries ~0.4-6% per-element relative error, so
+the fp8 forward is compared against the f32 eager oracle with quantization-
+scale tolerances, and gradients are checked for direction (cosine vs the
+oracle) rather than elementwise equality — the backward runs the exact bf16
+kernels on the fp8-forward's saved operands (straight-through/QAT semantics).
+Convergence at the flagship config is recorded in profiles/FP8.md.
+"""
+
+import pytest
+import torch
+
+from mpgcn_amd.graph import build_supports
+from mpgcn_amd.models import MPGCN
+from mpgcn_amd.ops import GraphOperator, bdgcn_layer_fp8, eager
+
+pytestmark = pytest.mark.gpu
+
+N, S, C, H, B = 256, 3, 32, 32, 2
+
+
+def _layer_inputs(dyn=False):
+    torch.manual_seed(0)
+    dev = "cuda:0"
+    X = (torch.rand(B, N, N, C, device=dev) - 0.3).bfloat16()
+    nb = B if dyn else 1
+    Go = build_supports(torch.rand(nb, N, N, device=dev),
+                        "random_walk_diffusion", S - 1)
+    Gd = build_supports(torch.rand(nb, N, N, device=dev),
+                        "random_walk_diffusion", S - 1)
+    if not dyn:
+        Go, Gd = Go.squeeze(0), Gd.squeeze(0)
+    Go, Gd = Go.bfloat16().contiguous(), Gd.bfloat16().contiguous()
+    W = (0.1 * torch.randn(C * S * S, H, device=dev)).bfloat16()
+    b = 0.05 * torch.randn(H, device=dev)
+    return X, Go, Gd, W, b
+
+
+@pytest.mark.parametrize("dyn", [False, True])
+def test_fp8_layer_forward_close_to_f32_eager(dyn):
+    X, Go, Gd, W, b = _layer_inputs(dyn)
+    gop = GraphOperator(Go, Gd)
+    Y, Y8 = bdgcn_layer_fp8(X, W, b, gop, relu=True)
+    ref = eager.bdgcn_layer_eager(
+        X.float(), Go.float(), Gd.float(), W.float(), b.float(), "relu"
+    )
+    rel = (Y.float() - ref).norm() / ref.norm()
+    assert rel < 0.05, f"fp8 forward rel err {rel:.4f}"
+    # the fp8 twin must be the quantized output
+    rel8 = (Y8.float() - Y.float()).norm() / (Y.float().norm() + 1e-9)
+    assert rel8 < 0.05, f"fp8 twin rel err {rel8:.4f}"
+
+
+def test_fp8_layer_gradients_aligned_with_oracle():
+    X, Go, Gd, W, b = _layer_inputs()
+    gop = GraphOperator(Go, Gd)
+    Xn = X.clone().requires_grad_(True)
+    Wn = W.clone().requires_grad_(True)
+    bn = b.clone().requires_grad_(True)
+    Y, _ = bdgcn_layer_fp8(Xn, Wn, bn, gop, relu=True)
+    Y.square().sum().backward()
+
+    Xe = X.float().detach().requires_grad_(True)
+    We = W.float().detach().requires_grad_(True)
+    be = b.float().detach().requires_grad_(True)
+    ref = eager.bdgcn_layer_eager(Xe, Go.float(), Gd.float(), We, be, "relu")
+    ref.square().sum().backward()
+
+    for g, ge, name in ((Xn.grad.float(), Xe.grad, "dX"),
+                        (Wn.grad.float(), We.grad, "dW"),
+                        (bn.grad.float(), be.grad, "db")):
+        cos = torch.nn.functional.cosine_similarity(
+            g.flatten(), ge.flatten(), dim=0
+        ).item()
+        assert cos > 0.98, f"{name} cosine {cos:.4f}"
+
+
+def test_fp8_twin_chains_between_layers():
+    """Layer 2 consuming layer 1's Y8 twin must match layer 2 quantizing
+    layer 1's bf16 Y itself (the chained twin IS the quantized output)."""
+    X, Go, Gd, W, b = _layer_inputs()
+    gop = GraphOperator(Go, Gd)
+    W2 = (0.1 * torch.randn(H * S * S, H, device=X.device)).bfloat16()
+    Y1, Y18 = bdgcn_layer_fp8(X, W, b, gop, relu=True)
+    Y2_chained, _ = bdgcn_layer_fp8(Y1, W2, None, gop, relu=True, X8=Y18)
+    Y2_requant, _ = bdgcn_layer_fp8(Y1.detach().clone(), W2, None, gop, relu=True)
+    torch.testing.assert_close(Y2_chained, Y2_requant, atol=1e-2, rtol=1e-2)
+
+
+def test_fp8_model_train_step_loss_decreases():
+    torch.manual_seed(3)
+    dev = "cuda:0"
+    model = MPGCN(M=2, K=S, input_dim=1, lstm_hidden_dim=H, lstm_num_layers=1,
+                  gcn_hidden_dim=H, gcn_num_layers=3, num_nodes=N,
+                  compute_dtype=torch.bfloat16, fp8_forward=True).to(dev)
+    opt = torch.optim.Adam(model.parameters(), lr=1e-3)
+    x = torch.rand(B, 7, N, N, 1, device=dev)
+    y = torch.rand(B, 1, N, N, 1, device=dev)
+    flow = torch.rand(B, N, N, device=dev)
+    Gs = build_supports(torch.rand(1, N, N, device=dev),
+                        "random_walk_diffusion", S - 1)[0]
+    Go = build_supports(flow, "random_walk_diffusion", S - 1)
+    Gd = build_supports(flow.transpose(-2, -1), "random_walk_diffusion", S - 1)
+    losses = []
+    for _ in range(8):
+        out = model(x, [Gs, (Go, Gd)])
+        loss = torch.nn.functional.mse_loss(out, y)
+        opt.zero_grad(set_to_none=True)
+        loss.backward()
+        opt.step()
+        losses.append(loss.item())
+    assert all(torch.isfinite(torch.tensor(losses)))
+    assert losses[-1] < losses[0], losses
+
+
+def test_fp8_shape_gate_raises():
+    with pytest.raises(ValueError, match="fp8_forward shape gate"):
+        MPGCN(M=2, K=S, input_dim=1, lstm_hidden_dim=24, lstm_num_layers=1,
+              gcn_hidden_dim=24, gcn_num_layers=3, num_nodes=100,
+              compute_dtype=torch.bfloat16, fp8_forward=True)
+    with pytest.raises(ValueError, match="requires compute_dtype=bf16"):
+        MPGCN(M=2, K=S, input_dim=1, lstm_hidden_dim=H, lstm_num_layers=1,
+              gcn_hidden_dim=H, gcn_num_layers=3, num_nodes=N,
+              compute_dtype=torch.float32, fp8_forward=True)
