@@ -34,15 +34,17 @@ class BDGCN(nn.Module):
             self.register_parameter("b", None)
 
     def forward(self, X: torch.Tensor, gop: GraphOperator, fp8: bool = False,
-                X8: torch.Tensor | None = None):
+                X8: torch.Tensor | None = None, emit_twin: bool = True):
         """X: (B, N, N, input_dim) -> (B, N, N, hidden_dim).
 
         fp8=True runs the fp8-forward/bf16-backward path and returns
         (Y, Y8_twin) so the caller can thread the fp8 twin into the next
-        layer without a quantize pass (ops/functional.py)."""
+        layer without a quantize pass (ops/functional.py); emit_twin=False
+        (last layer) returns Y alone and skips the twin write."""
         W = self.W.to(X.dtype if X.dtype != torch.float8_e4m3fn else torch.bfloat16)
         if fp8:
-            return bdgcn_layer_fp8(X, W, self.b, gop, relu=self.relu, X8=X8)
+            return bdgcn_layer_fp8(X, W, self.b, gop, relu=self.relu, X8=X8,
+                                   emit_twin=emit_twin)
         return bdgcn_layer(X, W, self.b, gop, relu=self.relu)
 
     def extra_repr(self) -> str:

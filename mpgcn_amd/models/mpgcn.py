@@ -163,10 +163,15 @@ class MPGCN(nn.Module):
             X = h_last.reshape(B, N, N, self.lstm_hidden_dim)
             if fp8:
                 # fp8 twins chain layer-to-layer through the dual-write
-                # epilogues; only the LSTM output needs a standalone cast
+                # epilogues; only the LSTM output needs a standalone cast,
+                # and the last layer (bf16 FC consumer) skips its twin
                 X8 = None
-                for layer in branch["spatial"]:
-                    X, X8 = layer(X, gops[m], fp8=True, X8=X8)
+                n_sp = len(branch["spatial"])
+                for i, layer in enumerate(branch["spatial"]):
+                    if i + 1 < n_sp:
+                        X, X8 = layer(X, gops[m], fp8=True, X8=X8)
+                    else:
+                        X = layer(X, gops[m], fp8=True, X8=X8, emit_twin=False)
             else:
                 for layer in branch["spatial"]:
                     X = layer(X, gops[m])

@@ -192,7 +192,8 @@ class _BDGCNLayerFp8Fn(torch.autograd.Function):
     """
 
     @staticmethod
-    def forward(ctx, X, W, bias, gop: GraphOperator, relu: bool, X8):
+    def forward(ctx, X, W, bias, gop: GraphOperator, relu: bool, X8,
+                emit_twin: bool):
         ext = _ops.get_ext()
         B, N = X.shape[0], X.shape[1]
         C = X.shape[-1]
@@ -200,26 +201,32 @@ class _BDGCNLayerFp8Fn(torch.autograd.Function):
         Hdim = W.shape[1]
         if X8 is None:
             X8 = X.to(torch.float8_e4m3fn)
-        U8, U = ext.bdgcn_mode1_fp8_train(X8, gop.GoT8)
+        # U stays fp8-ONLY: mode-1 writes half the bf16 path's output bytes
+        # (U is the step's largest tensor), and backward's dW reduction reads
+        # the fp8 U directly (red_gemm y_fp8 staging)
+        U8 = ext.bdgcn_mode1_fp8_train(X8, gop.GoT8)
         Wre = eager.reorder_projection_weight(W, S, C).contiguous()
         Wre8 = Wre.to(torch.float8_e4m3fn)
         V8 = ext.row_gemm_fp8(U8.reshape(B * N * N, S * C), Wre8)
         bias_f32 = bias.float().contiguous() if bias is not None else None
         Y, Y8 = ext.bdgcn_mode2_fp8_train(
-            V8.view(B, N, N * S, Hdim), gop.A2T8, bias_f32, relu, N, S
+            V8.view(B, N, N * S, Hdim), gop.A2T8, bias_f32, relu, N, S,
+            emit_twin,
         )
-        ctx.save_for_backward(U, Wre, Y)
+        ctx.save_for_backward(U8, Wre, Y)
         ctx.gop = gop
         ctx.relu = relu
         ctx.has_bias = bias is not None
         ctx.dims = (B, N, S, C, Hdim)
-        ctx.mark_non_differentiable(Y8)
-        return Y, Y8
+        if emit_twin:
+            ctx.mark_non_differentiable(Y8)
+            return Y, Y8
+        return Y
 
     @staticmethod
-    def backward(ctx, dH, _dY8):
+    def backward(ctx, dH, _dY8=None):
         dX, dW, db, _, _ = _BDGCNLayerFn.backward(ctx, dH)
-        return dX, dW, db, None, None, None
+        return dX, dW, db, None, None, None, None
 
 
 def fp8_forward_compatible(N: int, C: int, Hdim: int, S: int) -> bool:
@@ -229,12 +236,14 @@ def fp8_forward_compatible(N: int, C: int, Hdim: int, S: int) -> bool:
 
 
 def bdgcn_layer_fp8(X, W, bias, gop: GraphOperator, relu: bool = True,
-                    X8=None):
-    """fp8-forward BDGCN layer: (Y_bf16, Y8_twin). GPU-only; callers gate on
-    fp8_forward_compatible and fall back to bdgcn_layer otherwise."""
+                    X8=None, emit_twin: bool = True):
+    """fp8-forward BDGCN layer: (Y_bf16, Y8_twin), or Y_bf16 alone with
+    emit_twin=False (last layer — its consumer is the bf16 FC head).
+    GPU-only; callers gate on fp8_forward_compatible and fall back to
+    bdgcn_layer otherwise."""
     if not X.is_cuda:
         raise RuntimeError("fp8-forward mode requires a GPU")
-    return _BDGCNLayerFp8Fn.apply(X, W, bias, gop, relu, X8)
+    return _BDGCNLayerFp8Fn.apply(X, W, bias, gop, relu, X8, emit_twin)
 
 
 class _Mode1ProjFn(torch.autograd.Function):

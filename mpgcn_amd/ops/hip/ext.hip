@@ -233,7 +233,14 @@ std::vector<torch::Tensor> red_gemm(torch::Tensor X, torch::Tensor Y,
                                     c10::optional<torch::Tensor> xvec,
                                     long xv_stride, long xv_off) {
     check_in(X, "X");
-    check_in(Y, "Y");
+    const bool y8 = Y.scalar_type() == torch::kFloat8_e4m3fn;
+    if (y8) {
+        TORCH_CHECK(Y.is_cuda() && Y.is_contiguous(), "Y must be contiguous CUDA");
+        TORCH_CHECK(X.scalar_type() == torch::kBFloat16,
+                    "fp8 Y operand requires bf16 X");
+    } else {
+        check_in(Y, "Y");
+    }
     const long R = X.size(0), K = X.size(1), N = Y.size(1);
     TORCH_CHECK(Y.size(0) == R, "row mismatch");
     auto f32 = X.options().dtype(torch::kFloat);
@@ -258,6 +265,7 @@ std::vector<torch::Tensor> red_gemm(torch::Tensor X, torch::Tensor Y,
     p.xdot = has_xv ? xdot.data_ptr<float>() : nullptr;
     p.R = R; p.K = (int)K; p.N = (int)N;
     p.det = det ? 1 : 0;
+    p.y_fp8 = y8 ? 1 : 0;
     const int ch = chunk_elems(X);
     p.x_vec = (K % ch == 0);
     p.y_vec = (N % ch == 0);
@@ -468,16 +476,17 @@ void check_fp8(const torch::Tensor& t, const char* name) {
 }
 }  // namespace
 
-// fp8-forward training mode-1: U = mode1(X8, GT8) with DUAL outputs —
-// U8 (fp8, feeds the fp8 projection GEMM) and U_bf16 (the saved backward
-// operand for the dW reduction). Forward math runs entirely on fp8 operands
-// (half the staged bytes through the byte-bound load path — the measured
-// 1.30x kernel lever); backward stays bf16. Shape contract: the fp8 tile is
-// vector-only, so the column extent Nd*C must be a multiple of 256 and C a
-// multiple of 16 (flagship and large-N configs satisfy this; the Python
+// fp8-forward training mode-1: U8 = mode1(X8, GT8), fp8 OUTPUT ONLY — U8
+// both feeds the fp8 projection GEMM and is the saved backward operand (the
+// dW reduction red_gemm reads fp8 Y directly), so mode-1's output traffic is
+// HALF the bf16 path's (U is the largest tensor in the step) and no bf16
+// twin is written. Forward math runs entirely on fp8 operands (half the
+// staged bytes through the byte-bound load path — the measured 1.30x kernel
+// lever); the gradient CONTRACTIONS stay bf16. Shape contract: the fp8 tile
+// is vector-only, so the column extent Nd*C must be a multiple of 256 and C
+// a multiple of 16 (flagship and large-N configs satisfy this; the Python
 // layer falls back to bf16 otherwise).
-std::vector<torch::Tensor> bdgcn_mode1_fp8_train(torch::Tensor X8,
-                                                 torch::Tensor GT8) {
+torch::Tensor bdgcn_mode1_fp8_train(torch::Tensor X8, torch::Tensor GT8) {
     check_fp8(X8, "X8");
     check_fp8(GT8, "GT8");
     const bool dyn = GT8.dim() == 4;
@@ -489,14 +498,11 @@ std::vector<torch::Tensor> bdgcn_mode1_fp8_train(torch::Tensor X8,
     TORCH_CHECK((Nd * C) % 256 == 0 && C % 16 == 0,
                 "fp8 mode-1 needs (Nd*C) % 256 == 0 and C % 16 == 0");
     auto U8 = torch::empty({B, No, Nd, S, C}, X8.options());
-    auto Ubf = torch::empty({B, No, Nd, S, C},
-                            X8.options().dtype(torch::kBFloat16));
 
     AxisGemmParams p{};
     p.AT = GT8.data_ptr();
     p.X = X8.data_ptr();
     p.OUT = U8.data_ptr();
-    p.OUT2 = Ubf.data_ptr();
     p.M = (int)No; p.K = (int)No; p.L = (int)(Nd * C);
     p.a_div = (int)S; p.a_bs1 = dyn ? S * No * No : 0; p.a_bs2 = No * No;
     p.x_div = (int)S; p.x_bs1 = No * Nd * C; p.x_bs2 = 0;
@@ -507,8 +513,8 @@ std::vector<torch::Tensor> bdgcn_mode1_fp8_train(torch::Tensor X8,
     p.ogdiv = (int)C; p.og_hi = S * C;
     p.a_vec = (No % 16 == 0);
     p.x_vec = 1;
-    axis_gemm_fp8_launch(p, (int)(B * S), 1, stream());
-    return {U8, Ubf};
+    axis_gemm_fp8_launch(p, (int)(B * S), 0, stream());
+    return U8;
 }
 
 // fp8-forward training mode-2: Y = mode2(V8, A2T8) + bias + act with DUAL
@@ -517,7 +523,8 @@ std::vector<torch::Tensor> bdgcn_mode1_fp8_train(torch::Tensor X8,
 std::vector<torch::Tensor> bdgcn_mode2_fp8_train(torch::Tensor V8,
                                                  torch::Tensor A2T8,
                                                  c10::optional<torch::Tensor> bias,
-                                                 bool relu, long N, long S) {
+                                                 bool relu, long N, long S,
+                                                 bool want_twin) {
     check_fp8(V8, "V8");
     check_fp8(A2T8, "A2T8");
     const bool dyn = A2T8.dim() == 3;
@@ -527,13 +534,16 @@ std::vector<torch::Tensor> bdgcn_mode2_fp8_train(torch::Tensor V8,
     TORCH_CHECK((Nm * H) % 256 == 0 && H % 16 == 0,
                 "fp8 mode-2 needs (Nm*H) % 256 == 0 and H % 16 == 0");
     auto Y = torch::empty({B, Nm, N, H}, V8.options().dtype(torch::kBFloat16));
-    auto Y8 = torch::empty({B, Nm, N, H}, V8.options());
+    // want_twin=false (e.g. the LAST gcn layer, whose output feeds the bf16
+    // FC head) skips the fp8 twin write entirely
+    auto Y8 = want_twin ? torch::empty({B, Nm, N, H}, V8.options())
+                        : torch::Tensor();
 
     AxisGemmParams p{};
     p.AT = A2T8.data_ptr();
     p.X = V8.data_ptr();
     p.OUT = Y.data_ptr();
-    p.OUT2 = Y8.data_ptr();
+    p.OUT2 = want_twin ? Y8.data_ptr() : nullptr;
     p.bias = bias_ptr(bias);
     p.M = (int)N; p.K = (int)(N * S); p.L = (int)(Nm * H);
     p.a_div = 1; p.a_bs1 = dyn ? N * N * S : 0; p.a_bs2 = 0;

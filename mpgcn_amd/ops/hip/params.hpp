@@ -74,6 +74,7 @@ struct RedGemmParams {
     long R;
     int K, N;
     int x_vec, y_vec;
+    int y_fp8;          // Y operand is e4m3 (fp8-forward mode's saved U8)
     int det;            // deterministic: out/colsum/xdot are per-block
                         // workspaces (nblocks, ...) written with plain stores
 };

@@ -16,7 +16,13 @@
 #include "common.hpp"
 #include "params.hpp"
 
-template <typename T, int AKF, int ANF>
+typedef __attribute__((ext_vector_type(2))) float f32x2;
+
+// YF8: the Y operand is e4m3 fp8 in global memory (the fp8-forward mode's
+// U8 tensor — saving it fp8-only halves both the mode-1 write traffic and
+// this kernel's Y read bytes); converted to bf16 during LDS staging, MFMA
+// unchanged. v_cvt_pk_f32_fp8 unpacks byte pairs.
+template <typename T, int AKF, int ANF, bool YF8 = false>
 __launch_bounds__(256) __global__ void red_gemm_kernel(RedGemmParams p) {
     using MT = MfmaTraits<T>;
     constexpr int CH = 16 / sizeof(T);
@@ -73,11 +79,34 @@ __launch_bounds__(256) __global__ void red_gemm_kernel(RedGemmParams p) {
             alignas(16) T tmp[CH];
             for (int i = 0; i < CH; ++i) tmp[i] = (T)0.f;
             if (row < p.R) {
-                if (p.y_vec && n0 + CH <= p.N)
+                if constexpr (YF8) {
+                    const unsigned char* Y8p = (const unsigned char*)p.Y;
+                    if (p.y_vec && n0 + CH <= p.N) {
+                        // CH(=8) fp8 bytes -> 8 bf16 via packed converts
+                        unsigned long long raw;
+                        __builtin_memcpy(&raw, &Y8p[row * p.N + n0], 8);
+                        const int lo = (int)raw, hi = (int)(raw >> 32);
+                        const f32x2 f0 = __builtin_amdgcn_cvt_pk_f32_fp8(lo, false);
+                        const f32x2 f1 = __builtin_amdgcn_cvt_pk_f32_fp8(lo, true);
+                        const f32x2 f2 = __builtin_amdgcn_cvt_pk_f32_fp8(hi, false);
+                        const f32x2 f3 = __builtin_amdgcn_cvt_pk_f32_fp8(hi, true);
+                        tmp[0] = (T)f0[0]; tmp[1] = (T)f0[1];
+                        tmp[2] = (T)f1[0]; tmp[3] = (T)f1[1];
+                        tmp[4] = (T)f2[0]; tmp[5] = (T)f2[1];
+                        tmp[6] = (T)f3[0]; tmp[7] = (T)f3[1];
+                    } else {
+                        for (int i = 0; i < CH; ++i)
+                            if (n0 + i < p.N) {
+                                const int b = Y8p[row * p.N + n0 + i];
+                                tmp[i] = (T)__builtin_amdgcn_cvt_pk_f32_fp8(b, false)[0];
+                            }
+                    }
+                } else if (p.y_vec && n0 + CH <= p.N) {
                     *(Chunk16*)tmp = *(const Chunk16*)&Y[row * p.N + n0];
-                else
+                } else {
                     for (int i = 0; i < CH; ++i)
                         if (n0 + i < p.N) tmp[i] = Y[row * p.N + n0 + i];
+                }
             }
 #pragma unroll
             for (int i = 0; i < CH; ++i) {
@@ -176,20 +205,21 @@ extern "C" long red_gemm_nblocks(long R) {
 extern "C" void red_gemm_launch(RedGemmParams p, int is_f32, hipStream_t s) {
     long blocks = red_gemm_nblocks(p.R);
     dim3 grid((unsigned)blocks), block(256);
-#define DISPATCH(TT)                                                        \
+#define DISPATCH(TT, YF)                                                    \
     do {                                                                    \
         if (p.K <= 96 && p.N <= 96)                                         \
-            red_gemm_kernel<TT, 3, 3><<<grid, block, 0, s>>>(p);            \
+            red_gemm_kernel<TT, 3, 3, YF><<<grid, block, 0, s>>>(p);        \
         else if (p.K <= 128 && p.N <= 32)                                   \
-            red_gemm_kernel<TT, 4, 1><<<grid, block, 0, s>>>(p);            \
+            red_gemm_kernel<TT, 4, 1, YF><<<grid, block, 0, s>>>(p);        \
         else if (p.K <= 160 && p.N <= 160)                                  \
-            red_gemm_kernel<TT, 5, 5><<<grid, block, 0, s>>>(p);            \
+            red_gemm_kernel<TT, 5, 5, YF><<<grid, block, 0, s>>>(p);        \
         else {                                                              \
             fprintf(stderr, "red_gemm: K=%d N=%d unsupported\n", p.K, p.N); \
             abort();                                                        \
         }                                                                   \
     } while (0)
-    if (!is_f32) DISPATCH(__bf16);
-    else DISPATCH(float);
+    if (p.y_fp8) DISPATCH(__bf16, true);
+    else if (!is_f32) DISPATCH(__bf16, false);
+    else DISPATCH(float, false);
 #undef DISPATCH
 }
