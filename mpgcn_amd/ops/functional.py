@@ -44,6 +44,8 @@ class GraphOperator:
         self._A3T = None
         self._GoT8 = None
         self._A2T8 = None
+        self._A28 = None
+        self._A3T8 = None
 
     @property
     def GoT(self) -> torch.Tensor:
@@ -76,6 +78,18 @@ class GraphOperator:
         if self._A2T8 is None:
             self._A2T8 = self.A2T.to(torch.float8_e4m3fn)
         return self._A2T8
+
+    @property
+    def A28(self) -> torch.Tensor:
+        if self._A28 is None:
+            self._A28 = self.A2.to(torch.float8_e4m3fn)
+        return self._A28
+
+    @property
+    def A3T8(self) -> torch.Tensor:
+        if self._A3T8 is None:
+            self._A3T8 = self.A3T.to(torch.float8_e4m3fn)
+        return self._A3T8
 
     @property
     def A2(self) -> torch.Tensor:
@@ -176,6 +190,18 @@ def bdgcn_layer(X, W, bias, gop: GraphOperator, relu: bool = True):
     return eager.bdgcn_layer_eager(X, gop.Go, gop.Gd, W, bias, "relu" if relu else "none")
 
 
+def _scaled_fp8(t: torch.Tensor, margin: float = 224.0):
+    """Dynamic per-tensor fp8 quantization for gradients: returns
+    (fp8(t * s), 1/s as a device f32 scalar). s = margin/amax keeps the
+    tensor inside e4m3's range; everything stays on device (no host sync).
+    margin = half of e4m3 max (448) leaves headroom for the bf16 rounding
+    of the scale multiply."""
+    amax = t.abs().amax().float().clamp(min=1e-20)
+    scale = (margin / amax).to(torch.bfloat16).float()  # quantize-side value
+    t8 = (t * scale.to(t.dtype)).to(torch.float8_e4m3fn)
+    return t8, scale.reciprocal()
+
+
 class _BDGCNLayerFp8Fn(torch.autograd.Function):
     """fp8-forward / bf16-backward BDGCN layer.
 
@@ -225,8 +251,34 @@ class _BDGCNLayerFp8Fn(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dH, _dY8=None):
-        dX, dW, db, _, _ = _BDGCNLayerFn.backward(ctx, dH)
-        return dX, dW, db, None, None, None, None
+        # Scaled-fp8 gradient contractions: gradients live well below e4m3's
+        # 2^-9 subnormal floor, so each quantize uses a DEVICE-resident
+        # dynamic scale s = margin/amax (no host sync) and the kernel
+        # epilogue multiplies by 1/s. Weight-gradient reductions (red_gemm)
+        # and the dU projection stay bf16 — only the two big axis
+        # contractions (the byte-bound kernels) run on fp8 operands.
+        ext = _ops.get_ext()
+        U8, Wre, Y = ctx.saved_tensors
+        gop: GraphOperator = ctx.gop
+        B, N, S, C, Hdim = ctx.dims
+
+        dH = dH.contiguous()
+        dY, dbias = ext.relu_bwd_colsum(dH, Y, ctx.relu)
+        dY = dY.view_as(dH)
+        if not ctx.has_bias:
+            dbias = None
+
+        dY8, inv_y = _scaled_fp8(dY)
+        dV = ext.bdgcn_mode2_bwd_fp8(dY8, gop.A28, S, inv_y)  # (B,N,N,S,H)
+        R = B * N * N
+        dVflat = dV.reshape(R, S * Hdim)
+        dWreT, _, _ = ext.red_gemm(dVflat, U8.reshape(R, S * C), False, None, 0, 0)
+        dWre = dWreT.t().to(dH.dtype)
+        dW = dWre.reshape(S, C, S, Hdim).permute(0, 2, 1, 3).reshape(S * S * C, Hdim)
+        dU = _row_gemm_chunked(ext, dVflat, Wre.t().contiguous(), None, False)
+        dU8, inv_u = _scaled_fp8(dU)
+        dX = ext.bdgcn_mode1_bwd_fp8(dU8.view(B, N, N, S, C), gop.A3T8, inv_u)
+        return dX, dW, dbias, None, None, None, None
 
 
 def fp8_forward_compatible(N: int, C: int, Hdim: int, S: int) -> bool:

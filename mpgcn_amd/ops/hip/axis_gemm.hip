@@ -158,9 +158,10 @@ __launch_bounds__(WVM * WVN * 64) __global__ void axis_gemm_kernel(AxisGemmParam
         if (BUFS == 2) cur ^= 1;
     }
 
-    // ---- epilogue: bias + activation + strided store ----
+    // ---- epilogue: scale + bias + activation + strided store ----
     // per-lane q depends only on nf: hoist the column-offset division and the
     // bias load out of the (mf, r) loops
+    const float osc = p.scale ? *p.scale : 1.f;  // fp8 gradient descale
 #pragma unroll
     for (int nf = 0; nf < AN; ++nf) {
         const int q = l0 + wn + nf * 16 + lrow;
@@ -173,7 +174,7 @@ __launch_bounds__(WVM * WVN * 64) __global__ void axis_gemm_kernel(AxisGemmParam
             for (int r = 0; r < 4; ++r) {
                 const int m = m0 + wm + mf * 16 + kgrp * 4 + r;
                 if (m < p.M) {
-                    float v = acc[mf][nf][r] + bv;
+                    float v = acc[mf][nf][r] * osc + bv;
                     if (p.relu) v = fmaxf(v, 0.f);
                     O[oc + (long)m * p.o_row] = from_f32<OT>(v);
                     if (p.OUT2) O2[oc + (long)m * p.o_row] = from_f32<OT2>(v);

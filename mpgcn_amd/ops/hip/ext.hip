@@ -561,6 +561,79 @@ std::vector<torch::Tensor> bdgcn_mode2_fp8_train(torch::Tensor V8,
     return {Y, Y8};
 }
 
+// fp8 gradient contraction dV = mode2_bwd(dY8, A28), SCALED: dY was
+// quantized as dY8 = fp8(dY * s) with a device-resident dynamic scale
+// (s = margin/amax(dY) — gradients underflow e4m3's 2^-9 floor without it);
+// the epilogue multiplies by *inv_scale (device pointer, no host sync) and
+// writes bf16 dV. A28[cs, d] = fp8(Gd[s, c, d]) — supports are O(1), unit
+// scale. Same contraction as bdgcn_mode2_bwd (bf16 twin of this path).
+torch::Tensor bdgcn_mode2_bwd_fp8(torch::Tensor dY8, torch::Tensor A28,
+                                  long S, torch::Tensor inv_scale) {
+    check_fp8(dY8, "dY8");
+    check_fp8(A28, "A28");
+    TORCH_CHECK(inv_scale.is_cuda() && inv_scale.scalar_type() == torch::kFloat,
+                "inv_scale must be a CUDA f32 scalar tensor");
+    const bool dyn = A28.dim() == 3;
+    const long B = dY8.size(0), Nm = dY8.size(1), H = dY8.size(3);
+    const long N = A28.size(-1);
+    TORCH_CHECK(A28.size(-2) == N * S && dY8.size(2) == N, "A28 shape");
+    TORCH_CHECK(B <= 65535, "too many instances");
+    TORCH_CHECK((Nm * H) % 256 == 0 && H % 16 == 0, "fp8 bwd shape gate");
+    auto dV = torch::empty({B, Nm, N, S, H},
+                           dY8.options().dtype(torch::kBFloat16));
+    AxisGemmParams p{};
+    p.AT = A28.data_ptr();
+    p.X = dY8.data_ptr();
+    p.OUT = dV.data_ptr();
+    p.scale = inv_scale.data_ptr<float>();
+    p.M = (int)(N * S); p.K = (int)N; p.L = (int)(Nm * H);
+    p.a_div = 1; p.a_bs1 = dyn ? N * S * N : 0; p.a_bs2 = 0;
+    p.x_div = 1; p.x_bs1 = Nm * N * H; p.x_bs2 = 0;
+    p.o_div = 1; p.o_bs1 = Nm * N * S * H; p.o_bs2 = 0;
+    p.kdiv = 1; p.k_lo = H;
+    p.qdiv = (int)H; p.q_hi = N * H;
+    p.o_row = H;
+    p.ogdiv = (int)H; p.og_hi = N * S * H;
+    p.a_vec = (N % 16 == 0);
+    p.x_vec = 1;
+    axis_gemm_fp8_launch(p, (int)B, 2, stream());
+    return dV;
+}
+
+// fp8 gradient contraction dX = mode1_bwd(dU8, A3T8), scaled like
+// bdgcn_mode2_bwd_fp8. dU8 = fp8(dU * s); epilogue descales and writes bf16.
+torch::Tensor bdgcn_mode1_bwd_fp8(torch::Tensor dU8, torch::Tensor A3T8,
+                                  torch::Tensor inv_scale) {
+    check_fp8(dU8, "dU8");
+    check_fp8(A3T8, "A3T8");
+    TORCH_CHECK(inv_scale.is_cuda() && inv_scale.scalar_type() == torch::kFloat,
+                "inv_scale must be a CUDA f32 scalar tensor");
+    const bool dyn = A3T8.dim() == 3;
+    const long B = dU8.size(0), No = dU8.size(1), Nd = dU8.size(2);
+    const long S = dU8.size(3), C = dU8.size(4);
+    TORCH_CHECK(A3T8.size(-2) == No && A3T8.size(-1) == S * No, "A3T8 shape");
+    TORCH_CHECK(B <= 65535, "too many instances");
+    TORCH_CHECK((Nd * C) % 256 == 0 && C % 16 == 0, "fp8 bwd shape gate");
+    auto dX = torch::empty({B, No, Nd, C}, dU8.options().dtype(torch::kBFloat16));
+    AxisGemmParams p{};
+    p.AT = A3T8.data_ptr();
+    p.X = dU8.data_ptr();
+    p.OUT = dX.data_ptr();
+    p.scale = inv_scale.data_ptr<float>();
+    p.M = (int)No; p.K = (int)(S * No); p.L = (int)(Nd * C);
+    p.a_div = 1; p.a_bs1 = dyn ? No * S * No : 0; p.a_bs2 = 0;
+    p.x_div = 1; p.x_bs1 = No * Nd * S * C; p.x_bs2 = 0;
+    p.o_div = 1; p.o_bs1 = No * Nd * C; p.o_bs2 = 0;
+    p.kdiv = (int)No; p.k_hi = C; p.k_lo = Nd * S * C;  // k = o*No + m
+    p.qdiv = (int)C; p.q_hi = S * C;                    // q = d*C + l
+    p.o_row = Nd * C;
+    p.ogdiv = 0;
+    p.a_vec = ((S * No) % 16 == 0);
+    p.x_vec = 1;
+    axis_gemm_fp8_launch(p, (int)B, 2, stream());
+    return dX;
+}
+
 // fp8 projection GEMM: OUT8[r, n] = X8[r, :] @ W8 — the fp8-forward mode's
 // V = U @ Wre (output feeds fp8 mode-2; V is not needed by backward, so no
 // bf16 twin). K must be a multiple of 16 (vectorized fp8 row reads).
@@ -632,6 +705,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("bdgcn_mode2_fp8_train", &bdgcn_mode2_fp8_train,
           "fp8-forward mode-2 + bias + act, bf16 out + fp8 twin");
     m.def("row_gemm_fp8", &row_gemm_fp8, "fp8 projection GEMM");
+    m.def("bdgcn_mode2_bwd_fp8", &bdgcn_mode2_bwd_fp8,
+          "scaled fp8 gradient contraction dV");
+    m.def("bdgcn_mode1_bwd_fp8", &bdgcn_mode1_bwd_fp8,
+          "scaled fp8 gradient contraction dX");
     m.def("row_gemm_out", &row_gemm_out, "row GEMM into strided output slice");
     m.def("lstm_step_fwd", &lstm_step_fwd, "fused LSTM cell forward step (K4)");
     m.def("lstm_fused_fwd", &lstm_fused_fwd, "register-resident fused LSTM forward");

@@ -10,6 +10,8 @@ struct AxisGemmParams {
     void* OUT;
     void* OUT2;  // optional second epilogue copy (fp8/bf16 twin); nullable
     const float* bias;
+    const float* scale;  // optional epilogue multiplier (DEVICE pointer —
+                         // fp8 gradient descale, no host sync); nullable
     int M, K, L;
     int a_div; long a_bs1, a_bs2;
     int x_div; long x_bs1, x_bs2;
