@@ -13,6 +13,7 @@ import torch
 from torch import nn
 
 from mpgcn_amd.ops import GraphOperator, bdgcn_layer, bdgcn_layer_fp8
+from mpgcn_amd.ops.functional import make_fp8_state
 
 
 class BDGCN(nn.Module):
@@ -43,8 +44,11 @@ class BDGCN(nn.Module):
         (last layer) returns Y alone and skips the twin write."""
         W = self.W.to(X.dtype if X.dtype != torch.float8_e4m3fn else torch.bfloat16)
         if fp8:
+            st = getattr(self, "_fp8_state", None)
+            if st is None or st["amax_u"].device != X.device:
+                st = self._fp8_state = make_fp8_state(X.device)
             return bdgcn_layer_fp8(X, W, self.b, gop, relu=self.relu, X8=X8,
-                                   emit_twin=emit_twin)
+                                   emit_twin=emit_twin, fp8_state=st)
         return bdgcn_layer(X, W, self.b, gop, relu=self.relu)
 
     def extra_repr(self) -> str:

@@ -190,16 +190,20 @@ def bdgcn_layer(X, W, bias, gop: GraphOperator, relu: bool = True):
     return eager.bdgcn_layer_eager(X, gop.Go, gop.Gd, W, bias, "relu" if relu else "none")
 
 
-def _scaled_fp8(t: torch.Tensor, margin: float = 224.0):
-    """Dynamic per-tensor fp8 quantization for gradients: returns
-    (fp8(t * s), 1/s as a device f32 scalar). s = margin/amax keeps the
-    tensor inside e4m3's range; everything stays on device (no host sync).
-    margin = half of e4m3 max (448) leaves headroom for the bf16 rounding
-    of the scale multiply."""
-    amax = t.abs().amax().float().clamp(min=1e-20)
-    scale = (margin / amax).to(torch.bfloat16).float()  # quantize-side value
-    t8 = (t * scale.to(t.dtype)).to(torch.float8_e4m3fn)
-    return t8, scale.reciprocal()
+_FP8_MARGIN = 224.0  # half of e4m3 max: headroom over the recorded amax
+
+
+def make_fp8_state(device) -> dict:
+    """Per-layer delayed-scaling state for the fp8 dU path: amax recorded by
+    the producing kernel this step becomes next step's quantize scale
+    (device-resident; the whole schedule runs without host syncs). amax
+    bootstraps at 1.0 — the first step may clip outliers above the margin,
+    after which the scale tracks the real gradient magnitude."""
+    return {
+        "amax_u": torch.ones(1, device=device),
+        "scale_u": torch.full((1,), _FP8_MARGIN, device=device),
+        "inv_u": torch.full((1,), 1.0 / _FP8_MARGIN, device=device),
+    }
 
 
 class _BDGCNLayerFp8Fn(torch.autograd.Function):
@@ -219,7 +223,7 @@ class _BDGCNLayerFp8Fn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, X, W, bias, gop: GraphOperator, relu: bool, X8,
-                emit_twin: bool):
+                emit_twin: bool, fp8_state: dict):
         ext = _ops.get_ext()
         B, N = X.shape[0], X.shape[1]
         C = X.shape[-1]
@@ -244,6 +248,7 @@ class _BDGCNLayerFp8Fn(torch.autograd.Function):
         ctx.relu = relu
         ctx.has_bias = bias is not None
         ctx.dims = (B, N, S, C, Hdim)
+        ctx.fp8_state = fp8_state
         if emit_twin:
             ctx.mark_non_differentiable(Y8)
             return Y, Y8
@@ -252,33 +257,44 @@ class _BDGCNLayerFp8Fn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dH, _dY8=None):
         # Scaled-fp8 gradient contractions: gradients live well below e4m3's
-        # 2^-9 subnormal floor, so each quantize uses a DEVICE-resident
-        # dynamic scale s = margin/amax (no host sync) and the kernel
-        # epilogue multiplies by 1/s. Weight-gradient reductions (red_gemm)
-        # and the dU projection stay bf16 — only the two big axis
-        # contractions (the byte-bound kernels) run on fp8 operands.
+        # 2^-9 subnormal floor, so every quantize carries a DEVICE-resident
+        # dynamic scale and the axis-kernel epilogue descales — no host sync.
+        # Quantization is FUSED into the producers (a standalone
+        # amax+mul+cast chain costs more memory traffic than the fp8
+        # contraction saves — measured):
+        #   dY8: written by relu_bwd_colsum in its streaming pass, scale from
+        #        amax(|dH|) (a same-step upper bound of amax(|dY|));
+        #   dU8: the ONLY dU materialization — row_gemm emits scaled fp8
+        #        directly (its sole consumer is the fp8 dX contraction) with
+        #        a one-step-delayed amax (make_fp8_state).
+        # Weight-gradient reductions (red_gemm) stay bf16-accumulated-f32.
         ext = _ops.get_ext()
         U8, Wre, Y = ctx.saved_tensors
         gop: GraphOperator = ctx.gop
+        st = ctx.fp8_state
         B, N, S, C, Hdim = ctx.dims
 
         dH = dH.contiguous()
-        dY, dbias = ext.relu_bwd_colsum(dH, Y, ctx.relu)
-        dY = dY.view_as(dH)
+        amax_h = dH.abs().amax().float().clamp(min=1e-20)
+        scale_y = _FP8_MARGIN / amax_h
+        inv_y = amax_h / _FP8_MARGIN
+        dY, dY8, dbias = ext.relu_bwd_colsum_fp8(dH, Y, ctx.relu, scale_y)
         if not ctx.has_bias:
             dbias = None
 
-        dY8, inv_y = _scaled_fp8(dY)
         dV = ext.bdgcn_mode2_bwd_fp8(dY8, gop.A28, S, inv_y)  # (B,N,N,S,H)
         R = B * N * N
         dVflat = dV.reshape(R, S * Hdim)
         dWreT, _, _ = ext.red_gemm(dVflat, U8.reshape(R, S * C), False, None, 0, 0)
         dWre = dWreT.t().to(dH.dtype)
         dW = dWre.reshape(S, C, S, Hdim).permute(0, 2, 1, 3).reshape(S * S * C, Hdim)
-        dU = _row_gemm_chunked(ext, dVflat, Wre.t().contiguous(), None, False)
-        dU8, inv_u = _scaled_fp8(dU)
-        dX = ext.bdgcn_mode1_bwd_fp8(dU8.view(B, N, N, S, C), gop.A3T8, inv_u)
-        return dX, dW, dbias, None, None, None, None
+
+        ext.fp8_scale_update(st["amax_u"], st["scale_u"], st["inv_u"], _FP8_MARGIN)
+        dU8 = ext.row_gemm_fp8_out(dVflat, Wre.t().contiguous(),
+                                   st["scale_u"], st["amax_u"])
+        dX = ext.bdgcn_mode1_bwd_fp8(dU8.view(B, N, N, S, C), gop.A3T8,
+                                     st["inv_u"])
+        return dX, dW, dbias, None, None, None, None, None
 
 
 def fp8_forward_compatible(N: int, C: int, Hdim: int, S: int) -> bool:
@@ -288,14 +304,19 @@ def fp8_forward_compatible(N: int, C: int, Hdim: int, S: int) -> bool:
 
 
 def bdgcn_layer_fp8(X, W, bias, gop: GraphOperator, relu: bool = True,
-                    X8=None, emit_twin: bool = True):
+                    X8=None, emit_twin: bool = True, fp8_state: dict = None):
     """fp8-forward BDGCN layer: (Y_bf16, Y8_twin), or Y_bf16 alone with
     emit_twin=False (last layer — its consumer is the bf16 FC head).
+    fp8_state: per-layer make_fp8_state() dict (delayed dU scaling); a
+    transient one is created when omitted (single-shot calls/tests).
     GPU-only; callers gate on fp8_forward_compatible and fall back to
     bdgcn_layer otherwise."""
     if not X.is_cuda:
         raise RuntimeError("fp8-forward mode requires a GPU")
-    return _BDGCNLayerFp8Fn.apply(X, W, bias, gop, relu, X8, emit_twin)
+    if fp8_state is None:
+        fp8_state = make_fp8_state(X.device)
+    return _BDGCNLayerFp8Fn.apply(X, W, bias, gop, relu, X8, emit_twin,
+                                  fp8_state)
 
 
 class _Mode1ProjFn(torch.autograd.Function):

@@ -26,6 +26,9 @@ __launch_bounds__(256) __global__ void relu_bwd_colsum_kernel(ReluBwdParams p) {
     const __bf16* __restrict__ Y = (const __bf16*)p.Y;
     __bf16* __restrict__ dY = (__bf16*)p.dY;
 
+    unsigned char* __restrict__ dY8 = (unsigned char*)p.dY8;
+    const float qs = p.q_scale ? *p.q_scale : 1.f;
+
     float part[8] = {};
     const long chunks = p.total / 8;
     const long stride = (long)gridDim.x * 256;
@@ -38,13 +41,17 @@ __launch_bounds__(256) __global__ void relu_bwd_colsum_kernel(ReluBwdParams p) {
         const __bf16* h = (const __bf16*)&hv;
         const __bf16* y = (const __bf16*)&yv;
         __bf16* o = (__bf16*)&ov;
+        unsigned long long o8 = 0;
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
             const float v = (p.mask && !(to_f32(y[j]) > 0.f)) ? 0.f : to_f32(h[j]);
             o[j] = (__bf16)v;
             part[j] += v;
+            if (dY8)  // fused fp8 gradient quantize (scaled; see ext.hip)
+                o8 |= (unsigned long long)from_f32<unsigned char>(v * qs) << (8 * j);
         }
         *(Chunk16*)&dY[e0] = ov;
+        if (dY8) *(unsigned long long*)&dY8[e0] = o8;
     }
     // thread's column for slot j is fixed: (start*8 + j) % H
     if (p.det) {
@@ -108,4 +115,21 @@ extern "C" void slab_colsum_launch(const float* ws, float* out, long nb, long E,
                                    hipStream_t s) {
     slab_colsum_kernel<<<dim3((unsigned)((E + 255) / 256)), dim3(256), 0, s>>>(
         ws, out, nb, E);
+}
+
+// Delayed-scaling bookkeeping for the fp8 gradient path: derive this step's
+// quantize scale (and its exact descale pair) from LAST step's recorded
+// amax, then reset the amax accumulator — one thread, device-side only, so
+// the whole fp8 schedule needs no host synchronization.
+__global__ void fp8_scale_update_kernel(float* amax, float* scale, float* inv,
+                                        float margin) {
+    const float a = fmaxf(amax[0], 1e-20f);
+    scale[0] = margin / a;
+    inv[0] = a / margin;
+    amax[0] = 0.f;
+}
+
+extern "C" void fp8_scale_update_launch(float* amax, float* scale, float* inv,
+                                        float margin, hipStream_t s) {
+    fp8_scale_update_kernel<<<dim3(1), dim3(1), 0, s>>>(amax, scale, inv, margin);
 }

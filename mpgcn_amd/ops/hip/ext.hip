@@ -658,6 +658,81 @@ torch::Tensor row_gemm_fp8(torch::Tensor X8, torch::Tensor W8) {
     return OUT;
 }
 
+// relu_bwd_colsum with a FUSED fp8 gradient output: dY8 = fp8(dY * *q_scale)
+// written in the same streaming pass (the scale is derived from amax(|dH|),
+// a safe same-step upper bound of amax(|dY|) — see ops/functional.py).
+std::vector<torch::Tensor> relu_bwd_colsum_fp8(torch::Tensor dH, torch::Tensor Y,
+                                               bool mask, torch::Tensor q_scale) {
+    check_in(dH, "dH");
+    TORCH_CHECK(dH.scalar_type() == torch::kBFloat16, "bf16 only");
+    TORCH_CHECK(q_scale.is_cuda() && q_scale.scalar_type() == torch::kFloat);
+    const long H = dH.size(-1);
+    TORCH_CHECK(H >= 8 && (H & (H - 1)) == 0 && H <= 2048, "bad H");
+    auto dY = torch::empty_like(dH);
+    auto dY8 = torch::empty(dH.sizes(), dH.options().dtype(torch::kFloat8_e4m3fn));
+    const bool det = at::globalContext().deterministicAlgorithms();
+    const long nb = det ? relu_bwd_nblocks(dH.numel()) : 1;
+    auto colsum = det ? torch::empty({nb, H}, dH.options().dtype(torch::kFloat))
+                      : torch::zeros({H}, dH.options().dtype(torch::kFloat));
+    ReluBwdParams p{};
+    p.dH = dH.data_ptr();
+    p.Y = mask ? Y.data_ptr() : dH.data_ptr();
+    p.dY = dY.data_ptr();
+    p.dY8 = dY8.data_ptr();
+    p.q_scale = q_scale.data_ptr<float>();
+    p.colsum = colsum.data_ptr<float>();
+    p.total = dH.numel();
+    p.H = (int)H;
+    p.mask = mask ? 1 : 0;
+    p.det = det ? 1 : 0;
+    relu_bwd_colsum_launch(p, stream());
+    if (det) {
+        auto cs = torch::empty({H}, dH.options().dtype(torch::kFloat));
+        slab_colsum_launch(colsum.data_ptr<float>(), cs.data_ptr<float>(), nb,
+                           H, stream());
+        return {dY, dY8, cs};
+    }
+    return {dY, dY8, colsum};
+}
+
+// row_gemm emitting a SCALED fp8 output only (no bf16 store): the fp8-mode
+// dU projection — its sole consumer is the fp8 dX contraction, so the bf16
+// copy is never materialized. Tracks amax(|v|) into amax_out for the next
+// step's delayed scale.
+torch::Tensor row_gemm_fp8_out(torch::Tensor X, torch::Tensor W,
+                               torch::Tensor q_scale, torch::Tensor amax_out) {
+    check_in(X, "X");
+    check_in(W, "W");
+    TORCH_CHECK(X.scalar_type() == torch::kBFloat16, "bf16 only");
+    TORCH_CHECK(q_scale.is_cuda() && q_scale.scalar_type() == torch::kFloat);
+    TORCH_CHECK(amax_out.is_cuda() && amax_out.scalar_type() == torch::kFloat);
+    const long R = X.size(0), K = X.size(1), N = W.size(1);
+    TORCH_CHECK(W.size(0) == K, "W shape mismatch");
+    TORCH_CHECK(N <= 128 && K <= 2048, "row_gemm shape gate");
+    auto OUT8 = torch::empty({R, N}, X.options().dtype(torch::kFloat8_e4m3fn));
+    RowGemmParams p{};
+    p.X = X.data_ptr();
+    p.W = W.data_ptr();
+    p.OUT = nullptr;
+    p.OUT8 = OUT8.data_ptr();
+    p.q_scale = q_scale.data_ptr<float>();
+    p.amax_out = amax_out.data_ptr<float>();
+    p.R = R; p.K = (int)K; p.N = (int)N;
+    p.o_row = N; p.o_off = 0;
+    p.relu = 0;
+    p.x_vec = (K % chunk_elems(X) == 0);
+    row_gemm_launch(p, 0, stream());
+    return OUT8;
+}
+
+// Device-side delayed-scaling step: scale = margin/max(amax, eps);
+// inv = 1/scale; amax = 0. No host synchronization.
+void fp8_scale_update(torch::Tensor amax, torch::Tensor scale,
+                      torch::Tensor inv, double margin) {
+    fp8_scale_update_launch(amax.data_ptr<float>(), scale.data_ptr<float>(),
+                            inv.data_ptr<float>(), (float)margin, stream());
+}
+
 // Fused ReLU backward + bias-grad column sum: dY = dH * 1[Y>0], dbias=colsum.
 std::vector<torch::Tensor> relu_bwd_colsum(torch::Tensor dH, torch::Tensor Y,
                                            bool mask) {
@@ -709,6 +784,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "scaled fp8 gradient contraction dV");
     m.def("bdgcn_mode1_bwd_fp8", &bdgcn_mode1_bwd_fp8,
           "scaled fp8 gradient contraction dX");
+    m.def("relu_bwd_colsum_fp8", &relu_bwd_colsum_fp8,
+          "ReLU bwd + colsum + fused scaled fp8 dY8");
+    m.def("row_gemm_fp8_out", &row_gemm_fp8_out,
+          "row GEMM emitting scaled fp8 only, amax tracked");
+    m.def("fp8_scale_update", &fp8_scale_update,
+          "delayed-scaling scale/inv update from amax");
     m.def("row_gemm_out", &row_gemm_out, "row GEMM into strided output slice");
     m.def("lstm_step_fwd", &lstm_step_fwd, "fused LSTM cell forward step (K4)");
     m.def("lstm_fused_fwd", &lstm_fused_fwd, "register-resident fused LSTM forward");

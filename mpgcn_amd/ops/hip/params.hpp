@@ -29,7 +29,10 @@ struct AxisGemmParams {
 struct RowGemmParams {
     const void* X;
     const void* W;
-    void* OUT;
+    void* OUT;          // bf16/f32/fp8 primary output; nullable when OUT8 set
+    void* OUT8;         // optional fp8 e4m3 output = fp8(v * *q_scale); nullable
+    const float* q_scale;  // device ptr, quantize multiplier for OUT8
+    float* amax_out;    // device ptr, atomicMax'd with block amax(|v|); nullable
     const float* bias;
     long R;
     int K, N;
@@ -101,6 +104,8 @@ struct ReluBwdParams {
     const void* dH;  // (R, H) bf16
     const void* Y;   // (R, H) bf16 (forward output; ignored when !mask)
     void* dY;        // (R, H) bf16 out
+    void* dY8;       // optional fp8 out = fp8(dY * *q_scale); nullable
+    const float* q_scale;  // device ptr (fp8 gradient quantize scale)
     float* colsum;   // (H,) f32 zeroed, or (nblocks, H) workspace when det
     long total;      // R * H (multiple of 8)
     int H;           // power of two dividing 2048
@@ -123,4 +128,6 @@ int lstm_fused_bwd_blocks(long R);
 void relu_bwd_colsum_launch(ReluBwdParams p, hipStream_t s);
 long relu_bwd_nblocks(long total);
 long red_gemm_nblocks(long R);
+void fp8_scale_update_launch(float* amax, float* scale, float* inv,
+                             float margin, hipStream_t s);
 }

@@ -23,6 +23,9 @@ __launch_bounds__(256) __global__ void row_gemm_kernel(RowGemmParams p, int nf16
     const T* __restrict__ X = (const T*)p.X;
     const T* __restrict__ W = (const T*)p.W;
     T* __restrict__ O = (T*)p.OUT;
+    unsigned char* __restrict__ O8 = (unsigned char*)p.OUT8;
+    const float qs = p.q_scale ? *p.q_scale : 1.f;
+    float amax = 0.f;  // thread-local |v| max (fp8 delayed-scaling record)
 
     const int tid = threadIdx.x;
     // ---- stage W transposed: ldsWT[n][k] = W[k][n] (one time per block) ----
@@ -76,10 +79,23 @@ __launch_bounds__(256) __global__ void row_gemm_kernel(RowGemmParams p, int nf16
                     float v = acc[nf][r];
                     if (p.bias) v += p.bias[n];
                     if (p.relu) v = fmaxf(v, 0.f);
-                    O[m * p.o_row + p.o_off + n] = from_f32<T>(v);
+                    if (O) O[m * p.o_row + p.o_off + n] = from_f32<T>(v);
+                    if (O8) {
+                        amax = fmaxf(amax, fabsf(v));
+                        O8[m * p.o_row + p.o_off + n] =
+                            from_f32<unsigned char>(v * qs);
+                    }
                 }
             }
         }
+    }
+
+    if (p.amax_out) {
+        // wave max -> one atomic per wave (f32 >= 0: uint bit order matches)
+#pragma unroll
+        for (int s = 32; s >= 1; s >>= 1) amax = fmaxf(amax, __shfl_xor(amax, s));
+        if (lane == 0)
+            atomicMax((unsigned int*)p.amax_out, __float_as_uint(amax));
     }
 }
 
