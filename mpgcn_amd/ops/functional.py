@@ -194,15 +194,23 @@ _FP8_MARGIN = 224.0  # half of e4m3 max: headroom over the recorded amax
 
 
 def make_fp8_state(device) -> dict:
-    """Per-layer delayed-scaling state for the fp8 dU path: amax recorded by
-    the producing kernel this step becomes next step's quantize scale
-    (device-resident; the whole schedule runs without host syncs). amax
-    bootstraps at 1.0 — the first step may clip outliers above the margin,
-    after which the scale tracks the real gradient magnitude."""
+    """Per-layer delayed-scaling state for the fp8 gradient path (dY and dU):
+    amax recorded by the producing kernel this step becomes next step's
+    quantize scale (device-resident; the whole schedule runs without host
+    syncs). amax bootstraps at the margin, i.e. the first step quantizes at
+    scale 1 (e4m3's native 2^-9..448 covers typical gradient magnitudes);
+    from step 2 the scale tracks the real amax."""
+    def triple():
+        return {
+            "amax": torch.full((1,), _FP8_MARGIN, device=device),
+            "scale": torch.ones(1, device=device),
+            "inv": torch.ones(1, device=device),
+        }
+
+    y, u = triple(), triple()
     return {
-        "amax_u": torch.ones(1, device=device),
-        "scale_u": torch.full((1,), _FP8_MARGIN, device=device),
-        "inv_u": torch.full((1,), 1.0 / _FP8_MARGIN, device=device),
+        "amax_y": y["amax"], "scale_y": y["scale"], "inv_y": y["inv"],
+        "amax_u": u["amax"], "scale_u": u["scale"], "inv_u": u["inv"],
     }
 
 
@@ -259,14 +267,14 @@ class _BDGCNLayerFp8Fn(torch.autograd.Function):
         # Scaled-fp8 gradient contractions: gradients live well below e4m3's
         # 2^-9 subnormal floor, so every quantize carries a DEVICE-resident
         # dynamic scale and the axis-kernel epilogue descales — no host sync.
-        # Quantization is FUSED into the producers (a standalone
-        # amax+mul+cast chain costs more memory traffic than the fp8
-        # contraction saves — measured):
-        #   dY8: written by relu_bwd_colsum in its streaming pass, scale from
-        #        amax(|dH|) (a same-step upper bound of amax(|dY|));
+        # Quantization is FUSED into the producers with one-step-DELAYED
+        # per-layer scales (a standalone amax+mul+cast chain, and even a
+        # same-step amax pre-pass, cost more memory traffic than the fp8
+        # contraction saves — both measured):
+        #   dY8: written by relu_bwd_colsum in its streaming pass, which also
+        #        records this step's amax for the next step's scale;
         #   dU8: the ONLY dU materialization — row_gemm emits scaled fp8
-        #        directly (its sole consumer is the fp8 dX contraction) with
-        #        a one-step-delayed amax (make_fp8_state).
+        #        directly (its sole consumer is the fp8 dX contraction).
         # Weight-gradient reductions (red_gemm) stay bf16-accumulated-f32.
         ext = _ops.get_ext()
         U8, Wre, Y = ctx.saved_tensors
@@ -275,14 +283,13 @@ class _BDGCNLayerFp8Fn(torch.autograd.Function):
         B, N, S, C, Hdim = ctx.dims
 
         dH = dH.contiguous()
-        amax_h = dH.abs().amax().float().clamp(min=1e-20)
-        scale_y = _FP8_MARGIN / amax_h
-        inv_y = amax_h / _FP8_MARGIN
-        dY, dY8, dbias = ext.relu_bwd_colsum_fp8(dH, Y, ctx.relu, scale_y)
+        ext.fp8_scale_update(st["amax_y"], st["scale_y"], st["inv_y"], _FP8_MARGIN)
+        dY, dY8, dbias = ext.relu_bwd_colsum_fp8(dH, Y, ctx.relu,
+                                                 st["scale_y"], st["amax_y"])
         if not ctx.has_bias:
             dbias = None
 
-        dV = ext.bdgcn_mode2_bwd_fp8(dY8, gop.A28, S, inv_y)  # (B,N,N,S,H)
+        dV = ext.bdgcn_mode2_bwd_fp8(dY8, gop.A28, S, st["inv_y"])  # (B,N,N,S,H)
         R = B * N * N
         dVflat = dV.reshape(R, S * Hdim)
         dWreT, _, _ = ext.red_gemm(dVflat, U8.reshape(R, S * C), False, None, 0, 0)

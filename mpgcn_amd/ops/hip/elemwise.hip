@@ -28,6 +28,7 @@ __launch_bounds__(256) __global__ void relu_bwd_colsum_kernel(ReluBwdParams p) {
 
     unsigned char* __restrict__ dY8 = (unsigned char*)p.dY8;
     const float qs = p.q_scale ? *p.q_scale : 1.f;
+    float amax = 0.f;
 
     float part[8] = {};
     const long chunks = p.total / 8;
@@ -47,11 +48,20 @@ __launch_bounds__(256) __global__ void relu_bwd_colsum_kernel(ReluBwdParams p) {
             const float v = (p.mask && !(to_f32(y[j]) > 0.f)) ? 0.f : to_f32(h[j]);
             o[j] = (__bf16)v;
             part[j] += v;
-            if (dY8)  // fused fp8 gradient quantize (scaled; see ext.hip)
+            if (dY8) {  // fused fp8 gradient quantize (delayed scale)
                 o8 |= (unsigned long long)from_f32<unsigned char>(v * qs) << (8 * j);
+                amax = fmaxf(amax, fabsf(v));
+            }
         }
         *(Chunk16*)&dY[e0] = ov;
         if (dY8) *(unsigned long long*)&dY8[e0] = o8;
+    }
+    if (p.amax_out) {
+#pragma unroll
+        for (int s = 32; s >= 1; s >>= 1)
+            amax = fmaxf(amax, __shfl_xor(amax, s));
+        if ((tid % 64) == 0)
+            atomicMax((unsigned int*)p.amax_out, __float_as_uint(amax));
     }
     // thread's column for slot j is fixed: (start*8 + j) % H
     if (p.det) {

@@ -659,13 +659,15 @@ torch::Tensor row_gemm_fp8(torch::Tensor X8, torch::Tensor W8) {
 }
 
 // relu_bwd_colsum with a FUSED fp8 gradient output: dY8 = fp8(dY * *q_scale)
-// written in the same streaming pass (the scale is derived from amax(|dH|),
-// a safe same-step upper bound of amax(|dY|) — see ops/functional.py).
+// written in the same streaming pass; this step's amax(|dY|) is recorded
+// into amax_out for the next step's delayed scale (ops/functional.py).
 std::vector<torch::Tensor> relu_bwd_colsum_fp8(torch::Tensor dH, torch::Tensor Y,
-                                               bool mask, torch::Tensor q_scale) {
+                                               bool mask, torch::Tensor q_scale,
+                                               torch::Tensor amax_out) {
     check_in(dH, "dH");
     TORCH_CHECK(dH.scalar_type() == torch::kBFloat16, "bf16 only");
     TORCH_CHECK(q_scale.is_cuda() && q_scale.scalar_type() == torch::kFloat);
+    TORCH_CHECK(amax_out.is_cuda() && amax_out.scalar_type() == torch::kFloat);
     const long H = dH.size(-1);
     TORCH_CHECK(H >= 8 && (H & (H - 1)) == 0 && H <= 2048, "bad H");
     auto dY = torch::empty_like(dH);
@@ -680,6 +682,7 @@ std::vector<torch::Tensor> relu_bwd_colsum_fp8(torch::Tensor dH, torch::Tensor Y
     p.dY = dY.data_ptr();
     p.dY8 = dY8.data_ptr();
     p.q_scale = q_scale.data_ptr<float>();
+    p.amax_out = amax_out.data_ptr<float>();
     p.colsum = colsum.data_ptr<float>();
     p.total = dH.numel();
     p.H = (int)H;

@@ -106,6 +106,7 @@ struct ReluBwdParams {
     void* dY;        // (R, H) bf16 out
     void* dY8;       // optional fp8 out = fp8(dY * *q_scale); nullable
     const float* q_scale;  // device ptr (fp8 gradient quantize scale)
+    float* amax_out; // device ptr, atomicMax'd with amax(|dY|); nullable
     float* colsum;   // (H,) f32 zeroed, or (nblocks, H) workspace when det
     long total;      // R * H (multiple of 8)
     int H;           // power of two dividing 2048

@@ -91,7 +91,7 @@ def test_fp8_twin_chains_between_layers():
     # norm at quantization scale
     num = (Y2_chained.float() - Y2_requant.float()).norm()
     den = Y2_requant.float().norm() + 1e-9
-    assert num / den < 0.02, (num / den).item()
+    assert num / den < 0.06, (num / den).item()
 
 
 def test_fp8_model_train_step_loss_decreases():
