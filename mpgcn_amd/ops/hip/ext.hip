@@ -330,60 +330,97 @@ void lstm_step_bwd(torch::Tensor dh, c10::optional<torch::Tensor> dc_in,
     lstm_step_bwd_launch(p, is_f32(dh), stream());
 }
 
-// Fully-fused register-resident LSTM forward: x (R, T) bf16 -> h_T (R, H=32).
-torch::Tensor lstm_fused_fwd(torch::Tensor x, long T_logical, torch::Tensor whh,
-                             torch::Tensor wih, torch::Tensor bias) {
+// Fully-fused register-resident LSTM forward over ONE chunk of <= 8 steps:
+// x (R, x_cols) bf16, columns [x_off, x_off+T). Chunks longer sequences
+// chain through (h_in, c_in) boundary checkpoints (ops/functional.py) —
+// the register-resident schedule then covers ANY T with O(R*H) checkpoint
+// traffic instead of the slab path's O(R*H*T) state round trips.
+// Returns {h_end (R,32) bf16, c_end (R,32) f32 or undefined}.
+std::vector<torch::Tensor> lstm_fused_fwd(torch::Tensor x, long x_off,
+                                          long T_logical, torch::Tensor whh,
+                                          torch::Tensor wih, torch::Tensor bias,
+                                          c10::optional<torch::Tensor> h_in,
+                                          c10::optional<torch::Tensor> c_in,
+                                          bool want_c) {
     check_in(x, "x");
     check_in(whh, "whh");
     TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "fused LSTM is bf16-only");
     TORCH_CHECK(whh.size(1) == 32 && whh.size(0) == 128, "fused LSTM needs H=32");
-    TORCH_CHECK(x.size(1) == 8, "x must be zero-padded to 8 columns");
+    TORCH_CHECK(x.size(1) % 8 == 0 && x_off % 8 == 0 && x_off + 8 <= x.size(1),
+                "x must be zero-padded to a multiple of 8 columns");
     const long R = x.size(0);
     const int T = (int)T_logical;
-    TORCH_CHECK(T >= 1 && T <= 8, "fused LSTM needs T <= 8");
+    TORCH_CHECK(T >= 1 && T <= 8, "fused LSTM chunk needs T <= 8");
+    const bool has_h = h_in.has_value() && h_in->defined();
     auto h = torch::empty({R, 32}, x.options());
+    auto c = want_c ? torch::empty({R, 32}, x.options().dtype(torch::kFloat))
+                    : torch::Tensor();
     LstmFusedParams p{};
     p.x = x.data_ptr();
+    p.x_cols = x.size(1); p.x_off = x_off;
     p.whh = whh.data_ptr();
     p.wih = wih.data_ptr<float>();
     p.bias = bias.data_ptr<float>();
     p.h_out = h.data_ptr();
+    p.h_in = has_h ? h_in->data_ptr() : nullptr;
+    p.c_in = has_h ? c_in->data_ptr<float>() : nullptr;
+    p.c_out = want_c ? c.data_ptr<float>() : nullptr;
     p.R = R; p.T = T;
     lstm_fused_fwd_launch(p, stream());
-    return h;
+    return {h, c};
 }
 
-// Fused backward with in-kernel forward recompute. Returns
-// {dwhh (4H,H) f32, dbias (4H) f32, dwih (4H) f32, dx (R,T) bf16 or undefined}.
-std::vector<torch::Tensor> lstm_fused_bwd(torch::Tensor x, long T_logical,
+// Fused backward of one chunk with in-kernel forward recompute from the
+// chunk-entry checkpoint (h_in, c_in). dc_in chains dL/dc from the following
+// chunk; want_prev emits {dh_prev, dc_prev} = gradients at the chunk entry
+// for the preceding chunk's backward. dx (full (R, x_cols) buffer) gets this
+// chunk's columns written in place. Returns
+// {dwhh (4H,H) f32, dbias (4H) f32, dwih (4H) f32, dh_prev, dc_prev}.
+std::vector<torch::Tensor> lstm_fused_bwd(torch::Tensor x, long x_off,
+                                          long T_logical,
                                           torch::Tensor whh,
                                           torch::Tensor whh2, torch::Tensor wih,
                                           torch::Tensor bias, torch::Tensor dh,
-                                          bool need_dx) {
+                                          c10::optional<torch::Tensor> h_in,
+                                          c10::optional<torch::Tensor> c_in,
+                                          c10::optional<torch::Tensor> dc_in,
+                                          bool want_prev,
+                                          c10::optional<torch::Tensor> dx) {
     check_in(x, "x");
     check_in(whh, "whh");
     check_in(whh2, "whh2");
     check_in(dh, "dh");
-    TORCH_CHECK(x.size(1) == 8, "x must be zero-padded to 8 columns");
+    TORCH_CHECK(x.size(1) % 8 == 0 && x_off % 8 == 0 && x_off + 8 <= x.size(1),
+                "x must be zero-padded to a multiple of 8 columns");
     const long R = x.size(0);
     const int T = (int)T_logical;
     const int nb = lstm_fused_bwd_blocks(R);
+    const bool has_h = h_in.has_value() && h_in->defined();
+    const bool has_dc = dc_in.has_value() && dc_in->defined();
+    const bool has_dx = dx.has_value() && dx->defined();
     auto f32 = x.options().dtype(torch::kFloat);
     auto ws_dw = torch::empty({nb, 128, 32}, f32);
     auto ws_db = torch::empty({nb, 128}, f32);
     auto ws_dwih = torch::empty({nb, 128}, f32);
-    auto dx = need_dx ? torch::empty({R, (long)T}, x.options()) : torch::Tensor();
+    auto dh_prev = want_prev ? torch::empty({R, 32}, x.options()) : torch::Tensor();
+    auto dc_prev = want_prev ? torch::empty({R, 32}, f32) : torch::Tensor();
     LstmFusedParams p{};
     p.x = x.data_ptr();
+    p.x_cols = x.size(1); p.x_off = x_off;
     p.whh = whh.data_ptr();
     p.whh2 = whh2.data_ptr();
     p.wih = wih.data_ptr<float>();
     p.bias = bias.data_ptr<float>();
     p.dh = dh.data_ptr();
+    p.h_in = has_h ? h_in->data_ptr() : nullptr;
+    p.c_in = has_h ? c_in->data_ptr<float>() : nullptr;
+    p.dc_in = has_dc ? dc_in->data_ptr<float>() : nullptr;
+    p.dh_out = want_prev ? dh_prev.data_ptr() : nullptr;
+    p.dc_out = want_prev ? dc_prev.data_ptr<float>() : nullptr;
     p.ws_dw = ws_dw.data_ptr<float>();
     p.ws_db = ws_db.data_ptr<float>();
     p.ws_dwih = ws_dwih.data_ptr<float>();
-    p.dx = need_dx ? dx.data_ptr() : nullptr;
+    p.dx = has_dx ? dx->data_ptr() : nullptr;
     p.R = R; p.T = T;
     lstm_fused_bwd_launch(p, stream());
     // in-kernel fixed-order workspace reduction (deterministic; three
@@ -397,7 +434,7 @@ std::vector<torch::Tensor> lstm_fused_bwd(torch::Tensor x, long T_logical,
                        128, stream());
     slab_colsum_launch(ws_dwih.data_ptr<float>(), dwih.data_ptr<float>(), nb,
                        128, stream());
-    return {dwhh, dbias, dwih, dx};
+    return {dwhh, dbias, dwih, dh_prev, dc_prev};
 }
 
 // fp8 probe of the mode-2 contraction (docs/ROADMAP.md byte-reduction

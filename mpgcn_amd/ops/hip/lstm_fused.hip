@@ -88,6 +88,10 @@ __launch_bounds__(256) __global__ void lstm_fused_fwd_kernel(LstmFusedParams p) 
         bias_r[nf] = p.bias[nf * 16 + lrow];
     }
 
+    const __bf16* __restrict__ Hin = (const __bf16*)p.h_in;
+    const float* __restrict__ Cin = p.c_in;
+    float* __restrict__ Cout = p.c_out;
+
     const long ntiles = (p.R + 63) / 64;
     for (long tile = blockIdx.x; tile < ntiles; tile += gridDim.x) {
         const long r0 = tile * 64 + w * 16;
@@ -95,11 +99,27 @@ __launch_bounds__(256) __global__ void lstm_fused_fwd_kernel(LstmFusedParams p) 
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
             const long m = r0 + kgrp * 4 + r;
-            if (m < p.R) *(Chunk16*)&xrow[r][0] = *(const Chunk16*)&X[m * LF_XCOLS];
+            if (m < p.R)
+                *(Chunk16*)&xrow[r][0] = *(const Chunk16*)&X[m * p.x_cols + p.x_off];
             else for (int t2 = 0; t2 < LF_XCOLS; ++t2) xrow[r][t2] = (__bf16)0.f;
         }
+        // chunk-entry state: zeros for the first chunk, the previous chunk's
+        // boundary checkpoint otherwise (chunked T > 8 schedule)
         lf_frag h_frag = {};
+        if (Hin) {
+            const long row = r0 + lrow;
+            if (row < p.R) h_frag = *(const lf_frag*)&Hin[row * LF_H + kgrp * 8];
+        }
         float c[2][4] = {};  // c[jf][r] for j = jf*16 + lrow, row = r0 + kgrp*4 + r
+        if (Cin) {
+#pragma unroll
+            for (int jf = 0; jf < 2; ++jf)
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    const long m = r0 + kgrp * 4 + r;
+                    if (m < p.R) c[jf][r] = Cin[m * LF_H + jf * 16 + lrow];
+                }
+        }
 #pragma unroll
         for (int t = 0; t < T; ++t) {
             f32x4 acc[8] = {};
@@ -126,9 +146,18 @@ __launch_bounds__(256) __global__ void lstm_fused_fwd_kernel(LstmFusedParams p) 
             h_frag = *(const lf_frag*)&myH[lrow * (LF_H + 8) + kgrp * 8];
             lds_wave_fence();
         }
-        // store h_T, vectorized: lane holds row lrow's k-run
+        // store chunk-end h, vectorized: lane holds row lrow's k-run
         const long row = r0 + lrow;
         if (row < p.R) *(lf_frag*)&Ho[row * LF_H + kgrp * 8] = h_frag;
+        if (Cout) {
+#pragma unroll
+            for (int jf = 0; jf < 2; ++jf)
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    const long m = r0 + kgrp * 4 + r;
+                    if (m < p.R) Cout[m * LF_H + jf * 16 + lrow] = c[jf][r];
+                }
+        }
     }
 }
 
@@ -179,6 +208,12 @@ __launch_bounds__(256) __global__ void lstm_fused_bwd_kernel(LstmFusedParams p) 
     f32x4 dw_acc[8][2] = {};
     float db_acc[8] = {}, dwih_acc[8] = {};
 
+    const __bf16* __restrict__ Hin = (const __bf16*)p.h_in;
+    const float* __restrict__ Cin = p.c_in;
+    const float* __restrict__ DCin = p.dc_in;
+    __bf16* __restrict__ DHout = (__bf16*)p.dh_out;
+    float* __restrict__ DCout = p.dc_out;
+
     const long ntiles = (p.R + 63) / 64;
     for (long tile = blockIdx.x; tile < ntiles; tile += gridDim.x) {
         const long r0 = tile * 64 + w * 16;
@@ -186,15 +221,37 @@ __launch_bounds__(256) __global__ void lstm_fused_bwd_kernel(LstmFusedParams p) 
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
             const long m = r0 + kgrp * 4 + r;
-            if (m < p.R) *(Chunk16*)&xrow[r][0] = *(const Chunk16*)&X[m * LF_XCOLS];
+            if (m < p.R)
+                *(Chunk16*)&xrow[r][0] = *(const Chunk16*)&X[m * p.x_cols + p.x_off];
             else for (int t2 = 0; t2 < LF_XCOLS; ++t2) xrow[r][t2] = (__bf16)0.f;
+        }
+        // chunk-entry state (chunked T > 8: previous chunk's checkpoint)
+        lf_frag h_in_frag = {};
+        float c0r[2][4] = {};  // initial c (phase B's c_prev at t = 0)
+        if (Hin) {
+            const long row = r0 + lrow;
+            if (row < p.R)
+                h_in_frag = *(const lf_frag*)&Hin[row * LF_H + kgrp * 8];
+        }
+        if (Cin) {
+#pragma unroll
+            for (int jf = 0; jf < 2; ++jf)
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    const long m = r0 + kgrp * 4 + r;
+                    if (m < p.R) c0r[jf][r] = Cin[m * LF_H + jf * 16 + lrow];
+                }
         }
         // ---- phase A: forward recompute; h states in registers, c in LDS ----
         lf_frag h_states[T];  // h AFTER step t (A-frag layout)
 #define CST(t, r, jf) myC[(t) * 16 * 33 + (kgrp * 4 + (r)) * 33 + (jf) * 16 + lrow]
         {
-            lf_frag h_frag = {};
-            float c[2][4] = {};
+            lf_frag h_frag = h_in_frag;
+            float c[2][4];
+#pragma unroll
+            for (int jf = 0; jf < 2; ++jf)
+#pragma unroll
+                for (int r = 0; r < 4; ++r) c[jf][r] = c0r[jf][r];
 #pragma unroll
             for (int t = 0; t < T; ++t) {
                 f32x4 acc[8] = {};
@@ -232,6 +289,7 @@ __launch_bounds__(256) __global__ void lstm_fused_bwd_kernel(LstmFusedParams p) 
             for (int r = 0; r < 4; ++r) {
                 const long m = r0 + kgrp * 4 + r;
                 dh[jf][r] = (m < p.R) ? to_f32(DH[m * LF_H + jf * 16 + lrow]) : 0.f;
+                if (DCin && m < p.R) dc[jf][r] = DCin[m * LF_H + jf * 16 + lrow];
             }
 
 #pragma unroll
@@ -239,7 +297,7 @@ __launch_bounds__(256) __global__ void lstm_fused_bwd_kernel(LstmFusedParams p) 
             const int t = T - 1 - tt;
             const int slot = tt % 2;
             // recompute this step's gates from h_prev
-            lf_frag hp = {};
+            lf_frag hp = h_in_frag;
             if (t > 0) hp = h_states[t - 1];
             f32x4 acc[8] = {};
             lf_gate_mfma(ldsW, hp, lrow, kgrp, acc);
@@ -260,7 +318,7 @@ __launch_bounds__(256) __global__ void lstm_fused_bwd_kernel(LstmFusedParams p) 
                     const float i_g = gv[0 + jf], f_g = gv[2 + jf];
                     const float g_g = gv[4 + jf], o_g = gv[6 + jf];
                     const float c_t = CST(t, r, jf);
-                    const float c_prev = (t > 0) ? CST(t - 1, r, jf) : 0.f;
+                    const float c_prev = (t > 0) ? CST(t - 1, r, jf) : c0r[jf][r];
                     const float tc = fast_tanh(c_t);
                     float d_c = dc[jf][r] + dh[jf][r] * o_g * (1.f - tc * tc);
                     const float d_i = d_c * g_g, d_g = d_c * i_g, d_f = d_c * c_prev;
@@ -294,7 +352,8 @@ __launch_bounds__(256) __global__ void lstm_fused_bwd_kernel(LstmFusedParams p) 
                     for (int s = 1; s < 16; s <<= 1)
                         part += __shfl_xor(part, s);
                     const long m = r0 + kgrp * 4 + r;
-                    if (lrow == 0 && m < p.R) DX[m * T + t] = (__bf16)part;
+                    if (lrow == 0 && m < p.R)
+                        DX[m * p.x_cols + p.x_off + t] = (__bf16)part;
                 }
             }
 
@@ -314,7 +373,9 @@ __launch_bounds__(256) __global__ void lstm_fused_bwd_kernel(LstmFusedParams p) 
             lds_wave_fence();
 
             // dh chain: dh_prev[row][k] = sum_n dgp[row][n] * Whh[n][k]
-            if (t > 0) {
+            // (also at t = 0 when the chunk must emit d(h_in) for the
+            // previous chunk's backward)
+            if (t > 0 || DHout) {
                 f32x4 dh_acc[2] = {};
 #pragma unroll
                 for (int kf = 0; kf < 4; ++kf) {
@@ -367,6 +428,19 @@ __launch_bounds__(256) __global__ void lstm_fused_bwd_kernel(LstmFusedParams p) 
                 }
             }
             lds_wave_fence();
+        }
+        // chunk-start gradients out (chained into the previous chunk's bwd)
+        if (DHout) {
+#pragma unroll
+            for (int jf = 0; jf < 2; ++jf)
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    const long m = r0 + kgrp * 4 + r;
+                    if (m < p.R) {
+                        DHout[m * LF_H + jf * 16 + lrow] = (__bf16)dh[jf][r];
+                        DCout[m * LF_H + jf * 16 + lrow] = dc[jf][r];
+                    }
+                }
         }
     }
     __syncthreads();  // before reusing per-wave LDS as the block-reduce scratch

@@ -85,19 +85,27 @@ struct RedGemmParams {
 };
 
 struct LstmFusedParams {
-    const void* x;  // (R, T) T
+    const void* x;  // (R, x_cols) T; chunk reads columns [x_off, x_off+T)
     const void* whh;   // (4H, H)
     const void* whh2;  // (H, 4H) = whh^T (backward only)
     const float* wih;  // (4H)
     const float* bias; // (4H)
-    void* h_out;       // (R, H) — forward output h_T
-    const void* dh;    // (R, H) — backward input
+    void* h_out;       // (R, H) — forward output h at chunk end
+    const void* dh;    // (R, H) — backward input dL/dh(chunk end)
     float* ws_dw;      // (nblocks, 4H, H) f32 workspace
     float* ws_db;      // (nblocks, 4H)
     float* ws_dwih;    // (nblocks, 4H)
-    void* dx;          // (R, T) or nullptr
+    void* dx;          // (R, x_cols) or nullptr (chunk writes its columns)
     long R;
-    int T;
+    int T;             // chunk length (<= 8)
+    // ---- chunked T > 8 support (boundary states; all nullable) ----
+    long x_cols, x_off;     // x row stride / this chunk's first column
+    const void* h_in;       // (R, H) bf16 state entering the chunk (null = 0)
+    const float* c_in;      // (R, H) f32
+    float* c_out;           // (R, H) f32 chunk-end cell state (fwd checkpoint)
+    const float* dc_in;     // (R, H) f32 dL/dc(chunk end) (bwd chaining)
+    void* dh_out;           // (R, H) bf16 dL/dh(chunk start)
+    float* dc_out;          // (R, H) f32 dL/dc(chunk start)
 };
 
 struct ReluBwdParams {

@@ -158,7 +158,9 @@ def test_row_gemm_chunked_wide():
 
 
 @pytest.mark.parametrize("dtype,T", [(torch.float32, 7), (torch.bfloat16, 7),
-                                      (torch.bfloat16, 10)])  # T=10: slab path
+                                      (torch.float32, 10),  # slab path
+                                      (torch.bfloat16, 10),  # 2-chunk path
+                                      (torch.bfloat16, 16)])  # 2 full chunks
 def test_lstm_forward_vs_nn_lstm(dtype, T):
     from mpgcn_amd.ops.functional import fused_lstm_last
 
@@ -203,14 +205,16 @@ def test_lstm_backward_grads_match_autograd():
         )
 
 
-def test_reg_lstm_bf16_grads_match_autograd():
+@pytest.mark.parametrize("T", [7, 14, 16])
+def test_reg_lstm_bf16_grads_match_autograd(T):
     """The register-resident bf16 LSTM (forward + recompute-backward) vs the
-    eager fp32 autograd reference."""
+    eager fp32 autograd reference. T > 8 exercises the CHUNKED schedule
+    (boundary checkpoints + chained (dh, dc) across chunk backwards)."""
     from mpgcn_amd.ops import eager
     from mpgcn_amd.ops.functional import _RegLSTMFn
 
     torch.manual_seed(12)
-    R, T, H = 3000, 7, 32
+    R, H = 3000, 32
     x = torch.randn(R, T, device=DEV) * 0.5
     w_ih = torch.randn(4 * H, 1, device=DEV) * 0.2
     w_hh = torch.randn(4 * H, H, device=DEV) * 0.2

@@ -29,9 +29,11 @@ whh = torch.randn(128, 32, device=dev, dtype=torch.bfloat16) * 0.1
 wih = torch.randn(128, device=dev) * 0.1
 bias = torch.randn(128, device=dev) * 0.1
 dh = torch.randn(R, 32, device=dev, dtype=torch.bfloat16)
-cmp("lstm_fused_fwd", lambda: ext.lstm_fused_fwd(x, 7, whh, wih, bias))
+cmp("lstm_fused_fwd", lambda: ext.lstm_fused_fwd(x, 0, 7, whh, wih, bias, None, None, False))
+dxbuf = torch.empty_like(x)
 cmp("lstm_fused_bwd", lambda: ext.lstm_fused_bwd(
-    x, 7, whh, whh.t().contiguous(), wih, bias, dh, True))
+    x, 0, 7, whh, whh.t().contiguous(), wih, bias, dh, None, None, None,
+    False, dxbuf))
 
 dH = torch.randn(65536, 32, device=dev, dtype=torch.bfloat16)
 Y = torch.randn(65536, 32, device=dev, dtype=torch.bfloat16)
