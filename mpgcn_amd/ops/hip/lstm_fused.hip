@@ -164,7 +164,7 @@ __launch_bounds__(256) __global__ void lstm_fused_fwd_kernel(LstmFusedParams p) 
 // NOTE on gv indexing above: gate order i,f,g,o over n = 0..127 means
 // nf 0..1 = i, 2..3 = f, 4..5 = g, 6..7 = o; jf = n/16 % 2 selects the half.
 
-template <int T>
+template <int T, bool CH = false>
 __launch_bounds__(256) __global__ void lstm_fused_bwd_kernel(LstmFusedParams p) {
     __shared__ __bf16 ldsW[LF_G4 * (LF_H + 8)];   // [n][k] gate-GEMM B image
     __shared__ __bf16 ldsW2[LF_H * (LF_G4 + 8)];  // [k][n] dh-chain B image
@@ -208,11 +208,14 @@ __launch_bounds__(256) __global__ void lstm_fused_bwd_kernel(LstmFusedParams p) 
     f32x4 dw_acc[8][2] = {};
     float db_acc[8] = {}, dwih_acc[8] = {};
 
-    const __bf16* __restrict__ Hin = (const __bf16*)p.h_in;
-    const float* __restrict__ Cin = p.c_in;
-    const float* __restrict__ DCin = p.dc_in;
-    __bf16* __restrict__ DHout = (__bf16*)p.dh_out;
-    float* __restrict__ DCout = p.dc_out;
+    // CH=false (the whole-sequence T <= 8 path, incl. the flagship T=7)
+    // compiles the chunk-boundary logic OUT — the extra live state pushed
+    // the T=7 instantiation into VGPR spills otherwise.
+    const __bf16* __restrict__ Hin = CH ? (const __bf16*)p.h_in : nullptr;
+    const float* __restrict__ Cin = CH ? p.c_in : nullptr;
+    const float* __restrict__ DCin = CH ? p.dc_in : nullptr;
+    __bf16* __restrict__ DHout = CH ? (__bf16*)p.dh_out : nullptr;
+    float* __restrict__ DCout = CH ? p.dc_out : nullptr;
 
     const long ntiles = (p.R + 63) / 64;
     for (long tile = blockIdx.x; tile < ntiles; tile += gridDim.x) {
@@ -228,12 +231,12 @@ __launch_bounds__(256) __global__ void lstm_fused_bwd_kernel(LstmFusedParams p) 
         // chunk-entry state (chunked T > 8: previous chunk's checkpoint)
         lf_frag h_in_frag = {};
         float c0r[2][4] = {};  // initial c (phase B's c_prev at t = 0)
-        if (Hin) {
+        if (CH && Hin) {
             const long row = r0 + lrow;
             if (row < p.R)
                 h_in_frag = *(const lf_frag*)&Hin[row * LF_H + kgrp * 8];
         }
-        if (Cin) {
+        if (CH && Cin) {
 #pragma unroll
             for (int jf = 0; jf < 2; ++jf)
 #pragma unroll
@@ -289,7 +292,7 @@ __launch_bounds__(256) __global__ void lstm_fused_bwd_kernel(LstmFusedParams p) 
             for (int r = 0; r < 4; ++r) {
                 const long m = r0 + kgrp * 4 + r;
                 dh[jf][r] = (m < p.R) ? to_f32(DH[m * LF_H + jf * 16 + lrow]) : 0.f;
-                if (DCin && m < p.R) dc[jf][r] = DCin[m * LF_H + jf * 16 + lrow];
+                if (CH && DCin && m < p.R) dc[jf][r] = DCin[m * LF_H + jf * 16 + lrow];
             }
 
 #pragma unroll
@@ -430,7 +433,7 @@ __launch_bounds__(256) __global__ void lstm_fused_bwd_kernel(LstmFusedParams p) 
             lds_wave_fence();
         }
         // chunk-start gradients out (chained into the previous chunk's bwd)
-        if (DHout) {
+        if (CH && DHout) {
 #pragma unroll
             for (int jf = 0; jf < 2; ++jf)
 #pragma unroll
@@ -517,8 +520,16 @@ extern "C" int lstm_fused_bwd_blocks(long R) {
 
 extern "C" void lstm_fused_bwd_launch(LstmFusedParams p, hipStream_t s) {
     dim3 grid(lstm_fused_bwd_blocks(p.R)), block(256);
+    const bool ch = p.h_in || p.dc_in || p.dh_out;
     switch (p.T) {
-#define CASE(TT) case TT: lstm_fused_bwd_kernel<TT><<<grid, block, 0, s>>>(p); break;
+#define CASE(TT) \
+    case TT: { \
+        if (ch) { \
+            lstm_fused_bwd_kernel<TT, true><<<grid, block, 0, s>>>(p); \
+        } else { \
+            lstm_fused_bwd_kernel<TT, false><<<grid, block, 0, s>>>(p); \
+        } \
+    } break;
         CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
 #undef CASE
         default: fprintf(stderr, "lstm_fused_bwd: T=%d unsupported\n", p.T); abort();
