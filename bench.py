@@ -103,10 +103,13 @@ def main():
     # device-resident synthetic OD pool (log1p-scale magnitudes), windowed by index
     T_pool = 64
     pool = torch.log1p(20.0 * torch.rand(T_pool, N, N, 1, device=device))
+    from mpgcn_amd.graph.supports import tag_like
+
     adj = (torch.rand(N, N, device=device) < 0.1).float()
-    G_static = build_supports(adj.unsqueeze(0), kernel, K_order).squeeze(0)
-    G_corr = build_supports(torch.rand(1, N, N, device=device), kernel,
-                            K_order).squeeze(0)  # third perspective (static)
+    _gs = build_supports(adj.unsqueeze(0), kernel, K_order)
+    G_static = tag_like(_gs.squeeze(0), _gs)
+    _gc = build_supports(torch.rand(1, N, N, device=device), kernel, K_order)
+    G_corr = tag_like(_gc.squeeze(0), _gc)  # third perspective (static)
     # raw day-of-week correlation graphs (support build runs per-step, timed)
     O_dyn_raw = torch.rand(7, N, N, device=device)
     D_dyn_raw = torch.rand(7, N, N, device=device)

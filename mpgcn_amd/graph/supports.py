@@ -36,6 +36,14 @@ KERNEL_TYPES = (
 )
 
 
+def tag_like(dst: torch.Tensor, src: torch.Tensor) -> torch.Tensor:
+    """Propagate the _identity_first tag through views/squeezes/casts (tensor
+    attributes do not survive torch ops)."""
+    if getattr(src, "_identity_first", False):
+        dst._identity_first = True
+    return dst
+
+
 def get_support_K(kernel_type: str, cheby_order: int) -> int:
     """Support count per kernel type. Contract of Model_Trainer.py:24-36."""
     if kernel_type == "localpool":
@@ -148,4 +156,9 @@ def build_supports(
         )
 
     assert out.shape[1] == K, (out.shape, K)
+    if kernel_type != "localpool" and K >= 2:
+        # every Chebyshev-family stack starts with T_0 = I by construction —
+        # the GPU layer skips the identity support's products entirely
+        # (ops/functional.py GraphOperator.id_first)
+        out._identity_first = True
     return out

@@ -123,13 +123,19 @@ class MPGCN(nn.Module):
         forwards (dynamic graphs are data-dependent and rebuilt)."""
         if len(G_list) != self.M:
             raise ValueError(f"expected {self.M} graph inputs, got {len(G_list)}")
+        def cast(G):
+            Gc = G.to(self.compute_dtype)
+            if getattr(G, "_identity_first", False):
+                Gc._identity_first = True  # survive the dtype cast
+            return Gc
+
         gops = []
         for G in G_list:
             if isinstance(G, torch.Tensor):
                 key = (id(G), G.device, self.compute_dtype)
                 cached = self._gop_cache.get(key)
                 if cached is None or cached[0] is not G:
-                    Gc = G.to(self.compute_dtype)
+                    Gc = cast(G)
                     cached = (G, GraphOperator(Gc, Gc))
                     if len(self._gop_cache) >= 8:  # bound: one per static graph
                         self._gop_cache.clear()
@@ -137,8 +143,7 @@ class MPGCN(nn.Module):
                 gops.append(cached[1])
             else:
                 Go, Gd = G
-                gops.append(GraphOperator(Go.to(self.compute_dtype),
-                                          Gd.to(self.compute_dtype)))
+                gops.append(GraphOperator(cast(Go), cast(Gd)))
         return gops
 
     def forward(self, x_seq: torch.Tensor, G_list: list) -> torch.Tensor:

@@ -33,11 +33,22 @@ class GraphOperator:
     Go/Gd: (S, N, N) static or (B, S, N, N) dynamic, in compute dtype.
     """
 
-    def __init__(self, Go: torch.Tensor, Gd: torch.Tensor):
+    def __init__(self, Go: torch.Tensor, Gd: torch.Tensor,
+                 id_first: bool | None = None):
         self.Go = Go
         self.Gd = Gd
         self.S = Go.shape[-3]
         self.N = Go.shape[-1]
+        if id_first is None:
+            id_first = (getattr(Go, "_identity_first", False)
+                        and getattr(Gd, "_identity_first", False))
+        # identity-support skip: every Chebyshev-family support stack starts
+        # with T_0 = I (graph/supports.py tags its outputs), whose products
+        # are the inputs themselves — the kernel layouts then EXCLUDE support
+        # 0 and the bindings add the identity terms directly (epilogue adds /
+        # strided copies), cutting 1/S of every axis contraction's FLOPs and
+        # staged bytes. `id_first` is the id_skip flag passed to ext calls.
+        self.id_first = bool(id_first) and self.S >= 2
         self._GoT = None
         self._A2T = None
         self._A2 = None
@@ -47,22 +58,33 @@ class GraphOperator:
         self._A28 = None
         self._A3T8 = None
 
+    def _go_k(self) -> torch.Tensor:
+        """Origin supports entering the kernels (support 0 dropped in
+        id_first mode)."""
+        return self.Go.narrow(-3, 1, self.S - 1) if self.id_first else self.Go
+
+    def _gd_k(self) -> torch.Tensor:
+        return self.Gd.narrow(-3, 1, self.S - 1) if self.id_first else self.Gd
+
     @property
     def GoT(self) -> torch.Tensor:
-        """GT[..., m, n] = Go[..., n, m] — mode-1 A operand."""
+        """GT[..., m, n] = Go[..., n, m] — mode-1 A operand (reduced in
+        id_first mode)."""
         if self._GoT is None:
-            self._GoT = self.Go.transpose(-2, -1).contiguous()
+            self._GoT = self._go_k().transpose(-2, -1).contiguous()
         return self._GoT
 
     @property
     def A2T(self) -> torch.Tensor:
-        """A2T[d, c*S+s] = Gd[s, c, d] — mode-2 A operand, (N, N*S)."""
+        """A2T[d, c*Se+s] = Gd_k[s, c, d] — mode-2 A operand, (N, N*Se)."""
         if self._A2T is None:
-            if self.Gd.dim() == 3:
-                self._A2T = self.Gd.permute(2, 1, 0).reshape(self.N, self.N * self.S).contiguous()
+            gd = self._gd_k()
+            Se = gd.shape[-3]
+            if gd.dim() == 3:
+                self._A2T = gd.permute(2, 1, 0).reshape(self.N, self.N * Se).contiguous()
             else:
-                B = self.Gd.shape[0]
-                self._A2T = self.Gd.permute(0, 3, 2, 1).reshape(B, self.N, self.N * self.S).contiguous()
+                B = gd.shape[0]
+                self._A2T = gd.permute(0, 3, 2, 1).reshape(B, self.N, self.N * Se).contiguous()
         return self._A2T
 
     @property
@@ -93,24 +115,28 @@ class GraphOperator:
 
     @property
     def A2(self) -> torch.Tensor:
-        """A2[c*S+s, d] = Gd[s, c, d] — mode-2 backward A operand, (N*S, N)."""
+        """A2[c*Se+s, d] = Gd_k[s, c, d] — mode-2 backward A operand."""
         if self._A2 is None:
-            if self.Gd.dim() == 3:
-                self._A2 = self.Gd.permute(1, 0, 2).reshape(self.N * self.S, self.N).contiguous()
+            gd = self._gd_k()
+            Se = gd.shape[-3]
+            if gd.dim() == 3:
+                self._A2 = gd.permute(1, 0, 2).reshape(self.N * Se, self.N).contiguous()
             else:
-                B = self.Gd.shape[0]
-                self._A2 = self.Gd.permute(0, 2, 1, 3).reshape(B, self.N * self.S, self.N).contiguous()
+                B = gd.shape[0]
+                self._A2 = gd.permute(0, 2, 1, 3).reshape(B, self.N * Se, self.N).contiguous()
         return self._A2
 
     @property
     def A3T(self) -> torch.Tensor:
-        """A3T[n, o*N+m] = Go[o, n, m] — mode-1 backward A operand, (N, S*N)."""
+        """A3T[n, o*N+m] = Go_k[o, n, m] — mode-1 backward A operand."""
         if self._A3T is None:
-            if self.Go.dim() == 3:
-                self._A3T = self.Go.permute(1, 0, 2).reshape(self.N, self.S * self.N).contiguous()
+            go = self._go_k()
+            Se = go.shape[-3]
+            if go.dim() == 3:
+                self._A3T = go.permute(1, 0, 2).reshape(self.N, Se * self.N).contiguous()
             else:
-                B = self.Go.shape[0]
-                self._A3T = self.Go.permute(0, 2, 1, 3).reshape(B, self.N, self.S * self.N).contiguous()
+                B = go.shape[0]
+                self._A3T = go.permute(0, 2, 1, 3).reshape(B, self.N, Se * self.N).contiguous()
         return self._A3T
 
 
@@ -139,11 +165,11 @@ class _BDGCNLayerFn(torch.autograd.Function):
         S = gop.S
         Hdim = W.shape[1]
 
-        U = ext.bdgcn_mode1(X, gop.GoT)  # (B,N,N,S,C)
+        U = ext.bdgcn_mode1(X, gop.GoT, gop.id_first)  # (B,N,N,S,C)
         Wre = eager.reorder_projection_weight(W, S, C).contiguous()
         Vflat = _row_gemm_chunked(ext, U.reshape(B * N * N, S * C), Wre, None, False)
         bias_f32 = bias.float().contiguous() if bias is not None else None
-        Y = ext.bdgcn_mode2(Vflat.view(B, N, N * S, Hdim), gop.A2T, bias_f32, relu, N, S)
+        Y = ext.bdgcn_mode2(Vflat.view(B, N, N * S, Hdim), gop.A2T, bias_f32, relu, N, S, gop.id_first)
 
         ctx.save_for_backward(U, Wre, Y)
         ctx.gop = gop
@@ -170,7 +196,7 @@ class _BDGCNLayerFn(torch.autograd.Function):
             dY = dH * (Y > 0).to(dH.dtype) if ctx.relu else dH
             dbias = dY.sum(dim=(0, 1, 2)).to(torch.float32) if ctx.has_bias else None
 
-        dV = ext.bdgcn_mode2_bwd(dY, gop.A2, S)  # (B,N,N,S,H)
+        dV = ext.bdgcn_mode2_bwd(dY, gop.A2, S, gop.id_first)  # (B,N,N,S,H)
         R = B * N * N
         dVflat = dV.reshape(R, S * Hdim)
         # dWre^T = dV^T @ U via the fused reduction kernel (f32 accumulate)
@@ -178,7 +204,7 @@ class _BDGCNLayerFn(torch.autograd.Function):
         dWre = dWreT.t().to(dH.dtype)
         dW = dWre.reshape(S, C, S, Hdim).permute(0, 2, 1, 3).reshape(S * S * C, Hdim)
         dU = _row_gemm_chunked(ext, dVflat, Wre.t().contiguous(), None, False)
-        dX = ext.bdgcn_mode1_bwd(dU.view(B, N, N, S, C), gop.A3T)
+        dX = ext.bdgcn_mode1_bwd(dU.view(B, N, N, S, C), gop.A3T, gop.id_first)
         db = dbias if ctx.has_bias else None
         return dX, dW, db, None, None
 
@@ -242,14 +268,14 @@ class _BDGCNLayerFp8Fn(torch.autograd.Function):
         # U stays fp8-ONLY: mode-1 writes half the bf16 path's output bytes
         # (U is the step's largest tensor), and backward's dW reduction reads
         # the fp8 U directly (red_gemm y_fp8 staging)
-        U8 = ext.bdgcn_mode1_fp8_train(X8, gop.GoT8)
+        U8 = ext.bdgcn_mode1_fp8_train(X8, gop.GoT8, gop.id_first)
         Wre = eager.reorder_projection_weight(W, S, C).contiguous()
         Wre8 = Wre.to(torch.float8_e4m3fn)
         V8 = ext.row_gemm_fp8(U8.reshape(B * N * N, S * C), Wre8)
         bias_f32 = bias.float().contiguous() if bias is not None else None
         Y, Y8 = ext.bdgcn_mode2_fp8_train(
             V8.view(B, N, N * S, Hdim), gop.A2T8, bias_f32, relu, N, S,
-            emit_twin,
+            emit_twin, gop.id_first,
         )
         ctx.save_for_backward(U8, Wre, Y)
         ctx.gop = gop
@@ -289,7 +315,7 @@ class _BDGCNLayerFp8Fn(torch.autograd.Function):
         if not ctx.has_bias:
             dbias = None
 
-        dV = ext.bdgcn_mode2_bwd_fp8(dY8, gop.A28, S, st["inv_y"])  # (B,N,N,S,H)
+        dV = ext.bdgcn_mode2_bwd_fp8(dY8, gop.A28, S, st["inv_y"], dY, gop.id_first)
         R = B * N * N
         dVflat = dV.reshape(R, S * Hdim)
         dWreT, _, _ = ext.red_gemm(dVflat, U8.reshape(R, S * C), False, None, 0, 0)
@@ -300,7 +326,7 @@ class _BDGCNLayerFp8Fn(torch.autograd.Function):
         dU8 = ext.row_gemm_fp8_out(dVflat, Wre.t().contiguous(),
                                    st["scale_u"], st["amax_u"])
         dX = ext.bdgcn_mode1_bwd_fp8(dU8.view(B, N, N, S, C), gop.A3T8,
-                                     st["inv_u"])
+                                     st["inv_u"], gop.id_first)
         return dX, dW, dbias, None, None, None, None, None
 
 
@@ -338,7 +364,7 @@ class _Mode1ProjFn(torch.autograd.Function):
         B, No, Nd, C = X.shape
         S = gop.S
         Hdim = W.shape[1]
-        U = ext.bdgcn_mode1(X, gop.GoT)  # (B, No, Nd, S, C)
+        U = ext.bdgcn_mode1(X, gop.GoT, gop.id_first)  # (B, No, Nd, S, C)
         Wre = eager.reorder_projection_weight(W, S, C).contiguous()
         Vflat = _row_gemm_chunked(ext, U.reshape(B * No * Nd, S * C), Wre, None, False)
         ctx.save_for_backward(U, Wre)
@@ -357,7 +383,7 @@ class _Mode1ProjFn(torch.autograd.Function):
         dWre = dWreT.t().to(dV.dtype)
         dW = dWre.reshape(S, C, S, Hdim).permute(0, 2, 1, 3).reshape(S * S * C, Hdim)
         dU = _row_gemm_chunked(ext, dVflat, Wre.t().contiguous(), None, False)
-        dX = ext.bdgcn_mode1_bwd(dU.view(B, No, Nd, S, C), ctx.gop.A3T)
+        dX = ext.bdgcn_mode1_bwd(dU.view(B, No, Nd, S, C), ctx.gop.A3T, ctx.gop.id_first)
         return dX, dW, None
 
 
@@ -371,7 +397,7 @@ class _Mode2BiasActFn(torch.autograd.Function):
         B, Nm, N, S, Hdim = V.shape
         bias_f32 = bias.float().contiguous() if bias is not None else None
         Y = ext.bdgcn_mode2(V.reshape(B, Nm, N * S, Hdim).contiguous(),
-                            gop.A2T, bias_f32, relu, N, S)
+                            gop.A2T, bias_f32, relu, N, S, gop.id_first)
         ctx.save_for_backward(Y)
         ctx.gop = gop
         ctx.relu = relu
@@ -393,7 +419,7 @@ class _Mode2BiasActFn(torch.autograd.Function):
         else:
             dY = dH * (Y > 0).to(dH.dtype) if ctx.relu else dH
             dbias = dY.sum(dim=(0, 1, 2)).to(torch.float32) if ctx.has_bias else None
-        dV = ext.bdgcn_mode2_bwd(dY, ctx.gop.A2, S)  # (B, Nm, N, S, H)
+        dV = ext.bdgcn_mode2_bwd(dY, ctx.gop.A2, S, ctx.gop.id_first)  # (B, Nm, N, S, H)
         return dV, dbias if ctx.has_bias else None, None, None
 
 

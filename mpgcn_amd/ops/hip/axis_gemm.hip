@@ -24,8 +24,9 @@
 #include "params.hpp"
 
 __device__ __forceinline__ long xrow_off(const AxisGemmParams& p, int k) {
-    return p.kdiv > 1 ? (long)(k / p.kdiv) * p.k_hi + (long)(k % p.kdiv) * p.k_lo
-                      : (long)k * p.k_lo;
+    return p.k_base +
+           (p.kdiv > 1 ? (long)(k / p.kdiv) * p.k_hi + (long)(k % p.kdiv) * p.k_lo
+                       : (long)k * p.k_lo);
 }
 __device__ __forceinline__ long xcol_off(const AxisGemmParams& p, int q) {
     return p.qdiv ? (long)(q / p.qdiv) * p.q_hi + (long)(q % p.qdiv) : (long)q;
@@ -57,6 +58,10 @@ __launch_bounds__(WVM * WVN * 64) __global__ void axis_gemm_kernel(AxisGemmParam
     const long a_base = (long)(inst / p.a_div) * p.a_bs1 + (long)(inst % p.a_div) * p.a_bs2;
     const long x_base = (long)(inst / p.x_div) * p.x_bs1 + (long)(inst % p.x_div) * p.x_bs2;
     const long o_base = (long)(inst / p.o_div) * p.o_bs1 + (long)(inst % p.o_div) * p.o_bs2;
+    const T* __restrict__ CS = (const T*)p.CSUB;
+    const long cs_base = p.CSUB
+        ? (long)(inst / p.cs_div) * p.cs_bs1 + (long)(inst % p.cs_div) * p.cs_bs2
+        : 0;
 
     const int tm = blockIdx.x / p.tiles_l;
     const int tl = blockIdx.x % p.tiles_l;
@@ -158,15 +163,17 @@ __launch_bounds__(WVM * WVN * 64) __global__ void axis_gemm_kernel(AxisGemmParam
         if (BUFS == 2) cur ^= 1;
     }
 
-    // ---- epilogue: scale + bias + activation + strided store ----
-    // per-lane q depends only on nf: hoist the column-offset division and the
-    // bias load out of the (mf, r) loops
+    // ---- epilogue: v = (alpha*acc + cs_beta*csub)*scale + bias; act ----
+    // per-lane q depends only on nf: hoist the column-offset division, the
+    // bias load and the csub column base out of the (mf, r) loops
     const float osc = p.scale ? *p.scale : 1.f;  // fp8 gradient descale
+    const float alpha = p.alpha == 0.f ? 1.f : p.alpha;
 #pragma unroll
     for (int nf = 0; nf < AN; ++nf) {
         const int q = l0 + wn + nf * 16 + lrow;
         if (q >= p.L) continue;
         const long oc = o_base + ocol_off(p, q);
+        const long cs_col = CS ? cs_base + xcol_off(p, q) : 0;
         const float bv = p.bias ? p.bias[p.bias_mod ? q % p.bias_mod : q] : 0.f;
 #pragma unroll
         for (int mf = 0; mf < AM; ++mf) {
@@ -174,10 +181,17 @@ __launch_bounds__(WVM * WVN * 64) __global__ void axis_gemm_kernel(AxisGemmParam
             for (int r = 0; r < 4; ++r) {
                 const int m = m0 + wm + mf * 16 + kgrp * 4 + r;
                 if (m < p.M) {
-                    float v = acc[mf][nf][r] * osc + bv;
+                    float v = alpha * acc[mf][nf][r];
+                    if (CS)  // identity-support term / polynomial recurrence
+                        v += p.cs_beta * to_f32(CS[cs_col + (long)m * p.cs_row]);
+                    v = v * osc + bv;
                     if (p.relu) v = fmaxf(v, 0.f);
-                    O[oc + (long)m * p.o_row] = from_f32<OT>(v);
-                    if (p.OUT2) O2[oc + (long)m * p.o_row] = from_f32<OT2>(v);
+                    const long orow = p.o_mdiv
+                        ? (long)(m / p.o_mdiv) * p.o_m_hi +
+                              (long)(m % p.o_mdiv) * p.o_m_lo + p.o_m_base
+                        : (long)m * p.o_row;
+                    O[oc + orow] = from_f32<OT>(v);
+                    if (p.OUT2) O2[oc + orow] = from_f32<OT2>(v);
                 }
             }
         }

@@ -73,6 +73,11 @@ __device__ __forceinline__ unsigned char from_f32<unsigned char>(float v) {
 
 __device__ __forceinline__ float to_f32(__bf16 v) { return (float)v; }
 __device__ __forceinline__ float to_f32(float v) { return v; }
+__device__ __forceinline__ float to_f32(unsigned char v) {
+    // OCP e4m3 byte -> f32 (v_cvt_pk_f32_fp8, low half)
+    typedef __attribute__((ext_vector_type(2))) float _f2;
+    return ((_f2)__builtin_amdgcn_cvt_pk_f32_fp8((int)v, false))[0];
+}
 
 // Fast transcendentals: v_rcp_f32-based sigmoid/tanh (no IEEE division
 // sequences, no libm branches — hipcc otherwise emits v_div_scale/div_fixup

@@ -39,25 +39,32 @@ const float* bias_ptr(const c10::optional<torch::Tensor>& b) {
 // ('bncl,bnm->bmcl') pair (reference MPGCN.py:30,38), batched over supports.
 // Rectangular: X may be destination-sharded, (B, No, Nd, C) with Nd != No
 // (the region-partition path, mpgcn_amd/parallel/region.py).
-torch::Tensor bdgcn_mode1(torch::Tensor X, torch::Tensor GT) {
+// id_skip: the FIRST support is the identity matrix (T_0 = I for every
+// Chebyshev-family kernel type, GCN.py:128-138), so its product is X itself
+// — GT is then the REDUCED stack WITHOUT support 0 (S-1 supports), the GEMM
+// fills only slots 1..S-1 of U, and the caller copies X into slot 0
+// (ops/functional.py). Cuts 1/S of the contraction's FLOPs and bytes.
+torch::Tensor bdgcn_mode1(torch::Tensor X, torch::Tensor GT, bool id_skip) {
     check_in(X, "X");
     check_in(GT, "GT");
     const bool dyn = GT.dim() == 4;
     const long B = X.size(0), No = X.size(1), Nd = X.size(2), C = X.size(3);
-    const long S = dyn ? GT.size(1) : GT.size(0);
+    const long Se = dyn ? GT.size(1) : GT.size(0);  // supports in the GEMM
+    const long S = id_skip ? Se + 1 : Se;           // supports in U
     TORCH_CHECK(GT.size(-1) == No && GT.size(-2) == No, "shape mismatch");
     TORCH_CHECK(!dyn || GT.size(0) == B, "dynamic GT batch mismatch");
-    TORCH_CHECK(B * S <= 65535, "too many instances");
+    TORCH_CHECK(B * Se <= 65535, "too many instances");
     auto U = torch::empty({B, No, Nd, S, C}, X.options());
 
     AxisGemmParams p{};
     p.AT = GT.data_ptr();
     p.X = X.data_ptr();
-    p.OUT = U.data_ptr();
+    p.OUT = id_skip ? (void*)((char*)U.data_ptr() + C * U.element_size())
+                    : U.data_ptr();
     p.M = (int)No; p.K = (int)No; p.L = (int)(Nd * C);
-    p.a_div = (int)S; p.a_bs1 = dyn ? S * No * No : 0; p.a_bs2 = No * No;
-    p.x_div = (int)S; p.x_bs1 = No * Nd * C; p.x_bs2 = 0;
-    p.o_div = (int)S; p.o_bs1 = No * Nd * S * C; p.o_bs2 = C;
+    p.a_div = (int)Se; p.a_bs1 = dyn ? Se * No * No : 0; p.a_bs2 = No * No;
+    p.x_div = (int)Se; p.x_bs1 = No * Nd * C; p.x_bs2 = 0;
+    p.o_div = (int)Se; p.o_bs1 = No * Nd * S * C; p.o_bs2 = C;
     p.kdiv = 1; p.k_lo = Nd * C;
     p.qdiv = 0;
     p.o_row = Nd * S * C;
@@ -65,23 +72,29 @@ torch::Tensor bdgcn_mode1(torch::Tensor X, torch::Tensor GT) {
     const int ch = chunk_elems(X);
     p.a_vec = (No % ch == 0);
     p.x_vec = ((Nd * C) % ch == 0) && (C % ch == 0);
-    axis_gemm_launch(p, (int)(B * S), is_f32(X), stream());
+    axis_gemm_launch(p, (int)(B * Se), is_f32(X), stream());
+    if (id_skip)  // slot 0 = identity product = X
+        U.select(3, 0).copy_(X);
     return U;
 }
 
 // Y[b,m,d,h] = sum_{cs} A2T[(b,)d,cs] V[b,m,cs,h]  (+ bias + optional ReLU).
 // V: (B, N, N*S, H) flat (i.e. (B,N,N,S,H) contiguous); A2T: (N, N*S) static
 // or (B, N, N*S) dynamic, A2T[d, c*S+s] = Gd[s, c, d]. Out: (B, N, N, H).
+// id_skip: A2T is the REDUCED layout A2T'[d, c*(S-1)+s'] = Gd[s'+1, c, d];
+// the s = 0 (identity) term Y += V[b, m, (d, 0), h] is added by the epilogue
+// (CSUB) while the GEMM contracts only the s >= 1 columns.
 torch::Tensor bdgcn_mode2(torch::Tensor V, torch::Tensor A2T,
                           c10::optional<torch::Tensor> bias, bool relu,
-                          long N, long S) {
+                          long N, long S, bool id_skip) {
     check_in(V, "V");
     check_in(A2T, "A2T");
     const bool dyn = A2T.dim() == 3;
     const long B = V.size(0);
     const long Nm = V.size(1);  // origin rows held locally (== N unsharded)
     const long H = V.size(-1);
-    TORCH_CHECK(A2T.size(-1) == N * S && A2T.size(-2) == N, "A2T shape mismatch");
+    const long Se = id_skip ? S - 1 : S;
+    TORCH_CHECK(A2T.size(-1) == N * Se && A2T.size(-2) == N, "A2T shape mismatch");
     TORCH_CHECK(B <= 65535, "too many instances");
     auto Y = torch::empty({B, Nm, N, H}, V.options());
 
@@ -93,18 +106,26 @@ torch::Tensor bdgcn_mode2(torch::Tensor V, torch::Tensor A2T,
     p.X = V.data_ptr();
     p.OUT = Y.data_ptr();
     p.bias = bias_ptr(bias);
-    p.M = (int)N; p.K = (int)(N * S); p.L = (int)(Nm * H);
-    p.a_div = 1; p.a_bs1 = dyn ? N * N * S : 0; p.a_bs2 = 0;
+    p.M = (int)N; p.K = (int)(N * Se); p.L = (int)(Nm * H);
+    p.a_div = 1; p.a_bs1 = dyn ? N * N * Se : 0; p.a_bs2 = 0;
     p.x_div = 1; p.x_bs1 = Nm * N * S * H; p.x_bs2 = 0;
     p.o_div = 1; p.o_bs1 = Nm * N * H; p.o_bs2 = 0;
-    p.kdiv = 1; p.k_lo = H;
+    if (id_skip) {
+        p.kdiv = (int)Se; p.k_hi = S * H; p.k_lo = H; p.k_base = H;
+        p.CSUB = V.data_ptr();
+        p.cs_div = 1; p.cs_bs1 = Nm * N * S * H; p.cs_bs2 = 0;
+        p.cs_row = S * H;
+        p.cs_beta = 1.f;
+    } else {
+        p.kdiv = 1; p.k_lo = H;
+    }
     p.qdiv = (int)H; p.q_hi = N * S * H;   // q = (m, h)
     p.o_row = H;                           // out row = d
     p.ogdiv = (int)H; p.og_hi = N * H;
     p.relu = relu ? 1 : 0;
     p.bias_mod = (int)H;
     const int ch = chunk_elems(V);
-    p.a_vec = ((N * S) % ch == 0);
+    p.a_vec = ((N * Se) % ch == 0);
     p.x_vec = (H % ch == 0);
     axis_gemm_launch(p, (int)B, is_f32(V), stream());
     return Y;
@@ -114,13 +135,19 @@ torch::Tensor bdgcn_mode2(torch::Tensor V, torch::Tensor A2T,
 // dY: (B, N, N, H); A2: (N*S, N) or (B, N*S, N), A2[c*S+s, d] = Gd[s, c, d].
 // Rectangular: dY may be origin-sharded (B, Nm, N, H) with Nm != N
 // (region partition, mpgcn_amd/parallel/region.py); N comes from A2.
-torch::Tensor bdgcn_mode2_bwd(torch::Tensor dY, torch::Tensor A2, long S) {
+// id_skip: A2 is the REDUCED layout A2'[c*(S-1)+s', d] = Gd[s'+1, c, d];
+// the GEMM writes only the s >= 1 rows of dV (output-row remap) and the
+// caller copies dY into the s = 0 rows (dV[..., 0, :] = dY, the identity
+// support's exact gradient).
+torch::Tensor bdgcn_mode2_bwd(torch::Tensor dY, torch::Tensor A2, long S,
+                              bool id_skip) {
     check_in(dY, "dY");
     check_in(A2, "A2");
     const bool dyn = A2.dim() == 3;
     const long B = dY.size(0), Nm = dY.size(1), H = dY.size(3);
     const long N = A2.size(-1);
-    TORCH_CHECK(A2.size(-2) == N * S && dY.size(2) == N, "A2 shape mismatch");
+    const long Se = id_skip ? S - 1 : S;
+    TORCH_CHECK(A2.size(-2) == N * Se && dY.size(2) == N, "A2 shape mismatch");
     TORCH_CHECK(B <= 65535, "too many instances");
     auto dV = torch::empty({B, Nm, N, S, H}, dY.options());
 
@@ -130,31 +157,41 @@ torch::Tensor bdgcn_mode2_bwd(torch::Tensor dY, torch::Tensor A2, long S) {
     p.AT = A2.data_ptr();
     p.X = dY.data_ptr();
     p.OUT = dV.data_ptr();
-    p.M = (int)(N * S); p.K = (int)N; p.L = (int)(Nm * H);
-    p.a_div = 1; p.a_bs1 = dyn ? N * S * N : 0; p.a_bs2 = 0;
+    p.M = (int)(N * Se); p.K = (int)N; p.L = (int)(Nm * H);
+    p.a_div = 1; p.a_bs1 = dyn ? N * Se * N : 0; p.a_bs2 = 0;
     p.x_div = 1; p.x_bs1 = Nm * N * H; p.x_bs2 = 0;
     p.o_div = 1; p.o_bs1 = Nm * N * S * H; p.o_bs2 = 0;
     p.kdiv = 1; p.k_lo = H;
     p.qdiv = (int)H; p.q_hi = N * H;       // q = (m, h)
     p.o_row = H;                           // out row = cs
+    if (id_skip) {
+        // GEMM row m = (c, s') lands in dV slot (c, s'+1)
+        p.o_mdiv = (int)Se; p.o_m_hi = S * H; p.o_m_lo = H; p.o_m_base = H;
+    }
     p.ogdiv = (int)H; p.og_hi = N * S * H;
     const int ch = chunk_elems(dY);
     p.a_vec = (N % ch == 0);
     p.x_vec = (H % ch == 0);
     axis_gemm_launch(p, (int)B, is_f32(dY), stream());
+    if (id_skip)
+        dV.select(3, 0).copy_(dY);
     return dV;
 }
 
 // dX[b,n,d,l] = sum_{om} A3T[(b,)n,o*N+m] dU[b,m,d,o,l].
 // dU: (B, N, N, S, C); A3T: (N, S*N) or (B, N, S*N), A3T[n, o*N+m] = G[o,n,m].
 // Rectangular: dU may be destination-sharded (B, No, Nd, S, C), Nd != No.
-torch::Tensor bdgcn_mode1_bwd(torch::Tensor dU, torch::Tensor A3T) {
+// id_skip: A3T is the REDUCED layout A3T'[n, o'*No+m] = Go[o'+1, n, m]; the
+// o = 0 (identity) term dX += dU[b, n, d, 0, l] is added by the epilogue.
+torch::Tensor bdgcn_mode1_bwd(torch::Tensor dU, torch::Tensor A3T,
+                              bool id_skip) {
     check_in(dU, "dU");
     check_in(A3T, "A3T");
     const bool dyn = A3T.dim() == 3;
     const long B = dU.size(0), No = dU.size(1), Nd = dU.size(2);
     const long S = dU.size(3), C = dU.size(4);
-    TORCH_CHECK(A3T.size(-2) == No && A3T.size(-1) == S * No, "A3T shape mismatch");
+    const long Se = id_skip ? S - 1 : S;
+    TORCH_CHECK(A3T.size(-2) == No && A3T.size(-1) == Se * No, "A3T shape mismatch");
     TORCH_CHECK(B <= 65535, "too many instances");
     auto dX = torch::empty({B, No, Nd, C}, dU.options());
 
@@ -162,16 +199,23 @@ torch::Tensor bdgcn_mode1_bwd(torch::Tensor dU, torch::Tensor A3T) {
     p.AT = A3T.data_ptr();
     p.X = dU.data_ptr();
     p.OUT = dX.data_ptr();
-    p.M = (int)No; p.K = (int)(S * No); p.L = (int)(Nd * C);
-    p.a_div = 1; p.a_bs1 = dyn ? No * S * No : 0; p.a_bs2 = 0;
+    p.M = (int)No; p.K = (int)(Se * No); p.L = (int)(Nd * C);
+    p.a_div = 1; p.a_bs1 = dyn ? No * Se * No : 0; p.a_bs2 = 0;
     p.x_div = 1; p.x_bs1 = No * Nd * S * C; p.x_bs2 = 0;
     p.o_div = 1; p.o_bs1 = No * Nd * C; p.o_bs2 = 0;
     p.kdiv = (int)No; p.k_hi = C; p.k_lo = Nd * S * C;  // k = o*No + m
+    if (id_skip) {
+        p.k_base = C;  // k = (o', m) addresses support o'+1
+        p.CSUB = dU.data_ptr();
+        p.cs_div = 1; p.cs_bs1 = No * Nd * S * C; p.cs_bs2 = 0;
+        p.cs_row = Nd * S * C;
+        p.cs_beta = 1.f;
+    }
     p.qdiv = (int)C; p.q_hi = S * C;                    // q = d*C + l
     p.o_row = Nd * C;
     p.ogdiv = 0;
     const int ch = chunk_elems(dU);
-    p.a_vec = ((S * No) % ch == 0);
+    p.a_vec = ((Se * No) % ch == 0);
     p.x_vec = (C % ch == 0);
     axis_gemm_launch(p, (int)B, is_f32(dU), stream());
     return dX;
@@ -523,15 +567,17 @@ void check_fp8(const torch::Tensor& t, const char* name) {
 // is vector-only, so the column extent Nd*C must be a multiple of 256 and C
 // a multiple of 16 (flagship and large-N configs satisfy this; the Python
 // layer falls back to bf16 otherwise).
-torch::Tensor bdgcn_mode1_fp8_train(torch::Tensor X8, torch::Tensor GT8) {
+torch::Tensor bdgcn_mode1_fp8_train(torch::Tensor X8, torch::Tensor GT8,
+                                    bool id_skip) {
     check_fp8(X8, "X8");
     check_fp8(GT8, "GT8");
     const bool dyn = GT8.dim() == 4;
     const long B = X8.size(0), No = X8.size(1), Nd = X8.size(2), C = X8.size(3);
-    const long S = dyn ? GT8.size(1) : GT8.size(0);
+    const long Se = dyn ? GT8.size(1) : GT8.size(0);
+    const long S = id_skip ? Se + 1 : Se;
     TORCH_CHECK(GT8.size(-1) == No && GT8.size(-2) == No, "shape mismatch");
     TORCH_CHECK(!dyn || GT8.size(0) == B, "dynamic GT8 batch mismatch");
-    TORCH_CHECK(B * S <= 65535, "too many instances");
+    TORCH_CHECK(B * Se <= 65535, "too many instances");
     TORCH_CHECK((Nd * C) % 256 == 0 && C % 16 == 0,
                 "fp8 mode-1 needs (Nd*C) % 256 == 0 and C % 16 == 0");
     auto U8 = torch::empty({B, No, Nd, S, C}, X8.options());
@@ -539,18 +585,20 @@ torch::Tensor bdgcn_mode1_fp8_train(torch::Tensor X8, torch::Tensor GT8) {
     AxisGemmParams p{};
     p.AT = GT8.data_ptr();
     p.X = X8.data_ptr();
-    p.OUT = U8.data_ptr();
+    p.OUT = id_skip ? (void*)((char*)U8.data_ptr() + C) : U8.data_ptr();
     p.M = (int)No; p.K = (int)No; p.L = (int)(Nd * C);
-    p.a_div = (int)S; p.a_bs1 = dyn ? S * No * No : 0; p.a_bs2 = No * No;
-    p.x_div = (int)S; p.x_bs1 = No * Nd * C; p.x_bs2 = 0;
-    p.o_div = (int)S; p.o_bs1 = No * Nd * S * C; p.o_bs2 = C;
+    p.a_div = (int)Se; p.a_bs1 = dyn ? Se * No * No : 0; p.a_bs2 = No * No;
+    p.x_div = (int)Se; p.x_bs1 = No * Nd * C; p.x_bs2 = 0;
+    p.o_div = (int)Se; p.o_bs1 = No * Nd * S * C; p.o_bs2 = C;
     p.kdiv = 1; p.k_lo = Nd * C;
     p.qdiv = 0;
     p.o_row = Nd * S * C;
     p.ogdiv = (int)C; p.og_hi = S * C;
     p.a_vec = (No % 16 == 0);
     p.x_vec = 1;
-    axis_gemm_fp8_launch(p, (int)(B * S), 0, stream());
+    axis_gemm_fp8_launch(p, (int)(B * Se), 0, stream());
+    if (id_skip)
+        U8.select(3, 0).copy_(X8);
     return U8;
 }
 
@@ -561,12 +609,13 @@ std::vector<torch::Tensor> bdgcn_mode2_fp8_train(torch::Tensor V8,
                                                  torch::Tensor A2T8,
                                                  c10::optional<torch::Tensor> bias,
                                                  bool relu, long N, long S,
-                                                 bool want_twin) {
+                                                 bool want_twin, bool id_skip) {
     check_fp8(V8, "V8");
     check_fp8(A2T8, "A2T8");
     const bool dyn = A2T8.dim() == 3;
     const long B = V8.size(0), Nm = V8.size(1), H = V8.size(-1);
-    TORCH_CHECK(A2T8.size(-1) == N * S && A2T8.size(-2) == N, "A2T8 shape");
+    const long Se = id_skip ? S - 1 : S;
+    TORCH_CHECK(A2T8.size(-1) == N * Se && A2T8.size(-2) == N, "A2T8 shape");
     TORCH_CHECK(B <= 65535, "too many instances");
     TORCH_CHECK((Nm * H) % 256 == 0 && H % 16 == 0,
                 "fp8 mode-2 needs (Nm*H) % 256 == 0 and H % 16 == 0");
@@ -582,17 +631,25 @@ std::vector<torch::Tensor> bdgcn_mode2_fp8_train(torch::Tensor V8,
     p.OUT = Y.data_ptr();
     p.OUT2 = want_twin ? Y8.data_ptr() : nullptr;
     p.bias = bias_ptr(bias);
-    p.M = (int)N; p.K = (int)(N * S); p.L = (int)(Nm * H);
-    p.a_div = 1; p.a_bs1 = dyn ? N * N * S : 0; p.a_bs2 = 0;
+    p.M = (int)N; p.K = (int)(N * Se); p.L = (int)(Nm * H);
+    p.a_div = 1; p.a_bs1 = dyn ? N * N * Se : 0; p.a_bs2 = 0;
     p.x_div = 1; p.x_bs1 = Nm * N * S * H; p.x_bs2 = 0;
     p.o_div = 1; p.o_bs1 = Nm * N * H; p.o_bs2 = 0;
-    p.kdiv = 1; p.k_lo = H;
+    if (id_skip) {
+        p.kdiv = (int)Se; p.k_hi = S * H; p.k_lo = H; p.k_base = H;
+        p.CSUB = V8.data_ptr();
+        p.cs_div = 1; p.cs_bs1 = Nm * N * S * H; p.cs_bs2 = 0;
+        p.cs_row = S * H;
+        p.cs_beta = 1.f;
+    } else {
+        p.kdiv = 1; p.k_lo = H;
+    }
     p.qdiv = (int)H; p.q_hi = N * S * H;
     p.o_row = H;
     p.ogdiv = (int)H; p.og_hi = N * H;
     p.relu = relu ? 1 : 0;
     p.bias_mod = (int)H;
-    p.a_vec = ((N * S) % 16 == 0);
+    p.a_vec = ((N * Se) % 16 == 0);
     p.x_vec = 1;
     axis_gemm_fp8_launch(p, (int)B, 2, stream());
     return {Y, Y8};
@@ -605,7 +662,8 @@ std::vector<torch::Tensor> bdgcn_mode2_fp8_train(torch::Tensor V8,
 // writes bf16 dV. A28[cs, d] = fp8(Gd[s, c, d]) — supports are O(1), unit
 // scale. Same contraction as bdgcn_mode2_bwd (bf16 twin of this path).
 torch::Tensor bdgcn_mode2_bwd_fp8(torch::Tensor dY8, torch::Tensor A28,
-                                  long S, torch::Tensor inv_scale) {
+                                  long S, torch::Tensor inv_scale,
+                                  torch::Tensor dY_bf16, bool id_skip) {
     check_fp8(dY8, "dY8");
     check_fp8(A28, "A28");
     TORCH_CHECK(inv_scale.is_cuda() && inv_scale.scalar_type() == torch::kFloat,
@@ -613,7 +671,8 @@ torch::Tensor bdgcn_mode2_bwd_fp8(torch::Tensor dY8, torch::Tensor A28,
     const bool dyn = A28.dim() == 3;
     const long B = dY8.size(0), Nm = dY8.size(1), H = dY8.size(3);
     const long N = A28.size(-1);
-    TORCH_CHECK(A28.size(-2) == N * S && dY8.size(2) == N, "A28 shape");
+    const long Se = id_skip ? S - 1 : S;
+    TORCH_CHECK(A28.size(-2) == N * Se && dY8.size(2) == N, "A28 shape");
     TORCH_CHECK(B <= 65535, "too many instances");
     TORCH_CHECK((Nm * H) % 256 == 0 && H % 16 == 0, "fp8 bwd shape gate");
     auto dV = torch::empty({B, Nm, N, S, H},
@@ -623,24 +682,29 @@ torch::Tensor bdgcn_mode2_bwd_fp8(torch::Tensor dY8, torch::Tensor A28,
     p.X = dY8.data_ptr();
     p.OUT = dV.data_ptr();
     p.scale = inv_scale.data_ptr<float>();
-    p.M = (int)(N * S); p.K = (int)N; p.L = (int)(Nm * H);
-    p.a_div = 1; p.a_bs1 = dyn ? N * S * N : 0; p.a_bs2 = 0;
+    p.M = (int)(N * Se); p.K = (int)N; p.L = (int)(Nm * H);
+    p.a_div = 1; p.a_bs1 = dyn ? N * Se * N : 0; p.a_bs2 = 0;
     p.x_div = 1; p.x_bs1 = Nm * N * H; p.x_bs2 = 0;
     p.o_div = 1; p.o_bs1 = Nm * N * S * H; p.o_bs2 = 0;
     p.kdiv = 1; p.k_lo = H;
     p.qdiv = (int)H; p.q_hi = N * H;
     p.o_row = H;
+    if (id_skip) {
+        p.o_mdiv = (int)Se; p.o_m_hi = S * H; p.o_m_lo = H; p.o_m_base = H;
+    }
     p.ogdiv = (int)H; p.og_hi = N * S * H;
     p.a_vec = (N % 16 == 0);
     p.x_vec = 1;
     axis_gemm_fp8_launch(p, (int)B, 2, stream());
+    if (id_skip)  // identity-support gradient rows: exact bf16 dY, unquantized
+        dV.select(3, 0).copy_(dY_bf16);
     return dV;
 }
 
 // fp8 gradient contraction dX = mode1_bwd(dU8, A3T8), scaled like
 // bdgcn_mode2_bwd_fp8. dU8 = fp8(dU * s); epilogue descales and writes bf16.
 torch::Tensor bdgcn_mode1_bwd_fp8(torch::Tensor dU8, torch::Tensor A3T8,
-                                  torch::Tensor inv_scale) {
+                                  torch::Tensor inv_scale, bool id_skip) {
     check_fp8(dU8, "dU8");
     check_fp8(A3T8, "A3T8");
     TORCH_CHECK(inv_scale.is_cuda() && inv_scale.scalar_type() == torch::kFloat,
@@ -648,7 +712,8 @@ torch::Tensor bdgcn_mode1_bwd_fp8(torch::Tensor dU8, torch::Tensor A3T8,
     const bool dyn = A3T8.dim() == 3;
     const long B = dU8.size(0), No = dU8.size(1), Nd = dU8.size(2);
     const long S = dU8.size(3), C = dU8.size(4);
-    TORCH_CHECK(A3T8.size(-2) == No && A3T8.size(-1) == S * No, "A3T8 shape");
+    const long Se = id_skip ? S - 1 : S;
+    TORCH_CHECK(A3T8.size(-2) == No && A3T8.size(-1) == Se * No, "A3T8 shape");
     TORCH_CHECK(B <= 65535, "too many instances");
     TORCH_CHECK((Nd * C) % 256 == 0 && C % 16 == 0, "fp8 bwd shape gate");
     auto dX = torch::empty({B, No, Nd, C}, dU8.options().dtype(torch::kBFloat16));
@@ -657,15 +722,22 @@ torch::Tensor bdgcn_mode1_bwd_fp8(torch::Tensor dU8, torch::Tensor A3T8,
     p.X = dU8.data_ptr();
     p.OUT = dX.data_ptr();
     p.scale = inv_scale.data_ptr<float>();
-    p.M = (int)No; p.K = (int)(S * No); p.L = (int)(Nd * C);
-    p.a_div = 1; p.a_bs1 = dyn ? No * S * No : 0; p.a_bs2 = 0;
+    p.M = (int)No; p.K = (int)(Se * No); p.L = (int)(Nd * C);
+    p.a_div = 1; p.a_bs1 = dyn ? No * Se * No : 0; p.a_bs2 = 0;
     p.x_div = 1; p.x_bs1 = No * Nd * S * C; p.x_bs2 = 0;
     p.o_div = 1; p.o_bs1 = No * Nd * C; p.o_bs2 = 0;
     p.kdiv = (int)No; p.k_hi = C; p.k_lo = Nd * S * C;  // k = o*No + m
+    if (id_skip) {
+        p.k_base = C;
+        p.CSUB = dU8.data_ptr();
+        p.cs_div = 1; p.cs_bs1 = No * Nd * S * C; p.cs_bs2 = 0;
+        p.cs_row = Nd * S * C;
+        p.cs_beta = 1.f;
+    }
     p.qdiv = (int)C; p.q_hi = S * C;                    // q = d*C + l
     p.o_row = Nd * C;
     p.ogdiv = 0;
-    p.a_vec = ((S * No) % 16 == 0);
+    p.a_vec = ((Se * No) % 16 == 0);
     p.x_vec = 1;
     axis_gemm_fp8_launch(p, (int)B, 2, stream());
     return dX;

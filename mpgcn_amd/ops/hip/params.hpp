@@ -17,9 +17,17 @@ struct AxisGemmParams {
     int x_div; long x_bs1, x_bs2;
     int o_div; long o_bs1, o_bs2;
     int kdiv; long k_hi, k_lo;
+    long k_base;         // constant X-row offset (identity-support k skip)
     int qdiv; long q_hi;
     long o_row;
+    int o_mdiv; long o_m_hi, o_m_lo, o_m_base;  // OUT row mapping (0: m*o_row)
     int ogdiv; long og_hi;
+    // epilogue: v = (alpha*acc + cs_beta*csub[..]) * (*scale) + bias
+    const void* CSUB;    // same element type as X; nullable
+    int cs_div; long cs_bs1, cs_bs2;  // CSUB instance striding
+    long cs_row;         // CSUB row stride (index = cs_base + xcol_off(q) + m*cs_row)
+    float alpha;         // 0 is treated as 1 by the launchers
+    float cs_beta;
     int relu;
     int bias_mod;  // bias index = bias_mod ? q % bias_mod : q
     int a_vec, x_vec;

@@ -52,10 +52,13 @@ class Forecaster:
         self.model.load_state_dict(ckpt["state_dict"])
         self.model.eval()
 
+        from mpgcn_amd.graph.supports import tag_like
+
         adj = data["adj"].float().to(self.device)
-        self.G_static = build_supports(
+        sup = build_supports(
             adj.unsqueeze(0), params["kernel_type"], params["cheby_order"]
-        ).squeeze(0)
+        )
+        self.G_static = tag_like(sup.squeeze(0), sup)
         # (N, N, 7) -> per-dow support stacks, built once at startup
         O_dyn = data["O_dyn_G"].permute(2, 0, 1).float().to(self.device)
         D_dyn = data["D_dyn_G"].permute(2, 0, 1).float().to(self.device)
@@ -67,9 +70,10 @@ class Forecaster:
             # (train/trainer.py _graph_list), so a 3-perspective checkpoint is
             # served with the graphs it was trained on
             corr = data["O_dyn_G"].float().mean(dim=-1).to(self.device)
-            self.G_corr = build_supports(
+            csup = build_supports(
                 corr.unsqueeze(0), params["kernel_type"], params["cheby_order"]
-            ).squeeze(0)
+            )
+            self.G_corr = tag_like(csup.squeeze(0), csup)
 
     @torch.no_grad()
     def forecast(self, x_seq: torch.Tensor, dow: int, horizon: int = 1) -> torch.Tensor:

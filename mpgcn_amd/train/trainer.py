@@ -60,9 +60,12 @@ class ModelTrainer:
         if isinstance(adj, np.ndarray):
             adj = torch.from_numpy(adj)
         adj = adj.float().to(self.device)
-        return build_supports(
+        sup = build_supports(
             adj.unsqueeze(0), self.params["kernel_type"], self.params["cheby_order"]
-        ).squeeze(0)
+        )
+        from mpgcn_amd.graph.supports import tag_like
+
+        return tag_like(sup.squeeze(0), sup)
 
     def preprocess_dynamic_graph(self, dyn_G: torch.Tensor) -> torch.Tensor:
         """(B, N, N) raw flow -> (B, K, N, N) supports, batched, on device."""
@@ -121,11 +124,14 @@ class ModelTrainer:
         G_list = [self.G, dyn]
         if int(self.params.get("perspectives", 2)) == 3:
             if not hasattr(self, "_G_corr"):
+                from mpgcn_amd.graph.supports import tag_like
+
                 corr = self._data_O_dyn.mean(dim=-1).to(self.device)
-                self._G_corr = build_supports(
+                sup = build_supports(
                     corr.unsqueeze(0), self.params["kernel_type"],
                     self.params["cheby_order"],
-                ).squeeze(0)
+                )
+                self._G_corr = tag_like(sup.squeeze(0), sup)
             G_list.append(self._G_corr)
         return G_list
 

@@ -236,6 +236,72 @@ def test_reg_lstm_bf16_grads_match_autograd(T):
         assert err < 3e-2, f"{name}: rel grad err {err:.4f}"
 
 
+@pytest.mark.parametrize("dyn", [False, True])
+def test_identity_skip_matches_full_path(dyn):
+    """id_first mode (support 0 = I skipped: reduced operand layouts,
+    epilogue identity adds, strided identity copies) must reproduce the full
+    contraction bit-for... closely: the only difference is the f32 epilogue
+    add of the exact identity term vs an MFMA against a quantized-to-bf16 I
+    (which IS exact for I), so forward and all grads match tightly."""
+    from mpgcn_amd.graph import build_supports
+    from mpgcn_amd.ops import GraphOperator, bdgcn_layer
+
+    torch.manual_seed(9)
+    S, C, Hd, Bn, N_ = 3, 32, 32, 2, 64
+    X = torch.randn(Bn, N_, N_, C, device=DEV).bfloat16()
+    nb = Bn if dyn else 1
+    Go = build_supports(torch.rand(nb, N_, N_, device=DEV),
+                        "random_walk_diffusion", S - 1)
+    Gd = build_supports(torch.rand(nb, N_, N_, device=DEV),
+                        "random_walk_diffusion", S - 1)
+    if not dyn:
+        Go, Gd = Go.squeeze(0), Gd.squeeze(0)
+    Go, Gd = Go.bfloat16().contiguous(), Gd.bfloat16().contiguous()
+    W = (0.1 * torch.randn(C * S * S, Hd, device=DEV)).bfloat16()
+    b = 0.05 * torch.randn(Hd, device=DEV)
+
+    def run(id_first):
+        Xi = X.clone().requires_grad_(True)
+        Wi = W.clone().requires_grad_(True)
+        bi = b.clone().requires_grad_(True)
+        gop = GraphOperator(Go, Gd, id_first=id_first)
+        Y = bdgcn_layer(Xi, Wi, bi, gop, relu=True)
+        Y.square().sum().backward()
+        return Y, Xi.grad, Wi.grad, bi.grad
+
+    Yf, dXf, dWf, dbf = run(False)
+    Ys, dXs, dWs, dbs = run(True)
+    torch.testing.assert_close(Ys, Yf, atol=2e-2, rtol=2e-2)
+    torch.testing.assert_close(dXs.float(), dXf.float(), atol=1e-1, rtol=5e-2)
+    torch.testing.assert_close(dWs.float(), dWf.float(), atol=2.0, rtol=5e-2)
+    torch.testing.assert_close(dbs.float(), dbf.float(), atol=2.0, rtol=5e-2)
+
+
+def test_identity_skip_fp8_matches_full():
+    from mpgcn_amd.graph import build_supports
+    from mpgcn_amd.ops import GraphOperator, bdgcn_layer_fp8
+
+    torch.manual_seed(10)
+    S, C, Hd, Bn, N_ = 3, 32, 32, 2, 256
+    X = torch.randn(Bn, N_, N_, C, device=DEV).bfloat16()
+    Go = build_supports(torch.rand(1, N_, N_, device=DEV),
+                        "random_walk_diffusion", S - 1).squeeze(0)
+    Gd = build_supports(torch.rand(1, N_, N_, device=DEV),
+                        "random_walk_diffusion", S - 1).squeeze(0)
+    Go, Gd = Go.bfloat16().contiguous(), Gd.bfloat16().contiguous()
+    W = (0.1 * torch.randn(C * S * S, Hd, device=DEV)).bfloat16()
+
+    def run(id_first):
+        gop = GraphOperator(Go, Gd, id_first=id_first)
+        Y, _ = bdgcn_layer_fp8(X, W, None, gop, relu=True)
+        return Y
+
+    Yf = run(False)
+    Ys = run(True)
+    rel = (Ys.float() - Yf.float()).norm() / (Yf.float().norm() + 1e-9)
+    assert rel < 0.03, rel.item()
+
+
 @pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
 @pytest.mark.parametrize("dyn", [False, True])
 def test_bdgcn_layer_fwd_bwd_vs_eager(dtype, dyn):
