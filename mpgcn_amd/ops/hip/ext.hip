@@ -878,10 +878,15 @@ std::vector<torch::Tensor> relu_bwd_colsum(torch::Tensor dH, torch::Tensor Y,
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
-    m.def("bdgcn_mode1", &bdgcn_mode1, "origin-axis graph product (K1)");
-    m.def("bdgcn_mode2", &bdgcn_mode2, "dest-axis graph product + bias + act (K2)");
-    m.def("bdgcn_mode2_bwd", &bdgcn_mode2_bwd, "backward dV of mode2");
-    m.def("bdgcn_mode1_bwd", &bdgcn_mode1_bwd, "backward dX of mode1");
+    m.def("bdgcn_mode1", &bdgcn_mode1, "origin-axis graph product (K1)",
+          py::arg("X"), py::arg("GT"), py::arg("id_skip") = false);
+    m.def("bdgcn_mode2", &bdgcn_mode2, "dest-axis graph product + bias + act (K2)",
+          py::arg("V"), py::arg("A2T"), py::arg("bias"), py::arg("relu"),
+          py::arg("N"), py::arg("S"), py::arg("id_skip") = false);
+    m.def("bdgcn_mode2_bwd", &bdgcn_mode2_bwd, "backward dV of mode2",
+          py::arg("dY"), py::arg("A2"), py::arg("S"), py::arg("id_skip") = false);
+    m.def("bdgcn_mode1_bwd", &bdgcn_mode1_bwd, "backward dX of mode1",
+          py::arg("dU"), py::arg("A3T"), py::arg("id_skip") = false);
     m.def("row_gemm", &row_gemm, "fused row GEMM + bias + act (K3/K6)");
     m.def("red_gemm", &red_gemm, "fused reduction GEMM X^T@Y + colsum + xdot");
     m.def("relu_bwd_colsum", &relu_bwd_colsum, "fused ReLU bwd mask + bias colsum");
