@@ -127,6 +127,33 @@ extern "C" void slab_colsum_launch(const float* ws, float* out, long nb, long E,
         ws, out, nb, E);
 }
 
+// Strided identity-slot copy: dst rows at stride dst_stride_b receive the
+// contiguous src rows (row_b bytes each, multiple of 16). Fills the
+// identity-support slots of U/dV in id_skip mode — aten's strided
+// elementwise copy ran these at ~1 TB/s; 16-byte chunks with coalesced
+// reads recover streaming bandwidth.
+__launch_bounds__(256) __global__ void slot_copy_kernel(
+    const char* __restrict__ src, char* __restrict__ dst, long rows,
+    int row_b, long dst_stride_b) {
+    const int cpr = row_b / 16;
+    const long total = rows * (long)cpr;
+    for (long idx = (long)blockIdx.x * 256 + threadIdx.x; idx < total;
+         idx += (long)gridDim.x * 256) {
+        const long r = idx / cpr;
+        const int c = (int)(idx % cpr);
+        *(Chunk16*)(dst + r * dst_stride_b + (long)c * 16) =
+            *(const Chunk16*)(src + r * (long)row_b + (long)c * 16);
+    }
+}
+
+extern "C" void slot_copy_launch(const void* src, void* dst, long rows,
+                                 int row_b, long dst_stride_b, hipStream_t s) {
+    long blocks = (rows * (row_b / 16) + 255) / 256;
+    if (blocks > 8192) blocks = 8192;
+    slot_copy_kernel<<<dim3((unsigned)blocks), dim3(256), 0, s>>>(
+        (const char*)src, (char*)dst, rows, row_b, dst_stride_b);
+}
+
 // Delayed-scaling bookkeeping for the fp8 gradient path: derive this step's
 // quantize scale (and its exact descale pair) from LAST step's recorded
 // amax, then reset the amax accumulator — one thread, device-side only, so

@@ -147,4 +147,6 @@ long relu_bwd_nblocks(long total);
 long red_gemm_nblocks(long R);
 void fp8_scale_update_launch(float* amax, float* scale, float* inv,
                              float margin, hipStream_t s);
+void slot_copy_launch(const void* src, void* dst, long rows, int row_b,
+                      long dst_stride_b, hipStream_t s);
 }
