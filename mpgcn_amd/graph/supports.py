@@ -124,6 +124,17 @@ def build_supports(
         raise ValueError(f"flow must be (B, N, N); got {tuple(flow.shape)}")
     K = get_support_K(kernel_type, cheby_order)
 
+    if (kernel_type == "random_walk_diffusion" and flow.is_cuda
+            and cheby_order >= 1):
+        # fused HIP build (K8): 2 + (order-1) launches instead of the ~10
+        # stock-op chain below; the torch path remains the numerics oracle
+        # (tests/test_supports.py) and the CPU / other-kernel-type fallback
+        from mpgcn_amd import ops as _ops
+
+        out = _ops.get_ext().rwd_supports(flow.float().contiguous(), cheby_order)
+        out._identity_first = True
+        return out
+
     if kernel_type == "localpool":
         sup = _batched_eye(
             flow.shape[0], flow.shape[1], dtype=flow.dtype, device=flow.device

@@ -154,6 +154,69 @@ extern "C" void slot_copy_launch(const void* src, void* dst, long rows,
         (const char*)src, (char*)dst, rows, row_b, dst_stride_b);
 }
 
+// ---- fused dynamic-support build (random_walk_diffusion), K8 ----
+// Replaces the per-step stock-op chain (rowsum / where / reciprocal / mul /
+// transpose / eye-expand / stack — ~10 launches and as many tensor passes
+// per graph, reference hotspot GCN.py:56-100 + Model_Trainer.py:106) with:
+//   rwd_rowsum_kernel:  d[b,n] = sum_k A[b,n,k]
+//   rwd_norm_t_kernel:  OUT[b,0] = I,  OUT[b,1,m,n] = A[b,n,m] / d[b,n]
+//   + one axis_gemm per Chebyshev order >= 2 (T_k = 2*PT@T_{k-1} - T_{k-2}
+//     via the alpha/CSUB epilogue) — 2 + (order-1) launches per graph.
+
+__launch_bounds__(256) __global__ void rwd_rowsum_kernel(
+    const float* __restrict__ A, float* __restrict__ d, long rows, int N) {
+    const long r = (long)blockIdx.x * 4 + threadIdx.x / 64;
+    const int lane = threadIdx.x % 64;
+    if (r >= rows) return;
+    float s = 0.f;
+    for (int k = lane; k < N; k += 64) s += A[r * (long)N + k];
+#pragma unroll
+    for (int off = 32; off >= 1; off >>= 1) s += __shfl_xor(s, off);
+    if (lane == 0) d[r] = s;
+}
+
+extern "C" void rwd_rowsum_launch(const float* A, float* d, long rows, int N,
+                                  hipStream_t s) {
+    rwd_rowsum_kernel<<<dim3((unsigned)((rows + 3) / 4)), dim3(256), 0, s>>>(
+        A, d, rows, N);
+}
+
+__launch_bounds__(256) __global__ void rwd_norm_t_kernel(
+    const float* __restrict__ A, const float* __restrict__ d,
+    float* __restrict__ OUT, int N, long sOUT) {
+    __shared__ float tile[64][65];
+    const long b = blockIdx.z;
+    const int n0 = blockIdx.x * 64, m0 = blockIdx.y * 64;
+    const int tid = threadIdx.x;
+    // stage A[b, n0+i, m0+j] (coalesced rows)
+    for (int idx = tid; idx < 64 * 64; idx += 256) {
+        const int i = idx / 64, j = idx % 64;
+        const int n = n0 + i, m = m0 + j;
+        tile[i][j] = (n < N && m < N) ? A[b * (long)N * N + (long)n * N + m] : 0.f;
+    }
+    __syncthreads();
+    // write PT[b, m0+a, n0+c] = tile[c][a] / d[n0+c], and the I slot
+    float* O0 = OUT + b * sOUT;              // support 0: identity
+    float* O1 = O0 + (long)N * N;            // support 1: P^T
+    for (int idx = tid; idx < 64 * 64; idx += 256) {
+        const int a = idx / 64, c = idx % 64;
+        const int m = m0 + a, n = n0 + c;
+        if (m < N && n < N) {
+            const float dv = d[b * (long)N + n];
+            const float pt = dv != 0.f ? tile[c][a] / dv : 0.f;
+            O1[(long)m * N + n] = pt;
+            O0[(long)m * N + n] = (m == n) ? 1.f : 0.f;
+        }
+    }
+}
+
+extern "C" void rwd_norm_t_launch(const float* A, const float* d, float* OUT,
+                                  long B, int N, long sOUT, hipStream_t s) {
+    const unsigned t = (N + 63) / 64;
+    rwd_norm_t_kernel<<<dim3(t, t, (unsigned)B), dim3(256), 0, s>>>(
+        A, d, OUT, N, sOUT);
+}
+
 // Delayed-scaling bookkeeping for the fp8 gradient path: derive this step's
 // quantize scale (and its exact descale pair) from LAST step's recorded
 // amax, then reset the amax accumulator — one thread, device-side only, so

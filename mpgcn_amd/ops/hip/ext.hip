@@ -862,6 +862,51 @@ void fp8_scale_update(torch::Tensor amax, torch::Tensor scale,
                             inv.data_ptr<float>(), (float)margin, stream());
 }
 
+// Fused dynamic-support build (K8, random_walk_diffusion): raw flow
+// (B, N, N) f32 -> (B, order+1, N, N) f32 support stack in 2 + (order-1)
+// launches (rowsum, fused normalize+transpose+identity, then one
+// alpha/CSUB-epilogue GEMM per Chebyshev order). Replaces the per-step
+// ~10-launch stock chain (graph/supports.py torch path, kept as the
+// CPU/other-kernel-type fallback and the numerics oracle).
+torch::Tensor rwd_supports(torch::Tensor flow, long order) {
+    TORCH_CHECK(flow.is_cuda() && flow.is_contiguous() &&
+                flow.scalar_type() == torch::kFloat, "flow must be CUDA f32");
+    TORCH_CHECK(flow.dim() == 3 && flow.size(1) == flow.size(2), "flow (B,N,N)");
+    TORCH_CHECK(order >= 1, "rwd_supports needs order >= 1");
+    const long B = flow.size(0), N = flow.size(1), S = order + 1;
+    TORCH_CHECK(B <= 65535, "too many instances");
+    auto out = torch::empty({B, S, N, N}, flow.options());
+    auto d = torch::empty({B, N}, flow.options());
+    rwd_rowsum_launch(flow.data_ptr<float>(), d.data_ptr<float>(), B * N,
+                      (int)N, stream());
+    rwd_norm_t_launch(flow.data_ptr<float>(), d.data_ptr<float>(),
+                      out.data_ptr<float>(), B, (int)N, S * N * N, stream());
+    float* base = out.data_ptr<float>();
+    for (long k = 2; k <= order; ++k) {
+        // T_k = 2 * PT @ T_{k-1} - T_{k-2}
+        AxisGemmParams p{};
+        p.AT = base + 1 * N * N;        // PT (slot 1)
+        p.X = base + (k - 1) * N * N;   // T_{k-1}
+        p.OUT = base + k * N * N;       // T_k
+        p.CSUB = base + (k - 2) * N * N;
+        p.alpha = 2.f; p.cs_beta = -1.f;
+        p.M = (int)N; p.K = (int)N; p.L = (int)N;
+        p.a_div = 1; p.a_bs1 = S * N * N; p.a_bs2 = 0;
+        p.x_div = 1; p.x_bs1 = S * N * N; p.x_bs2 = 0;
+        p.o_div = 1; p.o_bs1 = S * N * N; p.o_bs2 = 0;
+        p.cs_div = 1; p.cs_bs1 = S * N * N; p.cs_bs2 = 0;
+        p.cs_row = N;
+        p.kdiv = 1; p.k_lo = N;
+        p.qdiv = 0;
+        p.o_row = N;
+        p.ogdiv = 0;
+        p.a_vec = (N % 4 == 0);
+        p.x_vec = (N % 4 == 0);
+        axis_gemm_launch(p, (int)B, /*is_f32=*/1, stream());
+    }
+    return out;
+}
+
 // Fused ReLU backward + bias-grad column sum: dY = dH * 1[Y>0], dbias=colsum.
 std::vector<torch::Tensor> relu_bwd_colsum(torch::Tensor dH, torch::Tensor Y,
                                            bool mask) {
@@ -914,6 +959,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("bdgcn_mode2_fp8_train", &bdgcn_mode2_fp8_train,
           "fp8-forward mode-2 + bias + act, bf16 out + fp8 twin");
     m.def("row_gemm_fp8", &row_gemm_fp8, "fp8 projection GEMM");
+    m.def("rwd_supports", &rwd_supports, "fused random-walk-diffusion support build (K8)");
     m.def("bdgcn_mode2_bwd_fp8", &bdgcn_mode2_bwd_fp8,
           "scaled fp8 gradient contraction dV");
     m.def("bdgcn_mode1_bwd_fp8", &bdgcn_mode1_bwd_fp8,

@@ -236,6 +236,22 @@ def test_reg_lstm_bf16_grads_match_autograd(T):
         assert err < 3e-2, f"{name}: rel grad err {err:.4f}"
 
 
+@pytest.mark.parametrize("order", [1, 2, 4])
+@pytest.mark.parametrize("N_", [47, 256])
+def test_fused_rwd_supports_matches_torch_path(order, N_):
+    """The fused K8 support build (rowsum + normalize/transpose/identity +
+    alpha/CSUB Chebyshev GEMMs) vs the stock torch chain on CPU."""
+    from mpgcn_amd.graph import build_supports
+
+    torch.manual_seed(order * 31 + N_)
+    flow = torch.rand(5, N_, N_) * 4
+    flow[0, 3] = 0  # empty row: rcp guard
+    ref = build_supports(flow, "random_walk_diffusion", order)  # CPU torch
+    got = build_supports(flow.to(DEV), "random_walk_diffusion", order)  # fused
+    assert getattr(got, "_identity_first", False)
+    torch.testing.assert_close(got.cpu(), ref, atol=1e-4, rtol=1e-4)
+
+
 @pytest.mark.parametrize("dyn", [False, True])
 def test_identity_skip_matches_full_path(dyn):
     """id_first mode (support 0 = I skipped: reduced operand layouts,
