@@ -140,6 +140,46 @@ class GraphOperator:
         return self._A3T
 
 
+_dw_streams: dict = {}
+
+
+def _dw_stream(device):
+    """Side stream for the weight-gradient reduction: red_gemm (dWre) is
+    independent of the dU -> dX chain, so it overlaps the rest of the layer
+    backward instead of serializing it (the b=32 flagship regime is
+    latency-bound on the per-layer chain, not device throughput)."""
+    s = _dw_streams.get(device)
+    if s is None:
+        s = _dw_streams[device] = torch.cuda.Stream(device)
+    return s
+
+
+def _dw_overlapped(ext, dVflat, U2d, dtype, S, C, Hdim):
+    """dW via red_gemm on the side stream; returns (dW, join_fn). Caller runs
+    the dX chain, then calls join_fn() before returning dW to autograd."""
+    cur = torch.cuda.current_stream()
+    s = _dw_stream(dVflat.device)
+    s.wait_stream(cur)
+    with torch.cuda.stream(s):
+        dWreT, _, _ = ext.red_gemm(dVflat, U2d, False, None, 0, 0)
+        dWre = dWreT.t().to(dtype)
+        dW = dWre.reshape(S, C, S, Hdim).permute(0, 2, 1, 3).reshape(S * S * C, Hdim)
+        dW = dW.contiguous()
+    # under hipGraph capture record_stream is not permitted; the wait edges
+    # alone express the dependency there (graph pool owns lifetimes)
+    capturing = torch.cuda.is_current_stream_capturing()
+    if not capturing:
+        dVflat.record_stream(s)
+        U2d.record_stream(s)
+
+    def join():
+        cur.wait_stream(s)
+        if not capturing:
+            dW.record_stream(cur)
+
+    return dW, join
+
+
 def _row_gemm_chunked(ext, X2d, W, bias, relu):
     """row_gemm with column chunking for N > 128 (e.g. dual-RWD S=5: S*H=160)."""
     R, _ = X2d.shape
@@ -199,12 +239,13 @@ class _BDGCNLayerFn(torch.autograd.Function):
         dV = ext.bdgcn_mode2_bwd(dY, gop.A2, S, gop.id_first)  # (B,N,N,S,H)
         R = B * N * N
         dVflat = dV.reshape(R, S * Hdim)
-        # dWre^T = dV^T @ U via the fused reduction kernel (f32 accumulate)
-        dWreT, _, _ = ext.red_gemm(dVflat, U.reshape(R, S * C), False, None, 0, 0)
-        dWre = dWreT.t().to(dH.dtype)
-        dW = dWre.reshape(S, C, S, Hdim).permute(0, 2, 1, 3).reshape(S * S * C, Hdim)
+        # dWre^T = dV^T @ U via the fused reduction kernel (f32 accumulate),
+        # overlapped with the dU -> dX chain on a side stream
+        dW, join_dw = _dw_overlapped(ext, dVflat, U.reshape(R, S * C),
+                                     dH.dtype, S, C, Hdim)
         dU = _row_gemm_chunked(ext, dVflat, Wre.t().contiguous(), None, False)
         dX = ext.bdgcn_mode1_bwd(dU.view(B, N, N, S, C), gop.A3T, gop.id_first)
+        join_dw()
         db = dbias if ctx.has_bias else None
         return dX, dW, db, None, None
 
@@ -318,15 +359,15 @@ class _BDGCNLayerFp8Fn(torch.autograd.Function):
         dV = ext.bdgcn_mode2_bwd_fp8(dY8, gop.A28, S, st["inv_y"], dY, gop.id_first)
         R = B * N * N
         dVflat = dV.reshape(R, S * Hdim)
-        dWreT, _, _ = ext.red_gemm(dVflat, U8.reshape(R, S * C), False, None, 0, 0)
-        dWre = dWreT.t().to(dH.dtype)
-        dW = dWre.reshape(S, C, S, Hdim).permute(0, 2, 1, 3).reshape(S * S * C, Hdim)
+        dW, join_dw = _dw_overlapped(ext, dVflat, U8.reshape(R, S * C),
+                                     dH.dtype, S, C, Hdim)
 
         ext.fp8_scale_update(st["amax_u"], st["scale_u"], st["inv_u"], _FP8_MARGIN)
         dU8 = ext.row_gemm_fp8_out(dVflat, Wre.t().contiguous(),
                                    st["scale_u"], st["amax_u"])
         dX = ext.bdgcn_mode1_bwd_fp8(dU8.view(B, N, N, S, C), gop.A3T8,
                                      st["inv_u"], gop.id_first)
+        join_dw()
         return dX, dW, dbias, None, None, None, None, None
 
 
