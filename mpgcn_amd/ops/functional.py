@@ -337,6 +337,13 @@ class _BDGCNLayerFp8Fn(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dH, _dY8=None):
+        import os
+
+        if os.environ.get("MPGCN_FP8_BWD", "1") == "0":
+            # debug/bisect: fp8 forward with the full bf16 backward (red_gemm
+            # reads the saved U8 directly either way)
+            dX, dW, db, _, _ = _BDGCNLayerFn.backward(ctx, dH)
+            return dX, dW, db, None, None, None, None, None
         # Scaled-fp8 gradient contractions: gradients live well below e4m3's
         # 2^-9 subnormal floor, so every quantize carries a DEVICE-resident
         # dynamic scale and the axis-kernel epilogue descales — no host sync.

@@ -67,7 +67,12 @@ template <>
 __device__ __forceinline__ float from_f32<float>(float v) { return v; }
 template <>
 __device__ __forceinline__ unsigned char from_f32<unsigned char>(float v) {
-    // v_cvt_pk_fp8_f32 (OCP e4m3 on gfx950); low byte of the packed pair
+    // v_cvt_pk_fp8_f32 (OCP e4m3 on gfx950) does NOT saturate: inputs past
+    // +-448 convert to NaN, which poisons fp8 gradient tensors whenever a
+    // delayed scale lags a >2x amax jump (and fmaxf-based amax recording
+    // silently drops NaN, locking the scale — a measured training doom
+    // loop). Clamp to the e4m3 range first; fminf/fmaxf also squash NaN.
+    v = fminf(fmaxf(v, -448.f), 448.f);
     return (unsigned char)(__builtin_amdgcn_cvt_pk_fp8_f32(v, 0.f, 0, false) & 0xff);
 }
 
