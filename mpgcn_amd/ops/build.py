@@ -40,6 +40,15 @@ def build(verbose: bool = True) -> Path:
     from torch.utils import cpp_extension
 
     BUILD_DIR.mkdir(parents=True, exist_ok=True)
+    # ninja's depfiles miss header deps through torch's hipify step (a .hpp
+    # edit silently left stale .o files linked into a "fresh" .so — measured
+    # the hard way); clear the objects whenever any header is newer than one
+    hpp_m = max((h.stat().st_mtime for h in HIP_DIR.glob("*.hpp")), default=0)
+    for o in BUILD_DIR.glob("*.o"):
+        if o.stat().st_mtime < hpp_m:
+            for f in list(BUILD_DIR.glob("*.o")) + list(HIP_DIR.glob("*_hip.hip")):
+                f.unlink(missing_ok=True)
+            break
     cpp_extension.load(
         name=EXT_NAME,
         sources=[str(HIP_DIR / s) for s in SOURCES],
