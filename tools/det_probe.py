@@ -2,6 +2,8 @@
 and report any drift; then two full model steps. Run on a GPU box."""
 import torch
 
+import os, sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from mpgcn_amd import ops
 from mpgcn_amd.graph import build_supports
 from mpgcn_amd.models import MPGCN
