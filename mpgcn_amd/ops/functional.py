@@ -163,6 +163,13 @@ def _dw_stream(device):
 def _dw_overlapped(ext, dVflat, U2d, dtype, S, C, Hdim):
     """dW via red_gemm on the side stream; returns (dW, join_fn). Caller runs
     the dX chain, then calls join_fn() before returning dW to autograd."""
+    if torch.cuda.is_current_stream_capturing():
+        # hipGraph capture: run inline (forking a non-capturing stream from
+        # inside a capture aborts the process)
+        dWreT, _, _ = ext.red_gemm(dVflat, U2d, False, None, 0, 0)
+        dWre = dWreT.t().to(dtype)
+        dW = dWre.reshape(S, C, S, Hdim).permute(0, 2, 1, 3).reshape(S * S * C, Hdim)
+        return dW.contiguous(), (lambda: None)
     cur = torch.cuda.current_stream()
     s = _dw_stream(dVflat.device)
     s.wait_stream(cur)
