@@ -173,12 +173,6 @@ __launch_bounds__(256) __global__ void lstm_fused_bwd_kernel(LstmFusedParams p) 
     // c states live in LDS (they would otherwise cost 8*T VGPRs per lane):
     // CST[w][t][row][j], row stride 33 breaks write conflicts
     __shared__ float ldsC[4][T * 16 * 33];
-    // tanh(c_t) cache (bf16): phase A already computes it for h — caching it
-    // drops T*8 exp+rcp pairs per lane from phase B's recompute (the kernel
-    // is execution/latency-bound; transcendentals are 4-cycle issues).
-    // T = 8 would exceed the 160 KB LDS: gated to T <= 7 (flagship T = 7).
-    constexpr bool TCC_ON = (T <= 7);
-    __shared__ __bf16 ldsTC[TCC_ON ? 4 : 1][(TCC_ON ? T : 1) * 16 * 33];
 
     const __bf16* __restrict__ X = (const __bf16*)p.x;
     const __bf16* __restrict__ Whh = (const __bf16*)p.whh;
@@ -203,7 +197,6 @@ __launch_bounds__(256) __global__ void lstm_fused_bwd_kernel(LstmFusedParams p) 
     __bf16* myHT = &ldsHT[w][0];
     __bf16* myH = &ldsHT[w][0];  // phase-A transpose scratch aliases the hT image
     float* myC = &ldsC[w][0];
-    __bf16* myTC = TCC_ON ? &ldsTC[w][0] : nullptr;
     float wih_r[8], bias_r[8];
 #pragma unroll
     for (int nf = 0; nf < 8; ++nf) {
@@ -255,7 +248,6 @@ __launch_bounds__(256) __global__ void lstm_fused_bwd_kernel(LstmFusedParams p) 
         // ---- phase A: forward recompute; h states in registers, c in LDS ----
         lf_frag h_states[T];  // h AFTER step t (A-frag layout)
 #define CST(t, r, jf) myC[(t) * 16 * 33 + (kgrp * 4 + (r)) * 33 + (jf) * 16 + lrow]
-#define TCC(t, r, jf) myTC[(t) * 16 * 33 + (kgrp * 4 + (r)) * 33 + (jf) * 16 + lrow]
         {
             lf_frag h_frag = h_in_frag;
             float c[2][4];
@@ -280,10 +272,8 @@ __launch_bounds__(256) __global__ void lstm_fused_bwd_kernel(LstmFusedParams p) 
                         const float c_new = gv[2 + jf] * c[jf][r] + gv[0 + jf] * gv[4 + jf];
                         c[jf][r] = c_new;
                         CST(t, r, jf) = c_new;
-                        const float th = fast_tanh(c_new);
-                        if (TCC_ON) TCC(t, r, jf) = (__bf16)th;
                         myH[(kgrp * 4 + r) * (LF_H + 8) + jf * 16 + lrow] =
-                            (__bf16)(gv[6 + jf] * th);
+                            (__bf16)(gv[6 + jf] * fast_tanh(c_new));
                     }
                 }
                 lds_wave_fence();
@@ -330,9 +320,9 @@ __launch_bounds__(256) __global__ void lstm_fused_bwd_kernel(LstmFusedParams p) 
                 for (int jf = 0; jf < 2; ++jf) {
                     const float i_g = gv[0 + jf], f_g = gv[2 + jf];
                     const float g_g = gv[4 + jf], o_g = gv[6 + jf];
+                    const float c_t = CST(t, r, jf);
                     const float c_prev = (t > 0) ? CST(t - 1, r, jf) : c0r[jf][r];
-                    const float tc = TCC_ON ? to_f32(TCC(t, r, jf))
-                                            : fast_tanh(CST(t, r, jf));
+                    const float tc = fast_tanh(c_t);
                     float d_c = dc[jf][r] + dh[jf][r] * o_g * (1.f - tc * tc);
                     const float d_i = d_c * g_g, d_g = d_c * i_g, d_f = d_c * c_prev;
                     const float d_o = dh[jf][r] * tc;
