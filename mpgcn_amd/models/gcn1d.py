@@ -33,10 +33,32 @@ class GCN(nn.Module):
 
     def forward(self, G: torch.Tensor, x: torch.Tensor) -> torch.Tensor:
         assert self.K == G.shape[0]
-        # support products: (K, N, N) x (B, N, C) -> (B, N, K*C)
+        B, N, C = x.shape
+        if (x.is_cuda and x.dtype in (torch.bfloat16, torch.float32)
+                and self.hidden_dim <= 128 and self.K * C <= 2048):
+            # HIP path (K5): the K-support contraction is a mode-1 axis GEMM
+            # with a singleton destination axis, and the projection + bias +
+            # act is one fused row_gemm — the same MFMA kernels as BDGCN
+            # (ext.hip bdgcn_mode1 / row_gemm; eager einsum is the oracle).
+            from mpgcn_amd.ops.functional import _ops
+
+            ext = _ops.get_ext()
+            # the kernel's A operand is consumed row-major as A[m, n] with n
+            # contracted; the 1-D GCN wants sum_j G[i, j] x[j], so G itself
+            # (NOT transposed — BDGCN's mode-1 passes G^T because its einsum
+            # contracts the FIRST graph index, GCN.py:34 vs MPGCN.py:30)
+            A = G.to(x.dtype).contiguous()
+            U = ext.bdgcn_mode1(x.reshape(B, N, 1, C).contiguous(), A, False)
+            # U: (B, N, 1, K, C) — concat over supports is the natural layout
+            bias_f = self.b.float().contiguous() if self.b is not None else None
+            out = ext.row_gemm(
+                U.reshape(B * N, self.K * C),
+                self.W.to(x.dtype).contiguous(), bias_f, self.relu,
+            )
+            return out.view(B, N, self.hidden_dim)
+        # eager fallback (CPU / exotic shapes): reference math (GCN.py:34-44)
         sup = torch.einsum("kij,bjp->bkip", G, x)
-        B, K, N, C = sup.shape
-        cat = sup.permute(0, 2, 1, 3).reshape(B, N, K * C)
+        cat = sup.permute(0, 2, 1, 3).reshape(B, N, self.K * C)
         out = cat @ self.W
         if self.b is not None:
             out = out + self.b
