@@ -35,7 +35,8 @@ class Forecaster:
         self.params = params
         self.device = torch.device(params.get("device", "cpu"))
         cd = str(params.get("compute_dtype", "float32"))
-        compute_dtype = (torch.bfloat16 if cd in ("bf16", "bfloat16")
+        fp8 = cd == "fp8"  # fp8-forward inference (weights stay fp32 masters)
+        compute_dtype = (torch.bfloat16 if cd in ("bf16", "bfloat16", "fp8")
                         else torch.float32)
         K = get_support_K(params["kernel_type"], params["cheby_order"])
         N = data["adj"].shape[-1]
@@ -46,6 +47,7 @@ class Forecaster:
             gcn_hidden_dim=params["hidden_dim"], gcn_num_layers=3,
             num_nodes=N, compute_dtype=compute_dtype,
             fusion=params.get("fusion", "mean"),
+            fp8_forward=fp8,
         ).to(self.device)
         ckpt = torch.load(params["checkpoint"], map_location=self.device,
                           weights_only=False)
