@@ -229,6 +229,10 @@ extern "C" void axis_gemm_fp8_launch(AxisGemmParams p, int instances,
 extern "C" void axis_gemm_launch(AxisGemmParams p, int instances, int is_f32,
                                  hipStream_t stream) {
     constexpr int BK = 64;
+    // (A/B round 2: a 128x128 4-wave tile for short-K contractions measured
+    // -6% at b32 and far worse at b128 — the byte-volume halving of the 256
+    // tile beats stage-latency overlap even at 2 K-stages; null kept in
+    // profiles/SUMMARY.md)
     int BN = (p.L >= 96 && p.M >= 96) ? 128 : (p.L >= 48 ? 64 : 32);
     if (!is_f32 && p.M >= 192 && p.L >= 192) BN = 256;
     if (is_f32 && BN > 64) BN = 64;  // f32 LDS budget caps the tile

@@ -120,6 +120,62 @@ def test_fp8_model_train_step_loss_decreases():
     assert losses[-1] < losses[0], losses
 
 
+def test_fp8_delayed_scale_adapts_from_mse_scale_gradients():
+    """Regression for the training doom loop: MSE-mean gradients (~1e-6)
+    underflow e4m3 at the bootstrap scale; the delayed per-layer scale must
+    adapt by the second call and produce gradients matching bf16."""
+    from mpgcn_amd.ops import bdgcn_layer
+    from mpgcn_amd.ops.functional import make_fp8_state
+
+    torch.manual_seed(0)
+    X, Go, Gd, W, b = _layer_inputs()
+    gop = GraphOperator(Go, Gd)
+    Xb = X.clone().requires_grad_(True)
+    Yb = bdgcn_layer(Xb, W.clone(), b.clone(), gop, relu=True)
+    (Yb.float() ** 2).mean().backward()
+    ref = Xb.grad.float()
+
+    st = make_fp8_state(X.device)
+    # warmup: call 0 underflows dY8 to zero at the bootstrap scale; the
+    # recorded amax adapts dY's scale for call 1, whose dU amax then fixes
+    # dX by call 2 (the id_skip path adapts one call earlier because its
+    # EXACT identity-gradient rows keep signal flowing through underflow —
+    # measured in tools/fp8_diag.py)
+    for it in range(3):
+        Xi = X.clone().requires_grad_(True)
+        Y, _ = bdgcn_layer_fp8(Xi, W.clone(), b.clone(), gop, relu=True,
+                               fp8_state=st)
+        (Y.float() ** 2).mean().backward()
+    got = Xi.grad.float()
+    assert got.norm() > 0
+    rel = (got - ref).norm() / (ref.norm() + 1e-12)
+    assert rel < 0.1, rel.item()
+
+
+def test_fp8_overflow_clips_instead_of_nan():
+    """gfx950's fp8 convert does NOT saturate (overflow -> NaN); the kernels
+    must clamp. A stale tiny amax (huge scale) against large gradients must
+    degrade to clipping, never NaN — the exact sequence that locked training
+    into a NaN/zero doom loop before the fix (profiles/FP8.md)."""
+    from mpgcn_amd.ops.functional import make_fp8_state
+
+    torch.manual_seed(1)
+    X, Go, Gd, W, b = _layer_inputs()
+    gop = GraphOperator(Go, Gd)
+    st = make_fp8_state(X.device)
+    # poison the delayed state exactly like the doom loop: recorded amax ~ 0
+    st["amax_y"].fill_(1e-20)
+    st["amax_u"].fill_(1e-20)
+    Xi = X.clone().requires_grad_(True)
+    Wi = W.clone().requires_grad_(True)
+    Y, _ = bdgcn_layer_fp8(Xi, Wi, b.clone(), gop, relu=True, fp8_state=st)
+    Y.square().sum().backward()  # large dH against scale ~ 2e22
+    assert torch.isfinite(Xi.grad.float()).all()
+    assert torch.isfinite(Wi.grad.float()).all()
+    # and the state must have recorded the TRUE amax so the next step recovers
+    assert st["amax_y"].item() > 1.0
+
+
 def test_fp8_shape_gate_raises():
     with pytest.raises(ValueError, match="fp8_forward shape gate"):
         MPGCN(M=2, K=S, input_dim=1, lstm_hidden_dim=24, lstm_num_layers=1,

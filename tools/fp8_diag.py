@@ -18,8 +18,12 @@ torch.manual_seed(0)
 X = torch.rand(B, N, N, C, device=dev).bfloat16()
 Go = build_supports(torch.rand(1, N, N, device=dev),
                     "random_walk_diffusion", 2).squeeze(0).bfloat16().contiguous()
-Go._identity_first = True
-gop = GraphOperator(Go, Go)
+import sys as _s
+if "--full" in _s.argv:
+    gop = GraphOperator(Go, Go, id_first=False)
+else:
+    Go._identity_first = True
+    gop = GraphOperator(Go, Go)
 W = (0.1 * torch.randn(C * S * S, H, device=dev)).bfloat16()
 st = make_fp8_state(dev)
 
