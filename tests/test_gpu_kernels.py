@@ -500,11 +500,14 @@ def test_deterministic_mode_bitwise_reproducible():
         torch.use_deterministic_algorithms(False)
 
 
-def test_rectangular_sharded_halves_match_eager():
+@pytest.mark.parametrize("id_first", [False, True])
+def test_rectangular_sharded_halves_match_eager(id_first):
     """Region-partition path on GPU: the layer split at the all-to-all seam
     (mode1_proj | mode2_bias_act) on rectangular shards — the HIP bindings'
     rectangular origin/destination extents — matches the fp32 eager layer,
-    forward and backward, with the re-shard emulated by slice/cat."""
+    forward and backward, with the re-shard emulated by slice/cat.
+    id_first=True exercises the identity-support skip on RECTANGULAR
+    extents (the path the region trainer takes with build_supports graphs)."""
     from mpgcn_amd.ops import GraphOperator, eager, mode1_proj, mode2_bias_act
 
     torch.manual_seed(11)
@@ -514,6 +517,10 @@ def test_rectangular_sharded_halves_match_eager():
     X32 = torch.randn(B, Nn, Nn, C, device=DEV)
     Go = torch.randn(S, Nn, Nn, device=DEV) / Nn**0.5
     Gd = torch.randn(S, Nn, Nn, device=DEV) / Nn**0.5
+    if id_first:  # the skip's contract: support 0 IS the identity
+        eye = torch.eye(Nn, device=DEV)
+        Go[0] = eye
+        Gd[0] = eye
     W32 = (torch.randn(C * S * S, Hd, device=DEV) / (C * S * S) ** 0.5).requires_grad_()
     bias32 = torch.randn(Hd, device=DEV, requires_grad=True)
 
@@ -524,7 +531,7 @@ def test_rectangular_sharded_halves_match_eager():
     X = X32.to(dt).requires_grad_()
     W = W32.detach().to(dt).requires_grad_()
     bias = bias32.detach().clone().requires_grad_()
-    gop = GraphOperator(Go.to(dt), Gd.to(dt))
+    gop = GraphOperator(Go.to(dt), Gd.to(dt), id_first=id_first)
     Vs = [mode1_proj(X[:, :, p * Nl:(p + 1) * Nl, :].contiguous(), W, gop)
           for p in range(P)]
     Vfull = torch.cat(Vs, dim=2)  # emulated all-to-all
