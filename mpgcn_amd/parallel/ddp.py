@@ -56,6 +56,8 @@ def init_distributed(device: str = "cuda", backend: str | None = None,
     if world <= 1:
         return DistContext()
     if backend is None:
+        backend = os.environ.get("MPGCN_DIST_BACKEND")  # test/bring-up override
+    if backend is None:
         backend = "nccl" if (device.startswith("cuda") and torch.cuda.is_available()) else "gloo"
     if not dist.is_initialized():
         if backend == "nccl":
