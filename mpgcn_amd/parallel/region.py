@@ -60,6 +60,21 @@ class _AllToAllShard(torch.autograd.Function):
         return _a2a(g.contiguous(), inv, ctx.group), None, None
 
 
+def _a2a_exchange(send: torch.Tensor, group) -> torch.Tensor:
+    """all_to_all_single with a CPU-staged fallback for gloo+CUDA (gloo's
+    CUDA transport covers all_reduce/broadcast but not all-to-all) — lets
+    the region path run multi-rank on shared/heterogeneous setups; RCCL
+    takes the direct device path."""
+    if send.is_cuda and dist.get_backend(group) == "gloo":
+        send_c = send.cpu()
+        recv_c = torch.empty_like(send_c)
+        dist.all_to_all_single(recv_c, send_c, group=group)
+        return recv_c.to(send.device, non_blocking=True)
+    recv = torch.empty_like(send)
+    dist.all_to_all_single(recv, send, group=group)
+    return recv
+
+
 def _a2a(x: torch.Tensor, direction: str, group) -> torch.Tensor:
     P = dist.get_world_size(group)
     if direction == "d2o":
@@ -67,8 +82,7 @@ def _a2a(x: torch.Tensor, direction: str, group) -> torch.Tensor:
         assert N == Nl * P, (N, Nl, P)
         # send chunk q = origin rows [q*Nl, (q+1)*Nl) of the local dest slice
         send = x.reshape(B, P, Nl, Nl, F).permute(1, 0, 2, 3, 4).contiguous()
-        recv = torch.empty_like(send)
-        dist.all_to_all_single(recv, send, group=group)
+        recv = _a2a_exchange(send, group)
         # recv[p] = origin rows (local) x dest cols of peer p
         out = recv.permute(1, 2, 0, 3, 4).reshape(B, Nl, N, F)
         return out.contiguous()
@@ -76,8 +90,7 @@ def _a2a(x: torch.Tensor, direction: str, group) -> torch.Tensor:
         B, Nl, N, F = x.shape
         assert N == Nl * P, (N, Nl, P)
         send = x.reshape(B, Nl, P, Nl, F).permute(2, 0, 1, 3, 4).contiguous()
-        recv = torch.empty_like(send)
-        dist.all_to_all_single(recv, send, group=group)
+        recv = _a2a_exchange(send, group)
         out = recv.permute(1, 0, 2, 3, 4).reshape(B, N, Nl, F)
         return out.contiguous()
 
