@@ -153,6 +153,13 @@ def mpgcn_forward_sharded(model, x_seq_shard, G_list, group=None):
         raise ValueError(
             f"region partition needs the region count ({N}) divisible by the "
             f"world size ({P}); pad the grid or change the rank count")
+    if getattr(model, "fp8_forward", False):
+        import warnings
+
+        warnings.warn(
+            "fp8_forward is not implemented for the region-sharded path; "
+            "running the shards in bf16 (same fp32-master weights)",
+            stacklevel=2)
     gops = model._graph_operators(G_list)
     cd = model.compute_dtype
     lstm_in = (
