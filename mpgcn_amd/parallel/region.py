@@ -112,6 +112,104 @@ def shard_dest(x_full: torch.Tensor, rank: int, P: int) -> torch.Tensor:
     return x_full[..., rank * Nl:(rank + 1) * Nl, :].contiguous()
 
 
+class _ShardedFp8Layer(torch.autograd.Function):
+    """One sharded BDGCN layer in fp8-forward mode, fused across the
+    exchange seams: mode1+proj (destination-sharded, fp8 kernels) ->
+    all-to-all of the fp8 V8 -> mode2+bias+act (origin-sharded) ->
+    all-to-all of the fp8 Y8 twin -> dequant. BOTH forward exchanges carry
+    1-BYTE payloads — the all-to-all is this workload's bandwidth-critical
+    collective, so fp8 halves it. The transported Y8 bytes
+    double as the next layer's X8 operand (zero extra quantization), exactly
+    like the square path's twin chaining. Backward transports bf16 gradients
+    through the inverse all-to-alls and reuses the square path's scaled-fp8
+    gradient machinery (delayed per-layer scales in `st`).
+    The LAST layer instead transports bf16 Y (its consumer is the bf16 FC
+    head — parity with the square path's emit_twin=False)."""
+
+    @staticmethod
+    def forward(ctx, X, W, bias, gop: GraphOperator, relu: bool, X8,
+                group, st: dict, last: bool):
+        from mpgcn_amd.ops import eager
+        from mpgcn_amd.ops.functional import _ops
+
+        ext = _ops.get_ext()
+        B, N, Nl, C = X.shape
+        S = gop.S
+        Hd = W.shape[1]
+        if X8 is None:
+            X8 = X.to(torch.float8_e4m3fn)
+        U8 = ext.bdgcn_mode1_fp8_train(X8, gop.GoT8, gop.id_first)
+        Wre = eager.reorder_projection_weight(W, S, C).contiguous()
+        Wre8 = Wre.to(torch.float8_e4m3fn)
+        V8 = ext.row_gemm_fp8(U8.reshape(B * N * Nl, S * C), Wre8)
+        Vo8 = _a2a(V8.view(B, N, Nl, S * Hd).view(torch.uint8), "d2o",
+                   group).view(torch.float8_e4m3fn)
+        bias_f32 = bias.float().contiguous() if bias is not None else None
+        Y, Y8 = ext.bdgcn_mode2_fp8_train(
+            Vo8.view(B, Nl, N * S, Hd).contiguous(), gop.A2T8, bias_f32,
+            relu, N, S, not last, gop.id_first,
+        )  # Y: (B, Nl, N, Hd) origin-sharded
+        if last:
+            Xd = _a2a(Y, "o2d", group)
+            Xd8 = Y8  # undefined tensor (not emitted)
+        else:
+            Xd8 = _a2a(Y8.view(torch.uint8), "o2d",
+                       group).view(torch.float8_e4m3fn)
+            Xd = Xd8.to(torch.bfloat16)
+        ctx.save_for_backward(U8, Wre, Y)
+        ctx.gop = gop
+        ctx.relu = relu
+        ctx.has_bias = bias is not None
+        ctx.dims = (B, N, Nl, S, C, Hd)
+        ctx.group = group
+        ctx.st = st
+        if last:
+            return Xd
+        ctx.mark_non_differentiable(Xd8)
+        return Xd, Xd8
+
+    @staticmethod
+    def backward(ctx, dXd, _dXd8=None):
+        from mpgcn_amd.ops.functional import _FP8_MARGIN, _ops
+
+        ext = _ops.get_ext()
+        U8, Wre, Y = ctx.saved_tensors
+        gop: GraphOperator = ctx.gop
+        st = ctx.st
+        B, N, Nl, S, C, Hd = ctx.dims
+        # inverse of the Y exchange: dest-sharded grad -> origin-sharded
+        dY = _a2a(dXd.contiguous(), "d2o", ctx.group).contiguous()
+        ext.fp8_scale_update(st["amax_y"], st["scale_y"], st["inv_y"], _FP8_MARGIN)
+        dYm, dY8, dbias = ext.relu_bwd_colsum_fp8(dY, Y, ctx.relu,
+                                                  st["scale_y"], st["amax_y"])
+        if not ctx.has_bias:
+            dbias = None
+        dV = ext.bdgcn_mode2_bwd_fp8(dY8.view(B, Nl, N, Hd), gop.A28, S,
+                                     st["inv_y"], dYm.view(B, Nl, N, Hd),
+                                     gop.id_first)  # (B, Nl, N, S, Hd)
+        # inverse of the V exchange: origin-sharded grad -> dest-sharded
+        dVd = _a2a(dV.reshape(B, Nl, N, S * Hd), "o2d", ctx.group)
+        R = B * N * Nl
+        dVflat = dVd.reshape(R, S * Hd).contiguous()
+        dWreT, _, _ = ext.red_gemm(dVflat, U8.reshape(R, S * C), False, None, 0, 0)
+        dWre = dWreT.t().to(dXd.dtype)
+        dW = dWre.reshape(S, C, S, Hd).permute(0, 2, 1, 3).reshape(S * S * C, Hd)
+        ext.fp8_scale_update(st["amax_u"], st["scale_u"], st["inv_u"], _FP8_MARGIN)
+        dU8 = ext.row_gemm_fp8_out(dVflat, Wre.t().contiguous(),
+                                   st["scale_u"], st["amax_u"])
+        dX = ext.bdgcn_mode1_bwd_fp8(dU8.view(B, N, Nl, S, C), gop.A3T8,
+                                     st["inv_u"], gop.id_first)
+        return dX, dW, dbias, None, None, None, None, None, None
+
+
+def region_fp8_compatible(N: int, P: int, C: int, Hd: int, S: int) -> bool:
+    """Shape gate for the fp8 sharded layer (vector-only fp8 tiles on the
+    shard extents Nl = N/P)."""
+    Nl = N // P
+    return ((Nl * C) % 256 == 0 and C % 16 == 0 and (Nl * Hd) % 256 == 0
+            and Hd % 16 == 0 and (S * C) % 16 == 0 and S * Hd <= 128)
+
+
 def bdgcn_layer_sharded(Xd, gop: GraphOperator, W, bias, group=None, relu=True):
     """One BDGCN layer on a destination-sharded input.
 
@@ -153,13 +251,16 @@ def mpgcn_forward_sharded(model, x_seq_shard, G_list, group=None):
         raise ValueError(
             f"region partition needs the region count ({N}) divisible by the "
             f"world size ({P}); pad the grid or change the rank count")
-    if getattr(model, "fp8_forward", False):
+    fp8 = bool(getattr(model, "fp8_forward", False))
+    if fp8 and not region_fp8_compatible(
+            N, P, model.lstm_hidden_dim,
+            model.branch_models[0]["spatial"][0].hidden_dim, model.K):
         import warnings
 
         warnings.warn(
-            "fp8_forward is not implemented for the region-sharded path; "
-            "running the shards in bf16 (same fp32-master weights)",
-            stacklevel=2)
+            "fp8_forward shard shapes incompatible (region_fp8_compatible); "
+            "running the region shards in bf16", stacklevel=2)
+        fp8 = False
     gops = model._graph_operators(G_list)
     cd = model.compute_dtype
     lstm_in = (
@@ -177,10 +278,29 @@ def mpgcn_forward_sharded(model, x_seq_shard, G_list, group=None):
         )
         X = h.reshape(B, N, Nl, model.lstm_hidden_dim)
         gop = gops[m]
-        for layer in branch["spatial"]:
-            X = bdgcn_layer_sharded(
-                X, gop, layer.W.to(cd), layer.b, group, relu=layer.relu
-            )
+        if fp8:
+            from mpgcn_amd.ops.functional import make_fp8_state
+
+            X8 = None
+            n_sp = len(branch["spatial"])
+            for i, layer in enumerate(branch["spatial"]):
+                st = getattr(layer, "_fp8_state", None)
+                if st is None or st["amax_u"].device != X.device:
+                    st = layer._fp8_state = make_fp8_state(X.device)
+                last = i + 1 == n_sp
+                out = _ShardedFp8Layer.apply(
+                    X, layer.W.to(cd), layer.b, gop, layer.relu, X8,
+                    group, st, last,
+                )
+                if last:
+                    X = out
+                else:
+                    X, X8 = out
+        else:
+            for layer in branch["spatial"]:
+                X = bdgcn_layer_sharded(
+                    X, gop, layer.W.to(cd), layer.b, group, relu=layer.relu
+                )
         fc = branch["fc"][0]
         out = linear_act(X.reshape(B * N * Nl, -1), fc.weight.to(cd), fc.bias, True)
         outs.append(out.view(B, N, Nl, 1))
