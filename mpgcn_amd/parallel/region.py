@@ -251,7 +251,7 @@ def mpgcn_forward_sharded(model, x_seq_shard, G_list, group=None):
         raise ValueError(
             f"region partition needs the region count ({N}) divisible by the "
             f"world size ({P}); pad the grid or change the rank count")
-    fp8 = bool(getattr(model, "fp8_forward", False))
+    fp8 = bool(getattr(model, "fp8_forward", False)) and x_seq_shard.is_cuda
     if fp8 and not region_fp8_compatible(
             N, P, model.lstm_hidden_dim,
             model.branch_models[0]["spatial"][0].hidden_dim, model.K):
