@@ -198,6 +198,21 @@ class ModelTrainer:
                         y_pred, y_tgt = self._step_forward(
                             x_seq, y_true, O_dyn_G, D_dyn_G
                         )
+                        if epoch == start_epoch and step == 0 and mode == "train":
+                            # The architecture (reference-faithful: stacked
+                            # ReLU GCN layers into a ReLU FC head, MPGCN.py
+                            # :47-49,74-76) can initialize DEAD: every output
+                            # zero => zero gradients forever (reproducible at
+                            # e.g. seed 4). Detect and say so instead of
+                            # silently "training" a constant model.
+                            if not bool((y_pred.detach() != 0).any().item()):
+                                log(
+                                    "[mpgcn] WARNING: model output is "
+                                    "identically zero at initialization "
+                                    "(dead ReLU chain) — gradients are zero "
+                                    "and training cannot progress; re-seed "
+                                    "(-seed) and restart."
+                                )
                         loss = self.criterion(y_pred, y_tgt)
                         if mode == "train":
                             self.optimizer.zero_grad(set_to_none=True)
