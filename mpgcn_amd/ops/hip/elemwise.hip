@@ -127,6 +127,43 @@ extern "C" void slab_colsum_launch(const float* ws, float* out, long nb, long E,
         ws, out, nb, E);
 }
 
+// Three slab column-sums in ONE launch (the fused-LSTM backward's dW_hh /
+// dbias / dw_ih workspace reductions are sequential tiny kernels on each
+// branch stream's critical path — one launch removes two launch+drain gaps).
+__launch_bounds__(256) __global__ void slab_colsum3_kernel(
+    const float* __restrict__ w1, float* __restrict__ o1, long E1,
+    const float* __restrict__ w2, float* __restrict__ o2, long E2,
+    const float* __restrict__ w3, float* __restrict__ o3, long E3, long nb) {
+    long e = (long)blockIdx.x * 256 + threadIdx.x;
+    const float* ws;
+    float* out;
+    long E;
+    if (e < E1) {
+        ws = w1; out = o1; E = E1;
+    } else if (e < E1 + E2) {
+        e -= E1; ws = w2; out = o2; E = E2;
+    } else {
+        e -= E1 + E2; ws = w3; out = o3; E = E3;
+        if (e >= E) return;
+    }
+    float s[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+    long b = 0, off = e;
+    for (; b + 8 <= nb; b += 8, off += 8 * E)
+#pragma unroll
+        for (int j = 0; j < 8; ++j) s[j] += ws[off + j * E];
+    for (int j = 0; b < nb; ++b, ++j, off += E) s[j] += ws[off];
+    out[e] = ((s[0] + s[1]) + (s[2] + s[3])) + ((s[4] + s[5]) + (s[6] + s[7]));
+}
+
+extern "C" void slab_colsum3_launch(const float* w1, float* o1, long E1,
+                                    const float* w2, float* o2, long E2,
+                                    const float* w3, float* o3, long E3,
+                                    long nb, hipStream_t s) {
+    const long tot = E1 + E2 + E3;
+    slab_colsum3_kernel<<<dim3((unsigned)((tot + 255) / 256)), dim3(256), 0, s>>>(
+        w1, o1, E1, w2, o2, E2, w3, o3, E3, nb);
+}
+
 // Strided identity-slot copy: dst rows at stride dst_stride_b receive the
 // contiguous src rows (row_b bytes each, multiple of 16). Fills the
 // identity-support slots of U/dV in id_skip mode — aten's strided

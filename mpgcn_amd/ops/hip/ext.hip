@@ -489,12 +489,10 @@ std::vector<torch::Tensor> lstm_fused_bwd(torch::Tensor x, long x_off,
     auto dwhh = torch::empty({128, 32}, f32);
     auto dbias = torch::empty({128}, f32);
     auto dwih = torch::empty({128}, f32);
-    slab_colsum_launch(ws_dw.data_ptr<float>(), dwhh.data_ptr<float>(), nb,
-                       128 * 32, stream());
-    slab_colsum_launch(ws_db.data_ptr<float>(), dbias.data_ptr<float>(), nb,
-                       128, stream());
-    slab_colsum_launch(ws_dwih.data_ptr<float>(), dwih.data_ptr<float>(), nb,
-                       128, stream());
+    slab_colsum3_launch(ws_dw.data_ptr<float>(), dwhh.data_ptr<float>(), 128 * 32,
+                        ws_db.data_ptr<float>(), dbias.data_ptr<float>(), 128,
+                        ws_dwih.data_ptr<float>(), dwih.data_ptr<float>(), 128,
+                        nb, stream());
     return {dwhh, dbias, dwih, dh_prev, dc_prev};
 }
 

@@ -149,6 +149,10 @@ void fp8_scale_update_launch(float* amax, float* scale, float* inv,
                              float margin, hipStream_t s);
 void slot_copy_launch(const void* src, void* dst, long rows, int row_b,
                       long dst_stride_b, hipStream_t s);
+void slab_colsum3_launch(const float* w1, float* o1, long E1,
+                         const float* w2, float* o2, long E2,
+                         const float* w3, float* o3, long E3,
+                         long nb, hipStream_t s);
 void rwd_rowsum_launch(const float* A, float* d, long rows, int N, hipStream_t s);
 void rwd_norm_t_launch(const float* A, const float* d, float* OUT, long B,
                        int N, long sOUT, hipStream_t s);
