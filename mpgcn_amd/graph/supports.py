@@ -124,14 +124,24 @@ def build_supports(
         raise ValueError(f"flow must be (B, N, N); got {tuple(flow.shape)}")
     K = get_support_K(kernel_type, cheby_order)
 
-    if (kernel_type == "random_walk_diffusion" and flow.is_cuda
-            and cheby_order >= 1):
-        # fused HIP build (K8): 2 + (order-1) launches instead of the ~10
+    if flow.is_cuda and cheby_order >= 1 and (
+            kernel_type == "random_walk_diffusion"
+            or kernel_type == "dual_random_walk_diffusion"
+            or (kernel_type == "chebyshev" and lambda_max is not None)):
+        # fused HIP builds (K8): a handful of launches instead of the ~10+
         # stock-op chain below; the torch path remains the numerics oracle
-        # (tests/test_supports.py) and the CPU / other-kernel-type fallback
+        # (tests/test_supports.py) and the CPU / localpool / power-iteration
+        # fallback
         from mpgcn_amd import ops as _ops
 
-        out = _ops.get_ext().rwd_supports(flow.float().contiguous(), cheby_order)
+        ext = _ops.get_ext()
+        f32 = flow.float().contiguous()
+        if kernel_type == "random_walk_diffusion":
+            out = ext.rwd_supports(f32, cheby_order)
+        elif kernel_type == "dual_random_walk_diffusion":
+            out = ext.dual_rwd_supports(f32, cheby_order)
+        else:
+            out = ext.cheby_supports(f32, cheby_order, float(lambda_max))
         out._identity_first = True
         return out
 

@@ -254,6 +254,86 @@ extern "C" void rwd_norm_t_launch(const float* A, const float* d, float* OUT,
         A, d, OUT, N, sOUT);
 }
 
+// Column sums (dual-RWD's backward series needs colsum_A: the row sums of
+// A^T without materializing the transpose). Coalesced: consecutive threads
+// cover consecutive columns n, looping rows m.
+__launch_bounds__(256) __global__ void rwd_colsum_kernel(
+    const float* __restrict__ A, float* __restrict__ dc, long B, int N) {
+    const long b = blockIdx.y;
+    const int n = blockIdx.x * 256 + threadIdx.x;
+    if (n >= N) return;
+    const float* Ab = A + b * (long)N * N;
+    float s = 0.f;
+    for (int m = 0; m < N; ++m) s += Ab[(long)m * N + n];
+    dc[b * (long)N + n] = s;
+}
+
+extern "C" void rwd_colsum_launch(const float* A, float* dc, long B, int N,
+                                  hipStream_t s) {
+    rwd_colsum_kernel<<<dim3((unsigned)((N + 255) / 256), (unsigned)B),
+                        dim3(256), 0, s>>>(A, dc, B, N);
+}
+
+// dual-RWD backward-series seed: OUT_slot[m, n] = A[m, n] / colsum_A[n]
+// (P_bwd^T in NATURAL orientation — no transpose needed). Fully coalesced.
+__launch_bounds__(256) __global__ void dual_bwd_norm_kernel(
+    const float* __restrict__ A, const float* __restrict__ dc,
+    float* __restrict__ OUT, int N, long sOUT) {
+    const long b = blockIdx.z;
+    const long idx = (long)blockIdx.y * (gridDim.x * 256L) +
+                     (long)blockIdx.x * 256 + threadIdx.x;
+    if (idx >= (long)N * N) return;
+    const int n = (int)(idx % N);
+    const float d = dc[b * (long)N + n];
+    OUT[b * sOUT + idx] =
+        d != 0.f ? A[b * (long)N * N + idx] / d : 0.f;
+}
+
+extern "C" void dual_bwd_norm_launch(const float* A, const float* dc,
+                                     float* OUT, long B, int N, long sOUT,
+                                     hipStream_t s) {
+    const long total = (long)N * N;
+    unsigned gx = (unsigned)((total + 255) / 256);
+    unsigned gy = 1;
+    while (gx > 65535) { gx = (gx + 1) / 2; gy *= 2; }
+    dual_bwd_norm_kernel<<<dim3(gx, gy, (unsigned)B), dim3(256), 0, s>>>(
+        A, dc, OUT, N, sOUT);
+}
+
+// Chebyshev seed (lambda_max fixed): OUT[0] = I and
+// OUT[1][m,n] = ((2/lam)-1)*I[m,n] - (2/lam)*A[m,n]*rsqrt(d[m])*rsqrt(d[n])
+// — the rescaled Laplacian of the symmetric-normalized graph (GCN.py:74-77,
+// 110-126 semantics with the always-taken lam=2 fallback). Natural
+// orientation, fully coalesced; empty rows guarded like the torch path.
+__launch_bounds__(256) __global__ void cheby_seed_kernel(
+    const float* __restrict__ A, const float* __restrict__ d,
+    float* __restrict__ OUT, int N, long sOUT, float lam) {
+    const long b = blockIdx.z;
+    const long idx = (long)blockIdx.y * (gridDim.x * 256L) +
+                     (long)blockIdx.x * 256 + threadIdx.x;
+    if (idx >= (long)N * N) return;
+    const int m = (int)(idx / N), n = (int)(idx % N);
+    const float dm = d[b * (long)N + m], dn = d[b * (long)N + n];
+    const float sym = (dm > 0.f && dn > 0.f)
+        ? A[b * (long)N * N + idx] * rsqrtf(dm) * rsqrtf(dn) : 0.f;
+    const float c = 2.f / lam;
+    const float eye = (m == n) ? 1.f : 0.f;
+    float* O0 = OUT + b * sOUT;
+    O0[idx] = eye;
+    O0[(long)N * N + idx] = (c - 1.f) * eye - c * sym;
+}
+
+extern "C" void cheby_seed_launch(const float* A, const float* d, float* OUT,
+                                  long B, int N, long sOUT, float lam,
+                                  hipStream_t s) {
+    const long total = (long)N * N;
+    unsigned gx = (unsigned)((total + 255) / 256);
+    unsigned gy = 1;
+    while (gx > 65535) { gx = (gx + 1) / 2; gy *= 2; }
+    cheby_seed_kernel<<<dim3(gx, gy, (unsigned)B), dim3(256), 0, s>>>(
+        A, d, OUT, N, sOUT, lam);
+}
+
 // Delayed-scaling bookkeeping for the fp8 gradient path: derive this step's
 // quantize scale (and its exact descale pair) from LAST step's recorded
 // amax, then reset the amax accumulator — one thread, device-side only, so

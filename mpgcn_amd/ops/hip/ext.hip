@@ -905,6 +905,97 @@ torch::Tensor rwd_supports(torch::Tensor flow, long order) {
     return out;
 }
 
+namespace {
+// shared Chebyshev-recurrence GEMM: OUT slot k = 2 * A_slot @ X_slot - C_slot
+void cheby_gemm(float* base, long a_off, long x_off, long o_off, long c_off,
+                long inst_stride, long B, long N) {
+    AxisGemmParams p{};
+    p.AT = base + a_off;
+    p.X = base + x_off;
+    p.OUT = base + o_off;
+    p.CSUB = base + c_off;
+    p.alpha = 2.f; p.cs_beta = -1.f;
+    p.M = (int)N; p.K = (int)N; p.L = (int)N;
+    p.a_div = 1; p.a_bs1 = inst_stride; p.a_bs2 = 0;
+    p.x_div = 1; p.x_bs1 = inst_stride; p.x_bs2 = 0;
+    p.o_div = 1; p.o_bs1 = inst_stride; p.o_bs2 = 0;
+    p.cs_div = 1; p.cs_bs1 = inst_stride; p.cs_bs2 = 0;
+    p.cs_row = N;
+    p.kdiv = 1; p.k_lo = N;
+    p.qdiv = 0;
+    p.o_row = N;
+    p.ogdiv = 0;
+    p.a_vec = (N % 4 == 0);
+    p.x_vec = (N % 4 == 0);
+    axis_gemm_launch(p, (int)B, /*is_f32=*/1, stream());
+}
+}  // namespace
+
+// Fused dual-random-walk-diffusion support build: fwd series (rowsum-
+// normalized P^T polynomials) at slots 0..order, bwd series (colsum-
+// normalized, i.e. P_bwd^T in natural orientation) at slots order+1..2order,
+// sharing T_0 = I (GCN.py:84-91 semantics).
+torch::Tensor dual_rwd_supports(torch::Tensor flow, long order) {
+    TORCH_CHECK(flow.is_cuda() && flow.is_contiguous() &&
+                flow.scalar_type() == torch::kFloat, "flow must be CUDA f32");
+    TORCH_CHECK(flow.dim() == 3 && flow.size(1) == flow.size(2), "flow (B,N,N)");
+    TORCH_CHECK(order >= 1, "needs order >= 1");
+    const long B = flow.size(0), N = flow.size(1), S = 2 * order + 1;
+    TORCH_CHECK(B <= 65535, "too many instances");
+    auto out = torch::empty({B, S, N, N}, flow.options());
+    auto d = torch::empty({B, N}, flow.options());
+    auto dc = torch::empty({B, N}, flow.options());
+    const long st = S * N * N;
+    float* base = out.data_ptr<float>();
+    rwd_rowsum_launch(flow.data_ptr<float>(), d.data_ptr<float>(), B * N,
+                      (int)N, stream());
+    rwd_colsum_launch(flow.data_ptr<float>(), dc.data_ptr<float>(), B, (int)N,
+                      stream());
+    // slots 0 (I) and 1 (P_fwd^T)
+    rwd_norm_t_launch(flow.data_ptr<float>(), d.data_ptr<float>(), base, B,
+                      (int)N, st, stream());
+    // slot order+1 (P_bwd^T, natural orientation)
+    dual_bwd_norm_launch(flow.data_ptr<float>(), dc.data_ptr<float>(),
+                         base + (order + 1) * N * N, B, (int)N, st, stream());
+    for (long k = 2; k <= order; ++k)
+        cheby_gemm(base, 1 * N * N, (k - 1) * N * N, k * N * N,
+                   (k - 2) * N * N, st, B, N);
+    for (long k = 2; k <= order; ++k) {
+        // bwd_j lives at slot order + j (j >= 1); bwd_0 is the shared I
+        const long a_s = order + 1;          // PT_bwd = bwd_1
+        const long x_s = order + (k - 1);    // bwd_{k-1}
+        const long o_s = order + k;          // bwd_k
+        const long c_s = (k == 2) ? 0 : order + (k - 2);
+        cheby_gemm(base, a_s * N * N, x_s * N * N, o_s * N * N,
+                   c_s * N * N, st, B, N);
+    }
+    return out;
+}
+
+// Fused Chebyshev support build (fixed lambda_max — the reference path,
+// since its torch.eig fallback always fires, GCN.py:116-126): rowsum ->
+// fused rescaled-Laplacian seed -> recurrence GEMMs.
+torch::Tensor cheby_supports(torch::Tensor flow, long order, double lam) {
+    TORCH_CHECK(flow.is_cuda() && flow.is_contiguous() &&
+                flow.scalar_type() == torch::kFloat, "flow must be CUDA f32");
+    TORCH_CHECK(flow.dim() == 3 && flow.size(1) == flow.size(2), "flow (B,N,N)");
+    TORCH_CHECK(order >= 1, "needs order >= 1");
+    const long B = flow.size(0), N = flow.size(1), S = order + 1;
+    TORCH_CHECK(B <= 65535, "too many instances");
+    auto out = torch::empty({B, S, N, N}, flow.options());
+    auto d = torch::empty({B, N}, flow.options());
+    const long st = S * N * N;
+    float* base = out.data_ptr<float>();
+    rwd_rowsum_launch(flow.data_ptr<float>(), d.data_ptr<float>(), B * N,
+                      (int)N, stream());
+    cheby_seed_launch(flow.data_ptr<float>(), d.data_ptr<float>(), base, B,
+                      (int)N, st, (float)lam, stream());
+    for (long k = 2; k <= order; ++k)
+        cheby_gemm(base, 1 * N * N, (k - 1) * N * N, k * N * N,
+                   (k - 2) * N * N, st, B, N);
+    return out;
+}
+
 // Fused ReLU backward + bias-grad column sum: dY = dH * 1[Y>0], dbias=colsum.
 std::vector<torch::Tensor> relu_bwd_colsum(torch::Tensor dH, torch::Tensor Y,
                                            bool mask) {
@@ -958,6 +1049,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "fp8-forward mode-2 + bias + act, bf16 out + fp8 twin");
     m.def("row_gemm_fp8", &row_gemm_fp8, "fp8 projection GEMM");
     m.def("rwd_supports", &rwd_supports, "fused random-walk-diffusion support build (K8)");
+    m.def("dual_rwd_supports", &dual_rwd_supports, "fused dual-RWD support build");
+    m.def("cheby_supports", &cheby_supports, "fused Chebyshev support build");
     m.def("bdgcn_mode2_bwd_fp8", &bdgcn_mode2_bwd_fp8,
           "scaled fp8 gradient contraction dV");
     m.def("bdgcn_mode1_bwd_fp8", &bdgcn_mode1_bwd_fp8,

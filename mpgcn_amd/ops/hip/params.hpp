@@ -154,6 +154,11 @@ void slab_colsum3_launch(const float* w1, float* o1, long E1,
                          const float* w3, float* o3, long E3,
                          long nb, hipStream_t s);
 void rwd_rowsum_launch(const float* A, float* d, long rows, int N, hipStream_t s);
+void rwd_colsum_launch(const float* A, float* dc, long B, int N, hipStream_t s);
+void dual_bwd_norm_launch(const float* A, const float* dc, float* OUT, long B,
+                          int N, long sOUT, hipStream_t s);
+void cheby_seed_launch(const float* A, const float* d, float* OUT, long B,
+                       int N, long sOUT, float lam, hipStream_t s);
 void rwd_norm_t_launch(const float* A, const float* d, float* OUT, long B,
                        int N, long sOUT, hipStream_t s);
 }

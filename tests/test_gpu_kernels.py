@@ -253,20 +253,24 @@ def test_gcn1d_hip_path_matches_eager(dtype):
     torch.testing.assert_close(got.float().cpu(), ref, **_tol(dtype))
 
 
+@pytest.mark.parametrize("kernel", ["random_walk_diffusion",
+                                    "dual_random_walk_diffusion", "chebyshev"])
 @pytest.mark.parametrize("order", [1, 2, 4])
 @pytest.mark.parametrize("N_", [47, 256])
-def test_fused_rwd_supports_matches_torch_path(order, N_):
-    """The fused K8 support build (rowsum + normalize/transpose/identity +
-    alpha/CSUB Chebyshev GEMMs) vs the stock torch chain on CPU."""
+def test_fused_supports_match_torch_path(kernel, order, N_):
+    """The fused K8 support builds (rowsum/colsum + fused seed kernels +
+    alpha/CSUB Chebyshev GEMMs) vs the stock torch chain on CPU, for all
+    three fused kernel types."""
     from mpgcn_amd.graph import build_supports
 
     torch.manual_seed(order * 31 + N_)
     flow = torch.rand(5, N_, N_) * 4
-    flow[0, 3] = 0  # empty row: rcp guard
-    ref = build_supports(flow, "random_walk_diffusion", order)  # CPU torch
-    got = build_supports(flow.to(DEV), "random_walk_diffusion", order)  # fused
+    flow[0, 3] = 0  # empty row: rcp/rsqrt guards
+    flow[1, :, 5] = 0  # empty column: dual-RWD backward-series guard
+    ref = build_supports(flow, kernel, order)  # CPU torch
+    got = build_supports(flow.to(DEV), kernel, order)  # fused
     assert getattr(got, "_identity_first", False)
-    torch.testing.assert_close(got.cpu(), ref, atol=1e-4, rtol=1e-4)
+    torch.testing.assert_close(got.cpu(), ref, atol=2e-4, rtol=2e-4)
 
 
 @pytest.mark.parametrize("dyn", [False, True])
