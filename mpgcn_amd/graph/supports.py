@@ -124,10 +124,12 @@ def build_supports(
         raise ValueError(f"flow must be (B, N, N); got {tuple(flow.shape)}")
     K = get_support_K(kernel_type, cheby_order)
 
-    if flow.is_cuda and cheby_order >= 1 and (
-            kernel_type == "random_walk_diffusion"
-            or kernel_type == "dual_random_walk_diffusion"
-            or (kernel_type == "chebyshev" and lambda_max is not None)):
+    if flow.is_cuda and (
+            (cheby_order >= 1 and (
+                kernel_type == "random_walk_diffusion"
+                or kernel_type == "dual_random_walk_diffusion"
+                or (kernel_type == "chebyshev" and lambda_max is not None)))
+            or kernel_type == "localpool"):
         # fused HIP builds (K8): a handful of launches instead of the ~10+
         # stock-op chain below; the torch path remains the numerics oracle
         # (tests/test_supports.py) and the CPU / localpool / power-iteration
@@ -136,6 +138,8 @@ def build_supports(
 
         ext = _ops.get_ext()
         f32 = flow.float().contiguous()
+        if kernel_type == "localpool":
+            return ext.localpool_supports(f32)  # single support: no id tag
         if kernel_type == "random_walk_diffusion":
             out = ext.rwd_supports(f32, cheby_order)
         elif kernel_type == "dual_random_walk_diffusion":

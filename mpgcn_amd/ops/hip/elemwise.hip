@@ -334,6 +334,32 @@ extern "C" void cheby_seed_launch(const float* A, const float* d, float* OUT,
         A, d, OUT, N, sOUT, lam);
 }
 
+// localpool seed (Kipf): OUT[0][m,n] = I[m,n] + A[m,n]*rsqrt(d[m])*rsqrt(d[n])
+__launch_bounds__(256) __global__ void localpool_seed_kernel(
+    const float* __restrict__ A, const float* __restrict__ d,
+    float* __restrict__ OUT, int N, long sOUT) {
+    const long b = blockIdx.z;
+    const long idx = (long)blockIdx.y * (gridDim.x * 256L) +
+                     (long)blockIdx.x * 256 + threadIdx.x;
+    if (idx >= (long)N * N) return;
+    const int m = (int)(idx / N), n = (int)(idx % N);
+    const float dm = d[b * (long)N + m], dn = d[b * (long)N + n];
+    const float sym = (dm > 0.f && dn > 0.f)
+        ? A[b * (long)N * N + idx] * rsqrtf(dm) * rsqrtf(dn) : 0.f;
+    OUT[b * sOUT + idx] = ((m == n) ? 1.f : 0.f) + sym;
+}
+
+extern "C" void localpool_seed_launch(const float* A, const float* d,
+                                      float* OUT, long B, int N, long sOUT,
+                                      hipStream_t s) {
+    const long total = (long)N * N;
+    unsigned gx = (unsigned)((total + 255) / 256);
+    unsigned gy = 1;
+    while (gx > 65535) { gx = (gx + 1) / 2; gy *= 2; }
+    localpool_seed_kernel<<<dim3(gx, gy, (unsigned)B), dim3(256), 0, s>>>(
+        A, d, OUT, N, sOUT);
+}
+
 // Delayed-scaling bookkeeping for the fp8 gradient path: derive this step's
 // quantize scale (and its exact descale pair) from LAST step's recorded
 // amax, then reset the amax accumulator — one thread, device-side only, so

@@ -996,6 +996,22 @@ torch::Tensor cheby_supports(torch::Tensor flow, long order, double lam) {
     return out;
 }
 
+// Fused localpool support build: (B,N,N) -> (B,1,N,N), I + sym_norm(A)
+// (GCN.py:71-72) in two launches.
+torch::Tensor localpool_supports(torch::Tensor flow) {
+    TORCH_CHECK(flow.is_cuda() && flow.is_contiguous() &&
+                flow.scalar_type() == torch::kFloat, "flow must be CUDA f32");
+    TORCH_CHECK(flow.dim() == 3 && flow.size(1) == flow.size(2), "flow (B,N,N)");
+    const long B = flow.size(0), N = flow.size(1);
+    auto out = torch::empty({B, 1, N, N}, flow.options());
+    auto d = torch::empty({B, N}, flow.options());
+    rwd_rowsum_launch(flow.data_ptr<float>(), d.data_ptr<float>(), B * N,
+                      (int)N, stream());
+    localpool_seed_launch(flow.data_ptr<float>(), d.data_ptr<float>(),
+                          out.data_ptr<float>(), B, (int)N, N * N, stream());
+    return out;
+}
+
 // Fused ReLU backward + bias-grad column sum: dY = dH * 1[Y>0], dbias=colsum.
 std::vector<torch::Tensor> relu_bwd_colsum(torch::Tensor dH, torch::Tensor Y,
                                            bool mask) {
@@ -1051,6 +1067,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("rwd_supports", &rwd_supports, "fused random-walk-diffusion support build (K8)");
     m.def("dual_rwd_supports", &dual_rwd_supports, "fused dual-RWD support build");
     m.def("cheby_supports", &cheby_supports, "fused Chebyshev support build");
+    m.def("localpool_supports", &localpool_supports, "fused localpool support build");
     m.def("bdgcn_mode2_bwd_fp8", &bdgcn_mode2_bwd_fp8,
           "scaled fp8 gradient contraction dV");
     m.def("bdgcn_mode1_bwd_fp8", &bdgcn_mode1_bwd_fp8,

@@ -157,6 +157,8 @@ void rwd_rowsum_launch(const float* A, float* d, long rows, int N, hipStream_t s
 void rwd_colsum_launch(const float* A, float* dc, long B, int N, hipStream_t s);
 void dual_bwd_norm_launch(const float* A, const float* dc, float* OUT, long B,
                           int N, long sOUT, hipStream_t s);
+void localpool_seed_launch(const float* A, const float* d, float* OUT,
+                           long B, int N, long sOUT, hipStream_t s);
 void cheby_seed_launch(const float* A, const float* d, float* OUT, long B,
                        int N, long sOUT, float lam, hipStream_t s);
 void rwd_norm_t_launch(const float* A, const float* d, float* OUT, long B,

@@ -253,6 +253,17 @@ def test_gcn1d_hip_path_matches_eager(dtype):
     torch.testing.assert_close(got.float().cpu(), ref, **_tol(dtype))
 
 
+def test_fused_localpool_supports_match_torch_path():
+    from mpgcn_amd.graph import build_supports
+
+    torch.manual_seed(2)
+    flow = torch.rand(5, 47, 47) * 4
+    flow[0, 3] = 0
+    ref = build_supports(flow, "localpool", 1)
+    got = build_supports(flow.to(DEV), "localpool", 1)
+    torch.testing.assert_close(got.cpu(), ref, atol=1e-5, rtol=1e-5)
+
+
 @pytest.mark.parametrize("kernel", ["random_walk_diffusion",
                                     "dual_random_walk_diffusion", "chebyshev"])
 @pytest.mark.parametrize("order", [1, 2, 4])
