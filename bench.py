@@ -96,7 +96,16 @@ def main():
                       num_nodes=N, compute_dtype=cdtype,
                       fusion=args.fusion, fp8_forward=fp8).to(device)
     use_graph = args.graph and is_cuda and world == 1 and args.impl == "native"
-    opt = torch.optim.Adam(model.parameters(), lr=1e-4, capturable=use_graph)
+    if args.impl == "native":
+        # fused flat-buffer Adam: one kernel per step instead of capturable
+        # Adam's ~45 tiny launches (ops/optim.py); must be built before the
+        # reducer registers its grad hooks since it repoints p.data/p.grad
+        from mpgcn_amd.ops.optim import FlatAdam
+
+        opt = FlatAdam(model.parameters(), lr=1e-4)
+    else:
+        opt = torch.optim.Adam(model.parameters(), lr=1e-4,
+                               capturable=use_graph)
     reducer = GradAllReducer(model, ctx)
     criterion = torch.nn.MSELoss()
 

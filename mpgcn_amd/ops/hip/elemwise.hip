@@ -376,3 +376,66 @@ extern "C" void fp8_scale_update_launch(float* amax, float* scale, float* inv,
                                         float margin, hipStream_t s) {
     fp8_scale_update_kernel<<<dim3(1), dim3(1), 0, s>>>(amax, scale, inv, margin);
 }
+
+// ---------------------------------------------------------------------------
+// Fused flat-buffer Adam: ONE vectorized kernel over a single packed f32
+// parameter buffer, replacing torch's per-tensor foreach/capturable Adam
+// (~30 tiny launches + bias-correction pow/div kernels per step — they stay
+// launch-bound even inside a hipGraph replay). L2-style weight decay
+// (g + wd*p) matches torch.optim.Adam's default, NOT AdamW. The step counter
+// lives on device so the whole update is capture-safe with no host math.
+__global__ void adam_bump_t_kernel(float* t) { t[0] += 1.0f; }
+
+__global__ void adam_flat_kernel(float* __restrict__ p,
+                                 const float* __restrict__ g,
+                                 float* __restrict__ m, float* __restrict__ v,
+                                 const float* __restrict__ t, long E, float lr,
+                                 float b1, float b2, float eps, float wd) {
+    const float tf = t[0];
+    // bias-corrected step size folded into one scalar per pass
+    const float bc1 = 1.0f - powf(b1, tf);
+    const float bc2 = 1.0f - powf(b2, tf);
+    const float step = lr / bc1;
+    const float vnorm = rsqrtf(bc2);  // vhat = v / bc2 -> sqrt(vhat) = sqrt(v)*vnorm
+    const long stride = (long)gridDim.x * blockDim.x * 4;
+    for (long i = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4; i < E;
+         i += stride) {
+        if (i + 4 <= E) {
+            float4 pv = *reinterpret_cast<const float4*>(p + i);
+            float4 gv = *reinterpret_cast<const float4*>(g + i);
+            float4 mv = *reinterpret_cast<const float4*>(m + i);
+            float4 vv = *reinterpret_cast<const float4*>(v + i);
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                float gj = (&gv.x)[j] + wd * (&pv.x)[j];
+                float mj = b1 * (&mv.x)[j] + (1.0f - b1) * gj;
+                float vj = b2 * (&vv.x)[j] + (1.0f - b2) * gj * gj;
+                (&mv.x)[j] = mj;
+                (&vv.x)[j] = vj;
+                (&pv.x)[j] -= step * mj / (sqrtf(vj) * vnorm + eps);
+            }
+            *reinterpret_cast<float4*>(p + i) = pv;
+            *reinterpret_cast<float4*>(m + i) = mv;
+            *reinterpret_cast<float4*>(v + i) = vv;
+        } else {
+            for (long k = i; k < E; ++k) {
+                float gj = g[k] + wd * p[k];
+                float mj = b1 * m[k] + (1.0f - b1) * gj;
+                float vj = b2 * v[k] + (1.0f - b2) * gj * gj;
+                m[k] = mj;
+                v[k] = vj;
+                p[k] -= step * mj / (sqrtf(vj) * vnorm + eps);
+            }
+        }
+    }
+}
+
+extern "C" void adam_flat_launch(float* p, const float* g, float* m, float* v,
+                                 float* t, long E, float lr, float b1, float b2,
+                                 float eps, float wd, hipStream_t s) {
+    adam_bump_t_kernel<<<dim3(1), dim3(1), 0, s>>>(t);
+    long blocks = (E + 4 * 256 - 1) / (4 * 256);
+    if (blocks > 1024) blocks = 1024;
+    adam_flat_kernel<<<dim3((unsigned)blocks), dim3(256), 0, s>>>(
+        p, g, m, v, t, E, lr, b1, b2, eps, wd);
+}

@@ -860,6 +860,24 @@ void fp8_scale_update(torch::Tensor amax, torch::Tensor scale,
                             inv.data_ptr<float>(), (float)margin, stream());
 }
 
+// Fused flat-buffer Adam step (one launch + a 1-thread step bump). All
+// operands are packed f32 device buffers; the step counter t is a 1-element
+// f32 tensor that lives on device, so the call is hipGraph-capture-safe.
+void adam_flat(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+               torch::Tensor v, torch::Tensor t, double lr, double b1,
+               double b2, double eps, double wd) {
+    TORCH_CHECK(p.is_cuda() && p.is_contiguous() &&
+                    p.scalar_type() == torch::kFloat32,
+                "adam_flat: p must be contiguous f32 CUDA");
+    TORCH_CHECK(g.numel() == p.numel() && m.numel() == p.numel() &&
+                    v.numel() == p.numel(),
+                "adam_flat: buffer size mismatch");
+    adam_flat_launch(p.data_ptr<float>(), g.data_ptr<float>(),
+                     m.data_ptr<float>(), v.data_ptr<float>(),
+                     t.data_ptr<float>(), (long)p.numel(), (float)lr, (float)b1,
+                     (float)b2, (float)eps, (float)wd, stream());
+}
+
 // Fused dynamic-support build (K8, random_walk_diffusion): raw flow
 // (B, N, N) f32 -> (B, order+1, N, N) f32 support stack in 2 + (order-1)
 // launches (rowsum, fused normalize+transpose+identity, then one
@@ -1078,6 +1096,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "row GEMM emitting scaled fp8 only, amax tracked");
     m.def("fp8_scale_update", &fp8_scale_update,
           "delayed-scaling scale/inv update from amax");
+    m.def("adam_flat", &adam_flat, "fused flat-buffer Adam step");
     m.def("row_gemm_out", &row_gemm_out, "row GEMM into strided output slice");
     m.def("lstm_step_fwd", &lstm_step_fwd, "fused LSTM cell forward step (K4)");
     m.def("lstm_fused_fwd", &lstm_fused_fwd, "register-resident fused LSTM forward");

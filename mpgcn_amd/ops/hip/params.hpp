@@ -147,6 +147,9 @@ long relu_bwd_nblocks(long total);
 long red_gemm_nblocks(long R);
 void fp8_scale_update_launch(float* amax, float* scale, float* inv,
                              float margin, hipStream_t s);
+void adam_flat_launch(float* p, const float* g, float* m, float* v, float* t,
+                      long E, float lr, float b1, float b2, float eps, float wd,
+                      hipStream_t s);
 void slot_copy_launch(const void* src, void* dst, long rows, int row_b,
                       long dst_stride_b, hipStream_t s);
 void slab_colsum3_launch(const float* w1, float* o1, long E1,

@@ -103,7 +103,18 @@ class ModelTrainer:
         raise NotImplementedError("Invalid loss function.")
 
     def get_optimizer(self):
-        if self.params.get("optimizer", "Adam") != "Adam":
+        name = self.params.get("optimizer", "Adam")
+        if name == "FusedAdam":
+            # single-kernel flat-buffer Adam (ops/optim.py) — same math as
+            # Adam below, repoints params/grads into packed buffers
+            from mpgcn_amd.ops.optim import FlatAdam
+
+            return FlatAdam(
+                self.model.parameters(),
+                lr=self.params["learn_rate"],
+                weight_decay=self.params.get("decay_rate", 0),
+            )
+        if name != "Adam":
             raise NotImplementedError("Invalid optimizer name.")
         return optim.Adam(
             self.model.parameters(),
