@@ -150,3 +150,25 @@ def test_dead_initialization_warning(tmp_path, capsys):
         _pytest.skip("this torch build's RNG stream did not produce a dead init")
     trainer.train(loaders, ["train", "validate"])
     assert "identically zero at initialization" in capsys.readouterr().out
+
+
+def test_fused_adam_trainer_option(tmp_path, capsys):
+    # optimizer: "FusedAdam" routes to FlatAdam (ops/optim.py) and trains to
+    # a similar validation loss as stock Adam on the same seed/config
+    params, trainer, loaders = _setup(tmp_path, optimizer="FusedAdam")
+    from mpgcn_amd.ops.optim import FlatAdam
+
+    assert isinstance(trainer.optimizer, FlatAdam)
+    trainer.train(loaders, ["train", "validate"])
+    out = capsys.readouterr().out
+    fused_losses = re.findall(r"validation loss drops from (\S+) to (\S+)\.", out)
+    assert fused_losses, "FusedAdam training never improved validation loss"
+
+    params2, trainer2, loaders2 = _setup(tmp_path, optimizer="Adam")
+    trainer2.train(loaders2, ["train", "validate"])
+    out2 = capsys.readouterr().out
+    ref_losses = re.findall(r"validation loss drops from (\S+) to (\S+)\.", out2)
+    final_fused = float(fused_losses[-1][1])
+    final_ref = float(ref_losses[-1][1])
+    assert abs(final_fused - final_ref) < 0.25 * abs(final_ref) + 1e-3, (
+        final_fused, final_ref)
