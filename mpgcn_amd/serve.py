@@ -77,17 +77,75 @@ class Forecaster:
             )
             self.G_corr = tag_like(csup.squeeze(0), csup)
 
+        # hipGraph capture state: obs_len -> (graph, x_static, y_static, dow)
+        self._graphs: dict = {}
+        self._use_graph = (self.device.type == "cuda"
+                           and bool(params.get("capture_graph", True)))
+
+    # ---- hipGraph-captured request path ------------------------------------
+    # A serving forward is launch-dense (hundreds of small kernels for one
+    # request) while its shapes are fixed per deployment, so on CUDA the
+    # forward is captured ONCE per observed obs_len into a hipGraph and every
+    # request replays it: x_seq is copied into a static input buffer and the
+    # day-of-week is a device int64 scalar the captured index_select re-reads
+    # at each replay (the same i_buf pattern bench.py uses for the train
+    # step). Falls back to eager on capture failure or off-CUDA.
+
+    def _dyn_g_list(self, dow_buf: torch.Tensor) -> list:
+        from mpgcn_amd.graph.supports import tag_like
+
+        go = tag_like(self.G_o.index_select(0, dow_buf), self.G_o)
+        gd = tag_like(self.G_d.index_select(0, dow_buf), self.G_d)
+        g_list = [self.G_static, (go, gd)]
+        if int(self.params.get("perspectives", 2)) == 3:
+            g_list.append(self.G_corr)
+        return g_list
+
+    def _captured(self, cur: torch.Tensor):
+        key = int(cur.shape[1])
+        entry = self._graphs.get(key)
+        if entry is None:
+            xs = cur.clone()
+            dow_buf = torch.zeros(1, dtype=torch.long, device=self.device)
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):  # allocator/lazy-state warmup
+                for _ in range(2):
+                    self.model(x_seq=xs, G_list=self._dyn_g_list(dow_buf))
+            torch.cuda.current_stream().wait_stream(side)
+            torch.cuda.synchronize()
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                ys = self.model(x_seq=xs, G_list=self._dyn_g_list(dow_buf))
+            entry = (graph, xs, ys, dow_buf)
+            self._graphs[key] = entry
+        return entry
+
     @torch.no_grad()
     def forecast(self, x_seq: torch.Tensor, dow: int, horizon: int = 1) -> torch.Tensor:
         """x_seq: (T, N, N) or (T, N, N, 1) -> (horizon, N, N)."""
         if x_seq.dim() == 3:
             x_seq = x_seq.unsqueeze(-1)
         cur = x_seq.unsqueeze(0).float().to(self.device)  # (1, T, N, N, 1)
+        if self._use_graph:
+            try:
+                graph, xs, ys, dow_buf = self._captured(cur)
+            except Exception:  # capture unsupported here — serve eager
+                self._use_graph = False
+        preds = []
+        if self._use_graph:
+            xs.copy_(cur)
+            dow_buf.fill_(dow % 7)
+            for _ in range(horizon):
+                graph.replay()
+                step = ys.clone()  # ys is overwritten by the next replay
+                xs.copy_(torch.cat([xs[:, 1:], step], dim=1))
+                preds.append(step)
+            return torch.cat(preds, dim=1)[0, :, :, :, 0].cpu()
         g_list = [self.G_static,
                   (self.G_o[dow % 7:dow % 7 + 1], self.G_d[dow % 7:dow % 7 + 1])]
         if int(self.params.get("perspectives", 2)) == 3:
             g_list.append(self.G_corr)
-        preds = []
         for _ in range(horizon):
             step = self.model(x_seq=cur, G_list=g_list)  # (1, 1, N, N, 1)
             cur = torch.cat([cur[:, 1:], step], dim=1)
