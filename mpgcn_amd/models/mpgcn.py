@@ -51,11 +51,13 @@ class TemporalEncoder(nn.Module):
         for p in self.parameters():
             nn.init.uniform_(p, -stdv, stdv)
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
-        """x: (R, T) scalar sequences -> (R, hidden) last hidden state."""
+    def forward(self, x: torch.Tensor, T: int | None = None) -> torch.Tensor:
+        """x: (R, T) scalar sequences -> (R, hidden) last hidden state.
+        T: logical length when x rows are pre-padded to the chunk width."""
         w_ih = self.weight_ih_l0.to(x.dtype)
         w_hh = self.weight_hh_l0.to(x.dtype)
-        return fused_lstm_last(x, w_ih, w_hh, self.bias_ih_l0, self.bias_hh_l0)
+        return fused_lstm_last(x, w_ih, w_hh, self.bias_ih_l0, self.bias_hh_l0,
+                               T=T)
 
 
 class MPGCN(nn.Module):
@@ -152,19 +154,32 @@ class MPGCN(nn.Module):
         B, T, N = x_seq.shape[0], x_seq.shape[1], self.num_nodes
         gops = self._graph_operators(G_list)
 
-        # (B, T, N, N, 1) -> (B*N*N, T) scalar sequences
-        lstm_in = (
-            x_seq.to(self.compute_dtype)
-            .permute(0, 2, 3, 1, 4)
-            .reshape(B * N * N, T)
-            .contiguous()
-        )
+        # (B, T, N, N, 1) -> (B*N*N, T) scalar sequences. On CUDA the rows
+        # are laid out at the LSTM kernel's padded width (multiple of 8) in
+        # ONE fused cast+transpose+pad copy — the separate .to() cast,
+        # .contiguous() transpose copy and per-branch F.pad copies this
+        # replaces were ~150 MB/step of pure traffic at the flagship shape
+        if x_seq.is_cuda:
+            Tpad = -(-T // 8) * 8
+            lstm_in = torch.empty(B * N * N, Tpad, dtype=self.compute_dtype,
+                                  device=x_seq.device)
+            lv = lstm_in.view(B, N, N, Tpad)
+            lv[..., :T].copy_(x_seq.squeeze(-1).permute(0, 2, 3, 1))
+            if Tpad != T:
+                lv[..., T:].zero_()
+        else:
+            lstm_in = (
+                x_seq.to(self.compute_dtype)
+                .permute(0, 2, 3, 1, 4)
+                .reshape(B * N * N, T)
+                .contiguous()
+            )
 
         fp8 = self.fp8_forward and x_seq.is_cuda
 
         def run_branch(m: int) -> torch.Tensor:
             branch = self.branch_models[m]
-            h_last = branch["temporal"](lstm_in)  # (B*N*N, H)
+            h_last = branch["temporal"](lstm_in, T)  # (B*N*N, H)
             X = h_last.reshape(B, N, N, self.lstm_hidden_dim)
             if fp8:
                 # fp8 twins chain layer-to-layer through the dual-write

@@ -596,15 +596,24 @@ class _RegLSTMFn(torch.autograd.Function):
     longer falls off the slab-path cliff (round-1: 2.3x slower per sample)."""
 
     @staticmethod
-    def forward(ctx, x, w_ih, w_hh, b_ih, b_hh):
+    def forward(ctx, x, w_ih, w_hh, b_ih, b_hh, T=None):
         ext = _ops.get_ext()
         whh = w_hh.contiguous()
         wih_f = w_ih.reshape(-1).float().contiguous()
         bias_f = (b_ih.float() + b_hh.float()).contiguous()
-        T = x.shape[1]
+        T = x.shape[1] if T is None else T
         n_chunks = (T + 7) // 8
-        pad = n_chunks * 8 - T
-        xp = torch.nn.functional.pad(x, (0, pad)).contiguous() if pad else x.contiguous()
+        width = n_chunks * 8
+        ctx.pre_padded = x.shape[1] == width and width != T
+        if ctx.pre_padded:
+            # caller already laid x out at the kernel's padded row width
+            # (models/mpgcn.py fuses cast+permute+pad into one copy) — no
+            # extra pad pass here
+            xp = x.contiguous()
+        else:
+            pad = width - T
+            xp = (torch.nn.functional.pad(x, (0, pad)).contiguous()
+                  if pad else x.contiguous())
         h = c = None
         starts = []  # (h, c) entering each chunk; chunk 0 enters at zeros
         for ci in range(n_chunks):
@@ -646,25 +655,34 @@ class _RegLSTMFn(torch.autograd.Function):
             dbias = dbias_i if dbias is None else dbias + dbias_i
             dwih = dwih_i if dwih is None else dwih + dwih_i
         wdt = whh.dtype
+        if need_dx and ctx.pre_padded:
+            dx[:, T:].zero_()  # grad must match the padded input layout
         return (
-            dx[:, :T].contiguous() if need_dx else None,
+            (dx if ctx.pre_padded else dx[:, :T].contiguous())
+            if need_dx else None,
             dwih.view(-1, 1).to(wdt),
             dwhh.to(wdt),
             dbias,
             dbias.clone(),
+            None,
         )
 
 
-def fused_lstm_last(x, w_ih, w_hh, b_ih, b_hh):
-    """Last hidden state of a 1-layer batch-first LSTM over (R, T) scalar inputs."""
+def fused_lstm_last(x, w_ih, w_hh, b_ih, b_hh, T=None):
+    """Last hidden state of a 1-layer batch-first LSTM over (R, T) scalar
+    inputs. T (optional) gives the logical sequence length when x rows are
+    pre-padded to the chunk width (multiple of 8) — the flagship path fuses
+    cast+permute+pad into one copy in models/mpgcn.py and passes T here."""
+    T_log = x.shape[1] if T is None else T
     Hd = w_hh.shape[1]
     if x.is_cuda and Hd == 32 and x.dtype == torch.bfloat16:
         # register-resident chunked schedule covers any T (chunks of <= 8)
-        return _RegLSTMFn.apply(x, w_ih, w_hh, b_ih, b_hh)
+        return _RegLSTMFn.apply(x, w_ih, w_hh, b_ih, b_hh, T_log)
+    xl = x if T_log == x.shape[1] else x[:, :T_log]
     kernel_ok = Hd == 32 or (Hd == 16 and x.dtype == torch.float32)
     if x.is_cuda and kernel_ok:
-        return _FusedLSTMLastFn.apply(x, w_ih, w_hh, b_ih, b_hh)
-    out, _, _ = eager.lstm_forward_eager(x.unsqueeze(-1), w_ih, w_hh, b_ih, b_hh)
+        return _FusedLSTMLastFn.apply(xl.contiguous(), w_ih, w_hh, b_ih, b_hh)
+    out, _, _ = eager.lstm_forward_eager(xl.unsqueeze(-1), w_ih, w_hh, b_ih, b_hh)
     return out[:, -1, :]
 
 
