@@ -263,9 +263,21 @@ def mpgcn_forward_sharded(model, x_seq_shard, G_list, group=None):
         fp8 = False
     gops = model._graph_operators(G_list)
     cd = model.compute_dtype
-    lstm_in = (
-        x_seq_shard.to(cd).permute(0, 2, 3, 1, 4).reshape(B * N * Nl, T).contiguous()
-    )
+    if x_seq_shard.is_cuda:
+        # one fused cast+transpose+pad copy at the LSTM kernel's row width
+        # (same as models/mpgcn.py forward)
+        Tpad = -(-T // 8) * 8
+        lstm_in = torch.empty(B * N * Nl, Tpad, dtype=cd,
+                              device=x_seq_shard.device)
+        lv = lstm_in.view(B, N, Nl, Tpad)
+        lv[..., :T].copy_(x_seq_shard.squeeze(-1).permute(0, 2, 3, 1))
+        if Tpad != T:
+            lv[..., T:].zero_()
+    else:
+        lstm_in = (
+            x_seq_shard.to(cd).permute(0, 2, 3, 1, 4)
+            .reshape(B * N * Nl, T).contiguous()
+        )
     outs = []
     for m in range(model.M):
         branch = model.branch_models[m]
@@ -275,6 +287,7 @@ def mpgcn_forward_sharded(model, x_seq_shard, G_list, group=None):
             branch["temporal"].weight_hh_l0.to(cd),
             branch["temporal"].bias_ih_l0,
             branch["temporal"].bias_hh_l0,
+            T=T,
         )
         X = h.reshape(B, N, Nl, model.lstm_hidden_dim)
         gop = gops[m]
