@@ -81,6 +81,11 @@ class Forecaster:
         self._graphs: dict = {}
         self._use_graph = (self.device.type == "cuda"
                            and bool(params.get("capture_graph", True)))
+        # FastAPI serves sync endpoints from a threadpool; replays share the
+        # static input/output buffers, so the captured path must serialize
+        import threading
+
+        self._graph_lock = threading.Lock()
 
     # ---- hipGraph-captured request path ------------------------------------
     # A serving forward is launch-dense (hundreds of small kernels for one
@@ -129,19 +134,21 @@ class Forecaster:
         cur = x_seq.unsqueeze(0).float().to(self.device)  # (1, T, N, N, 1)
         if self._use_graph:
             try:
-                graph, xs, ys, dow_buf = self._captured(cur)
+                with self._graph_lock:
+                    graph, xs, ys, dow_buf = self._captured(cur)
             except Exception:  # capture unsupported here — serve eager
                 self._use_graph = False
         preds = []
         if self._use_graph:
-            xs.copy_(cur)
-            dow_buf.fill_(dow % 7)
-            for _ in range(horizon):
-                graph.replay()
-                step = ys.clone()  # ys is overwritten by the next replay
-                xs.copy_(torch.cat([xs[:, 1:], step], dim=1))
-                preds.append(step)
-            return torch.cat(preds, dim=1)[0, :, :, :, 0].cpu()
+            with self._graph_lock:
+                xs.copy_(cur)
+                dow_buf.fill_(dow % 7)
+                for _ in range(horizon):
+                    graph.replay()
+                    step = ys.clone()  # ys is overwritten by the next replay
+                    xs.copy_(torch.cat([xs[:, 1:], step], dim=1))
+                    preds.append(step)
+                return torch.cat(preds, dim=1)[0, :, :, :, 0].cpu()
         g_list = [self.G_static,
                   (self.G_o[dow % 7:dow % 7 + 1], self.G_d[dow % 7:dow % 7 + 1])]
         if int(self.params.get("perspectives", 2)) == 3:
