@@ -215,3 +215,61 @@ def test_noop_context_without_env():
     out = model(x, [Gs, (Go, Gd)])
     torch.nn.functional.mse_loss(out, y).backward()
     reducer.finalize()  # must be a no-op, not raise
+
+
+def _flatadam_rank_worker(rank, world, file_name, out_file):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world), LOCAL_RANK=str(rank))
+    dist.init_process_group("gloo", init_method=f"file://{file_name}",
+                            rank=rank, world_size=world)
+    ctx = DistContext(rank=rank, world_size=world, local_rank=rank, backend="gloo")
+    from mpgcn_amd.ops.optim import FlatAdam
+
+    model = _make_model(seed=100 + rank)  # broadcast must equalize
+    # bench order: FlatAdam repoints params/grads BEFORE the reducer hooks
+    opt = FlatAdam(model.parameters(), lr=3e-3)
+    reducer = GradAllReducer(model, ctx)
+
+    x, y, Gs, Go, Gd = _make_inputs()
+    half = B // world
+    sl = slice(rank * half, (rank + 1) * half)
+    for i in range(3):
+        opt.zero_grad()
+        out = model(x[sl], [Gs, (Go[sl], Gd[sl])])
+        torch.nn.functional.mse_loss(out, y[sl]).backward()
+        reducer.finalize()
+        opt.step()
+    if rank == 0:
+        torch.save({n: p.detach().clone() for n, p in model.named_parameters()},
+                   out_file)
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_two_rank_dp_flat_adam_matches_full_batch(tmp_path):
+    """The bench DP path (FlatAdam flat grads as reducer targets): 3 steps of
+    2-rank half-batch training must land on the same weights as 1 process on
+    the full batch with the same optimizer."""
+    file_name = str(tmp_path / "pg_init_fa")
+    out_file = str(tmp_path / "rank0_weights.pt")
+    ctxm = mp.get_context("spawn")
+    procs = [ctxm.Process(target=_flatadam_rank_worker,
+                          args=(r, 2, file_name, out_file)) for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=240)
+        assert p.exitcode == 0
+    dp_weights = torch.load(out_file, weights_only=True)
+
+    from mpgcn_amd.ops.optim import FlatAdam
+
+    model = _make_model(seed=100)
+    opt = FlatAdam(model.parameters(), lr=3e-3)
+    x, y, Gs, Go, Gd = _make_inputs()
+    for i in range(3):
+        opt.zero_grad()
+        out = model(x, [Gs, (Go, Gd)])
+        torch.nn.functional.mse_loss(out, y).backward()
+        opt.step()
+    for n, p in model.named_parameters():
+        assert torch.allclose(dp_weights[n], p.detach(), atol=1e-5), n
