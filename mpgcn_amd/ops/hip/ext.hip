@@ -61,13 +61,19 @@ void fill_slot0(torch::Tensor& dst5, const torch::Tensor& src4, long S) {
 // — GT is then the REDUCED stack WITHOUT support 0 (S-1 supports), the GEMM
 // fills only slots 1..S-1 of U, and the caller copies X into slot 0
 // (ops/functional.py). Cuts 1/S of the contraction's FLOPs and bytes.
-torch::Tensor bdgcn_mode1(torch::Tensor X, torch::Tensor GT, bool id_skip) {
+torch::Tensor bdgcn_mode1(torch::Tensor X, torch::Tensor GT, bool id_skip,
+                          bool no_fill) {
     check_in(X, "X");
     check_in(GT, "GT");
     const bool dyn = GT.dim() == 4;
     const long B = X.size(0), No = X.size(1), Nd = X.size(2), C = X.size(3);
     const long Se = dyn ? GT.size(1) : GT.size(0);  // supports in the GEMM
-    const long S = id_skip ? Se + 1 : Se;           // supports in U
+    // no_fill (id_skip only): return U with ONLY the Se computed slots — no
+    // materialized identity slot. Consumers read the identity block straight
+    // from X via row_gemm_split / red_gemm_split, so the slot_copy read+write
+    // disappears from the step.
+    const bool red = id_skip && no_fill;
+    const long S = (id_skip && !red) ? Se + 1 : Se;  // slots in U
     TORCH_CHECK(GT.size(-1) == No && GT.size(-2) == No, "shape mismatch");
     TORCH_CHECK(!dyn || GT.size(0) == B, "dynamic GT batch mismatch");
     TORCH_CHECK(B * Se <= 65535, "too many instances");
@@ -76,8 +82,9 @@ torch::Tensor bdgcn_mode1(torch::Tensor X, torch::Tensor GT, bool id_skip) {
     AxisGemmParams p{};
     p.AT = GT.data_ptr();
     p.X = X.data_ptr();
-    p.OUT = id_skip ? (void*)((char*)U.data_ptr() + C * U.element_size())
-                    : U.data_ptr();
+    p.OUT = (id_skip && !red)
+                ? (void*)((char*)U.data_ptr() + C * U.element_size())
+                : U.data_ptr();
     p.M = (int)No; p.K = (int)No; p.L = (int)(Nd * C);
     p.a_div = (int)Se; p.a_bs1 = dyn ? Se * No * No : 0; p.a_bs2 = No * No;
     p.x_div = (int)Se; p.x_bs1 = No * Nd * C; p.x_bs2 = 0;
@@ -90,7 +97,7 @@ torch::Tensor bdgcn_mode1(torch::Tensor X, torch::Tensor GT, bool id_skip) {
     p.a_vec = (No % ch == 0);
     p.x_vec = ((Nd * C) % ch == 0) && (C % ch == 0);
     axis_gemm_launch(p, (int)(B * Se), is_f32(X), stream());
-    if (id_skip)  // slot 0 = identity product = X
+    if (id_skip && !red)  // slot 0 = identity product = X
         fill_slot0(U, X, S);
     return U;
 }
@@ -157,16 +164,21 @@ torch::Tensor bdgcn_mode2(torch::Tensor V, torch::Tensor A2T,
 // caller copies dY into the s = 0 rows (dV[..., 0, :] = dY, the identity
 // support's exact gradient).
 torch::Tensor bdgcn_mode2_bwd(torch::Tensor dY, torch::Tensor A2, long S,
-                              bool id_skip) {
+                              bool id_skip, bool no_fill) {
     check_in(dY, "dY");
     check_in(A2, "A2");
     const bool dyn = A2.dim() == 3;
     const long B = dY.size(0), Nm = dY.size(1), H = dY.size(3);
     const long N = A2.size(-1);
     const long Se = id_skip ? S - 1 : S;
+    // no_fill (id_skip only): dV carries the Se computed slots without the
+    // materialized identity-gradient slot (dY); split-row consumers read dY
+    // directly (see bdgcn_mode1 no_fill).
+    const bool red = id_skip && no_fill;
+    const long So = red ? Se : S;  // slots in dV
     TORCH_CHECK(A2.size(-2) == N * Se && dY.size(2) == N, "A2 shape mismatch");
     TORCH_CHECK(B <= 65535, "too many instances");
-    auto dV = torch::empty({B, Nm, N, S, H}, dY.options());
+    auto dV = torch::empty({B, Nm, N, So, H}, dY.options());
 
     // One GEMM per batch element (m folded into L, as in bdgcn_mode2):
     // dV[cs, (m,h)] = sum_d A2[cs,d] dY[b,m,d,h]
@@ -177,20 +189,20 @@ torch::Tensor bdgcn_mode2_bwd(torch::Tensor dY, torch::Tensor A2, long S,
     p.M = (int)(N * Se); p.K = (int)N; p.L = (int)(Nm * H);
     p.a_div = 1; p.a_bs1 = dyn ? N * Se * N : 0; p.a_bs2 = 0;
     p.x_div = 1; p.x_bs1 = Nm * N * H; p.x_bs2 = 0;
-    p.o_div = 1; p.o_bs1 = Nm * N * S * H; p.o_bs2 = 0;
+    p.o_div = 1; p.o_bs1 = Nm * N * So * H; p.o_bs2 = 0;
     p.kdiv = 1; p.k_lo = H;
     p.qdiv = (int)H; p.q_hi = N * H;       // q = (m, h)
     p.o_row = H;                           // out row = cs
-    if (id_skip) {
+    if (id_skip && !red) {
         // GEMM row m = (c, s') lands in dV slot (c, s'+1)
         p.o_mdiv = (int)Se; p.o_m_hi = S * H; p.o_m_lo = H; p.o_m_base = H;
     }
-    p.ogdiv = (int)H; p.og_hi = N * S * H;
+    p.ogdiv = (int)H; p.og_hi = N * So * H;
     const int ch = chunk_elems(dY);
     p.a_vec = (N % ch == 0);
     p.x_vec = (H % ch == 0);
     axis_gemm_launch(p, (int)B, is_f32(dY), stream());
-    if (id_skip)  // identity-support gradient rows are dY itself
+    if (id_skip && !red)  // identity-support gradient rows are dY itself
         fill_slot0(dV, dY, S);
     return dV;
 }
@@ -284,6 +296,82 @@ void row_gemm_out(torch::Tensor X, torch::Tensor W,
     p.relu = relu ? 1 : 0;
     p.x_vec = (K % chunk_elems(X) == 0);
     row_gemm_launch(p, is_f32(X), stream());
+}
+
+// Split-row row_gemm: logical X row r = [XA[r, :kA] | XB[r, :kB]] — the
+// identity-support block is consumed straight from its source tensor (X or
+// dY) instead of a materialized slot-0 copy (slot_copy traffic removed).
+// Both part widths must be vector-chunk-aligned so fragments never straddle
+// the seam (row_gemm.hip split addressing).
+torch::Tensor row_gemm_split(torch::Tensor XA, torch::Tensor XB,
+                             torch::Tensor W,
+                             c10::optional<torch::Tensor> bias, bool relu) {
+    check_in(XA, "XA");
+    check_in(XB, "XB");
+    check_in(W, "W");
+    const long R = XA.size(0), kA = XA.size(1), kB = XB.size(1);
+    const long K = kA + kB, N = W.size(1);
+    TORCH_CHECK(XB.size(0) == R, "row mismatch");
+    TORCH_CHECK(XA.scalar_type() == XB.scalar_type(), "dtype mismatch");
+    TORCH_CHECK(W.size(0) == K, "W shape mismatch");
+    TORCH_CHECK(N <= 128, "row_gemm: N must be <= 128");
+    TORCH_CHECK(K <= 2048, "row_gemm: K too large for LDS staging");
+    const int ch = chunk_elems(XA);
+    TORCH_CHECK(kA % ch == 0 && kB % ch == 0,
+                "row_gemm_split: part widths must be chunk-aligned");
+    auto OUT = torch::empty({R, N}, XA.options());
+    RowGemmParams p{};
+    p.X = XB.data_ptr();
+    p.X2 = XA.data_ptr();
+    p.k0 = (int)kA;
+    p.W = W.data_ptr();
+    p.OUT = OUT.data_ptr();
+    p.bias = bias_ptr(bias);
+    p.R = R; p.K = (int)K; p.N = (int)N;
+    p.o_row = N; p.o_off = 0;
+    p.relu = relu ? 1 : 0;
+    p.x_vec = 1;
+    row_gemm_launch(p, is_f32(XA), stream());
+    return OUT;
+}
+
+// Split-row red_gemm for the projection weight gradient: logical rows
+// X = [XA | XB] (dY | reduced dV) and Y = [YA | YB] (X | reduced U); both
+// identity blocks read from their source tensors, no slot-0 fills.
+torch::Tensor red_gemm_split(torch::Tensor XA, torch::Tensor XB,
+                             torch::Tensor YA, torch::Tensor YB) {
+    check_in(XA, "XA");
+    check_in(XB, "XB");
+    check_in(YA, "YA");
+    check_in(YB, "YB");
+    const long R = XA.size(0);
+    const long xkA = XA.size(1), xkB = XB.size(1);
+    const long ykA = YA.size(1), ykB = YB.size(1);
+    const long K = xkA + xkB, N = ykA + ykB;
+    TORCH_CHECK(XB.size(0) == R && YA.size(0) == R && YB.size(0) == R,
+                "row mismatch");
+    const int ch = chunk_elems(XA);
+    TORCH_CHECK(xkA % ch == 0 && xkB % ch == 0 && ykA % ch == 0 &&
+                    ykB % ch == 0,
+                "red_gemm_split: part widths must be chunk-aligned");
+    auto f32 = XA.options().dtype(torch::kFloat);
+    const bool det = at::globalContext().deterministicAlgorithms();
+    const long nb = det ? red_gemm_nblocks(R) : 1;
+    auto out = det ? torch::empty({nb, K, N}, f32) : torch::zeros({K, N}, f32);
+    RedGemmParams p{};
+    p.X = XB.data_ptr();
+    p.x2 = XA.data_ptr();
+    p.x_k0 = (int)xkA;
+    p.Y = YB.data_ptr();
+    p.y2 = YA.data_ptr();
+    p.y_k0 = (int)ykA;
+    p.out = out.data_ptr<float>();
+    p.R = R; p.K = (int)K; p.N = (int)N;
+    p.det = det ? 1 : 0;
+    p.x_vec = 1;
+    p.y_vec = 1;
+    red_gemm_launch(p, is_f32(XA), stream());
+    return det ? out.sum(0) : out;
 }
 
 // Reduction GEMM: out = X^T @ Y (f32), plus optional colsum(X) and
@@ -1064,16 +1152,22 @@ std::vector<torch::Tensor> relu_bwd_colsum(torch::Tensor dH, torch::Tensor Y,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("bdgcn_mode1", &bdgcn_mode1, "origin-axis graph product (K1)",
-          py::arg("X"), py::arg("GT"), py::arg("id_skip") = false);
+          py::arg("X"), py::arg("GT"), py::arg("id_skip") = false,
+          py::arg("no_fill") = false);
     m.def("bdgcn_mode2", &bdgcn_mode2, "dest-axis graph product + bias + act (K2)",
           py::arg("V"), py::arg("A2T"), py::arg("bias"), py::arg("relu"),
           py::arg("N"), py::arg("S"), py::arg("id_skip") = false);
     m.def("bdgcn_mode2_bwd", &bdgcn_mode2_bwd, "backward dV of mode2",
-          py::arg("dY"), py::arg("A2"), py::arg("S"), py::arg("id_skip") = false);
+          py::arg("dY"), py::arg("A2"), py::arg("S"), py::arg("id_skip") = false,
+          py::arg("no_fill") = false);
     m.def("bdgcn_mode1_bwd", &bdgcn_mode1_bwd, "backward dX of mode1",
           py::arg("dU"), py::arg("A3T"), py::arg("id_skip") = false);
     m.def("row_gemm", &row_gemm, "fused row GEMM + bias + act (K3/K6)");
+    m.def("row_gemm_split", &row_gemm_split,
+          "row GEMM over split-source rows [XA | XB] (identity-slot-free)");
     m.def("red_gemm", &red_gemm, "fused reduction GEMM X^T@Y + colsum + xdot");
+    m.def("red_gemm_split", &red_gemm_split,
+          "reduction GEMM over split rows [XA|XB]^T @ [YA|YB]");
     m.def("relu_bwd_colsum", &relu_bwd_colsum, "fused ReLU bwd mask + bias colsum");
     m.def("bdgcn_mode2_fp8", &bdgcn_mode2_fp8, "fp8 e4m3 mode-2 probe (measurement only)");
     m.def("bdgcn_mode1_fp8", &bdgcn_mode1_fp8, "fp8 e4m3 mode-1 probe (measurement only)");

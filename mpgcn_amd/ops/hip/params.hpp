@@ -36,6 +36,11 @@ struct AxisGemmParams {
 
 struct RowGemmParams {
     const void* X;
+    const void* X2;     // optional identity-block source: row k < k0 elements
+                        // come from X2 (row stride k0), k >= k0 from X (row
+                        // stride K - k0). Avoids materializing the identity
+                        // support slot (slot_copy) just to re-read it here.
+    int k0;
     const void* W;
     void* OUT;          // bf16/f32/fp8 primary output; nullable when OUT8 set
     void* OUT8;         // optional fp8 e4m3 output = fp8(v * *q_scale); nullable
@@ -78,6 +83,11 @@ struct LstmBwdParams {
 
 struct RedGemmParams {
     const void* X;      // (R, K) row-major T
+    const void* x2;     // optional identity-block source for X rows: k < x_k0
+                        // from x2 (row stride x_k0), else X (stride K - x_k0)
+    int x_k0;
+    const void* y2;     // same for Y rows at y_k0
+    int y_k0;
     const void* Y;      // (R, N) row-major T
     const void* xvec;   // per-row scalar T at xvec[r*xv_stride + xv_off]; nullable
     long xv_stride, xv_off;

@@ -57,11 +57,20 @@ __launch_bounds__(256) __global__ void red_gemm_kernel(RedGemmParams p) {
             alignas(16) T tmp[CH];
             for (int i = 0; i < CH; ++i) tmp[i] = (T)0.f;
             if (row < p.R) {
+                // split-source rows: [x2 row (x_k0) | X row (K - x_k0)];
+                // x_k0 is CH-aligned (binding contract), so a chunk never
+                // straddles the boundary
+                const T* xs = X;
+                long xb = row * (long)p.K + k0;
+                if (p.x2) {
+                    if (k0 < p.x_k0) { xs = (const T*)p.x2; xb = row * (long)p.x_k0 + k0; }
+                    else xb = row * (long)(p.K - p.x_k0) + (k0 - p.x_k0);
+                }
                 if (p.x_vec && k0 + CH <= p.K)
-                    *(Chunk16*)tmp = *(const Chunk16*)&X[row * p.K + k0];
+                    *(Chunk16*)tmp = *(const Chunk16*)&xs[xb];
                 else
                     for (int i = 0; i < CH; ++i)
-                        if (k0 + i < p.K) tmp[i] = X[row * p.K + k0 + i];
+                        if (k0 + i < p.K) tmp[i] = xs[xb + i];
             }
             // lane-rotated write order: consecutive lanes write different
             // LDS rows per instruction (banks 4*{0..7}, no 16-way conflict)
@@ -101,11 +110,18 @@ __launch_bounds__(256) __global__ void red_gemm_kernel(RedGemmParams p) {
                                 tmp[i] = (T)__builtin_amdgcn_cvt_pk_f32_fp8(b, false)[0];
                             }
                     }
-                } else if (p.y_vec && n0 + CH <= p.N) {
-                    *(Chunk16*)tmp = *(const Chunk16*)&Y[row * p.N + n0];
                 } else {
-                    for (int i = 0; i < CH; ++i)
-                        if (n0 + i < p.N) tmp[i] = Y[row * p.N + n0 + i];
+                    const T* ys = Y;
+                    long yb = row * (long)p.N + n0;
+                    if (p.y2) {  // [y2 row (y_k0) | Y row (N - y_k0)]
+                        if (n0 < p.y_k0) { ys = (const T*)p.y2; yb = row * (long)p.y_k0 + n0; }
+                        else yb = row * (long)(p.N - p.y_k0) + (n0 - p.y_k0);
+                    }
+                    if (p.y_vec && n0 + CH <= p.N)
+                        *(Chunk16*)tmp = *(const Chunk16*)&ys[yb];
+                    else
+                        for (int i = 0; i < CH; ++i)
+                            if (n0 + i < p.N) tmp[i] = ys[yb + i];
                 }
             }
 #pragma unroll

@@ -21,6 +21,8 @@ __launch_bounds__(256) __global__ void row_gemm_kernel(RowGemmParams p, int nf16
     T* ldsWT = (T*)smem;  // [nf16*16][K + PAD]
 
     const T* __restrict__ X = (const T*)p.X;
+    const T* __restrict__ X2 = (const T*)p.X2;  // identity block (k < k0)
+    const int k0 = p.k0, kb = p.K - p.k0;       // part widths (X2 | X)
     const T* __restrict__ W = (const T*)p.W;
     T* __restrict__ O = (T*)p.OUT;
     unsigned char* __restrict__ O8 = (unsigned char*)p.OUT8;
@@ -53,13 +55,20 @@ __launch_bounds__(256) __global__ void row_gemm_kernel(RowGemmParams p, int nf16
         for (int kf = 0; kf < kfrags; ++kf) {
             const int k = kf * MT::MFMA_K + kgrp * MT::FRAG_ELEMS;
             typename MT::frag_t af;
+            // split-source rows: [X2 row (k0) | X row (K - k0)]. Callers
+            // guarantee k0 % FRAG_ELEMS == 0 when X2 is set, so a fragment
+            // never straddles the boundary on the vector path.
+            const T* src = (X2 && k < k0) ? X2 : X;
+            const long base = (X2 && k < k0) ? row * (long)k0 + k
+                              : (X2 ? row * (long)kb + (k - k0)
+                                    : row * (long)p.K + k);
             if (row_ok && k + MT::FRAG_ELEMS <= p.K && p.x_vec) {
-                af = *(const typename MT::frag_t*)&X[row * p.K + k];
+                af = *(const typename MT::frag_t*)&src[base];
             } else {
                 alignas(16) T tmp[MT::FRAG_ELEMS];
 #pragma unroll
                 for (int i = 0; i < MT::FRAG_ELEMS; ++i)
-                    tmp[i] = (row_ok && k + i < p.K) ? X[row * p.K + k + i] : (T)0.f;
+                    tmp[i] = (row_ok && k + i < p.K) ? src[base + i] : (T)0.f;
                 af = *(const typename MT::frag_t*)tmp;
             }
             for (int nf = 0; nf < nf16; ++nf) {
