@@ -387,7 +387,10 @@ class _BDGCNLayerFp8Fn(torch.autograd.Function):
             V8.view(B, N, N * S, Hdim), gop.A2T8, bias_f32, relu, N, S,
             emit_twin, gop.id_first,
         )
-        ctx.save_for_backward(U8, Wre, Y)
+        # X saved (and nofill pinned off) so the MPGCN_FP8_BWD=0 bisect path
+        # can delegate to _BDGCNLayerFn.backward, which unpacks (X, U, Wre, Y)
+        ctx.save_for_backward(X, U8, Wre, Y)
+        ctx.nofill = False
         ctx.gop = gop
         ctx.relu = relu
         ctx.has_bias = bias is not None
@@ -420,7 +423,7 @@ class _BDGCNLayerFp8Fn(torch.autograd.Function):
         #        directly (its sole consumer is the fp8 dX contraction).
         # Weight-gradient reductions (red_gemm) stay bf16-accumulated-f32.
         ext = _ops.get_ext()
-        U8, Wre, Y = ctx.saved_tensors
+        _X, U8, Wre, Y = ctx.saved_tensors
         gop: GraphOperator = ctx.gop
         st = ctx.fp8_state
         B, N, S, C, Hdim = ctx.dims

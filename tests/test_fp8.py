@@ -222,3 +222,31 @@ def test_fp8_shape_gate_raises():
         MPGCN(M=2, K=S, input_dim=1, lstm_hidden_dim=H, lstm_num_layers=1,
               gcn_hidden_dim=H, gcn_num_layers=3, num_nodes=N,
               compute_dtype=torch.float32, fp8_forward=True)
+
+
+def test_fp8_bisect_bwd_delegates_to_bf16_backward(monkeypatch):
+    # MPGCN_FP8_BWD=0 routes the fp8 layer's backward through
+    # _BDGCNLayerFn.backward (bf16 contractions over the saved fp8 U8);
+    # pin the ctx contract between the two Functions (saved-tensor tuple
+    # and the nofill flag)
+    monkeypatch.setenv("MPGCN_FP8_BWD", "0")
+    X, Go, Gd, W, b = _layer_inputs()
+    gop = GraphOperator(Go, Gd)
+    Xn = X.clone().requires_grad_(True)
+    Wn = W.clone().requires_grad_(True)
+    bn = b.clone().requires_grad_(True)
+    Y, _ = bdgcn_layer_fp8(Xn, Wn, bn, gop, relu=True)
+    Y.square().sum().backward()
+    Xe = X.float().detach().requires_grad_(True)
+    We = W.float().detach().requires_grad_(True)
+    be = b.float().detach().requires_grad_(True)
+    ref = eager.bdgcn_layer_eager(Xe, Go.float(), Gd.float(), We, be, "relu")
+    ref.square().sum().backward()
+    for g, ge, name in ((Xn.grad.float(), Xe.grad, "dX"),
+                        (Wn.grad.float(), We.grad, "dW"),
+                        (bn.grad.float(), be.grad, "db")):
+        assert torch.isfinite(g).all(), name
+        cos = torch.nn.functional.cosine_similarity(
+            g.flatten(), ge.flatten(), dim=0
+        ).item()
+        assert cos > 0.98, f"{name} cosine {cos:.4f}"
