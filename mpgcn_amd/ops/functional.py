@@ -368,6 +368,8 @@ class _BDGCNLayerFp8Fn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, X, W, bias, gop: GraphOperator, relu: bool, X8,
                 emit_twin: bool, fp8_state: dict):
+        import os
+
         ext = _ops.get_ext()
         B, N = X.shape[0], X.shape[1]
         C = X.shape[-1]
@@ -375,22 +377,40 @@ class _BDGCNLayerFp8Fn(torch.autograd.Function):
         Hdim = W.shape[1]
         if X8 is None:
             X8 = X.to(torch.float8_e4m3fn)
+        # identity-slot-free fp8 schedule (see _BDGCNLayerFn): U8/dV carry no
+        # materialized identity slot; consumers read X8/dY directly via the
+        # split-row kernels. Disabled under MPGCN_FP8_BWD=0 so the bisect
+        # path can delegate to _BDGCNLayerFn.backward's fill layout.
+        nofill = (gop.id_first and C % 16 == 0 and Hdim % 8 == 0
+                  and S * Hdim <= _ROW_GEMM_MAX_N and S * C <= _ROW_GEMM_MAX_N
+                  and os.environ.get("MPGCN_FP8_BWD", "1") != "0")
         # U stays fp8-ONLY: mode-1 writes half the bf16 path's output bytes
         # (U is the step's largest tensor), and backward's dW reduction reads
         # the fp8 U directly (red_gemm y_fp8 staging)
-        U8 = ext.bdgcn_mode1_fp8_train(X8, gop.GoT8, gop.id_first)
+        U8 = ext.bdgcn_mode1_fp8_train(X8, gop.GoT8, gop.id_first, nofill)
         Wre = eager.reorder_projection_weight(W, S, C).contiguous()
         Wre8 = Wre.to(torch.float8_e4m3fn)
-        V8 = ext.row_gemm_fp8(U8.reshape(B * N * N, S * C), Wre8)
+        R = B * N * N
+        if nofill:  # U8: (B,N,N,S-1,C); logical rows are [X8 | U8]
+            V8 = ext.row_gemm_fp8_split(X8.reshape(R, C),
+                                        U8.reshape(R, (S - 1) * C), Wre8)
+        else:
+            V8 = ext.row_gemm_fp8(U8.reshape(R, S * C), Wre8)
         bias_f32 = bias.float().contiguous() if bias is not None else None
         Y, Y8 = ext.bdgcn_mode2_fp8_train(
             V8.view(B, N, N * S, Hdim), gop.A2T8, bias_f32, relu, N, S,
             emit_twin, gop.id_first,
         )
-        # X saved (and nofill pinned off) so the MPGCN_FP8_BWD=0 bisect path
-        # can delegate to _BDGCNLayerFn.backward, which unpacks (X, U, Wre, Y)
-        ctx.save_for_backward(X, U8, Wre, Y)
+        # X saved first (and ctx.nofill pinned False) so the MPGCN_FP8_BWD=0
+        # bisect path can delegate to _BDGCNLayerFn.backward, which unpacks
+        # (X, U, Wre, Y); nofill mode additionally saves X8 for the dW split
+        # (bisect is env-gated off there, so the layouts never mix)
+        if nofill:
+            ctx.save_for_backward(X, X8, U8, Wre, Y)
+        else:
+            ctx.save_for_backward(X, U8, Wre, Y)
         ctx.nofill = False
+        ctx.nofill_fp8 = nofill
         ctx.gop = gop
         ctx.relu = relu
         ctx.has_bias = bias is not None
@@ -405,9 +425,11 @@ class _BDGCNLayerFp8Fn(torch.autograd.Function):
     def backward(ctx, dH, _dY8=None):
         import os
 
-        if os.environ.get("MPGCN_FP8_BWD", "1") == "0":
+        if os.environ.get("MPGCN_FP8_BWD", "1") == "0" and not ctx.nofill_fp8:
             # debug/bisect: fp8 forward with the full bf16 backward (red_gemm
-            # reads the saved U8 directly either way)
+            # reads the saved U8 directly either way). Forward disables the
+            # identity-slot-free layout under this env, so the ctx layouts
+            # match; a mid-run env flip keeps the normal fp8 backward instead.
             dX, dW, db, _, _ = _BDGCNLayerFn.backward(ctx, dH)
             return dX, dW, db, None, None, None, None, None
         # Scaled-fp8 gradient contractions: gradients live well below e4m3's
@@ -423,7 +445,10 @@ class _BDGCNLayerFp8Fn(torch.autograd.Function):
         #        directly (its sole consumer is the fp8 dX contraction).
         # Weight-gradient reductions (red_gemm) stay bf16-accumulated-f32.
         ext = _ops.get_ext()
-        _X, U8, Wre, Y = ctx.saved_tensors
+        if ctx.nofill_fp8:
+            _X, X8s, U8, Wre, Y = ctx.saved_tensors
+        else:
+            _X, U8, Wre, Y = ctx.saved_tensors
         gop: GraphOperator = ctx.gop
         st = ctx.fp8_state
         B, N, S, C, Hdim = ctx.dims
@@ -435,15 +460,30 @@ class _BDGCNLayerFp8Fn(torch.autograd.Function):
         if not ctx.has_bias:
             dbias = None
 
-        dV = ext.bdgcn_mode2_bwd_fp8(dY8, gop.A28, S, st["inv_y"], dY, gop.id_first)
         R = B * N * N
-        dVflat = dV.reshape(R, S * Hdim)
-        dW, join_dw = _dw_overlapped(ext, dVflat, U8.reshape(R, S * C),
-                                     dH.dtype, S, C, Hdim)
-
+        WreT = Wre.t().contiguous()
         ext.fp8_scale_update(st["amax_u"], st["scale_u"], st["inv_u"], _FP8_MARGIN)
-        dU8 = ext.row_gemm_fp8_out(dVflat, Wre.t().contiguous(),
-                                   st["scale_u"], st["amax_u"])
+        if ctx.nofill_fp8:
+            # identity-slot-free: reduced dV; logical rows [dY | dV] / the
+            # dW reduction reads [X8 | U8] — numerics identical to the fill
+            # layout (the fills wrote exactly dY and X8)
+            dV = ext.bdgcn_mode2_bwd_fp8(dY8, gop.A28, S, st["inv_y"], dY,
+                                         True, True)  # (B,N,N,S-1,H)
+            dYf = dY.reshape(R, Hdim)
+            dVred = dV.reshape(R, (S - 1) * Hdim)
+            dW, join_dw = _dw_overlapped_split(
+                ext, dYf, dVred, X8s.reshape(R, C),
+                U8.reshape(R, (S - 1) * C), dH.dtype, S, C, Hdim)
+            dU8 = ext.row_gemm_fp8_out_split(dYf, dVred, WreT,
+                                             st["scale_u"], st["amax_u"])
+        else:
+            dV = ext.bdgcn_mode2_bwd_fp8(dY8, gop.A28, S, st["inv_y"], dY,
+                                         gop.id_first)
+            dVflat = dV.reshape(R, S * Hdim)
+            dW, join_dw = _dw_overlapped(ext, dVflat, U8.reshape(R, S * C),
+                                         dH.dtype, S, C, Hdim)
+            dU8 = ext.row_gemm_fp8_out(dVflat, WreT,
+                                       st["scale_u"], st["amax_u"])
         dX = ext.bdgcn_mode1_bwd_fp8(dU8.view(B, N, N, S, C), gop.A3T8,
                                      st["inv_u"], gop.id_first)
         join_dw()

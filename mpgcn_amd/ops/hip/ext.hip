@@ -342,8 +342,20 @@ torch::Tensor red_gemm_split(torch::Tensor XA, torch::Tensor XB,
                              torch::Tensor YA, torch::Tensor YB) {
     check_in(XA, "XA");
     check_in(XB, "XB");
-    check_in(YA, "YA");
-    check_in(YB, "YB");
+    // fp8 Y parts (fp8-forward mode: YA = X8, YB = reduced U8) route through
+    // the YF8 staging with the same split addressing
+    const bool y8 = YA.scalar_type() == torch::kFloat8_e4m3fn;
+    if (y8) {
+        TORCH_CHECK(YB.scalar_type() == torch::kFloat8_e4m3fn &&
+                        YA.is_cuda() && YA.is_contiguous() &&
+                        YB.is_cuda() && YB.is_contiguous(),
+                    "red_gemm_split: fp8 Y parts must both be contiguous fp8");
+        TORCH_CHECK(XA.scalar_type() == torch::kBFloat16,
+                    "fp8 Y operand requires bf16 X");
+    } else {
+        check_in(YA, "YA");
+        check_in(YB, "YB");
+    }
     const long R = XA.size(0);
     const long xkA = XA.size(1), xkB = XB.size(1);
     const long ykA = YA.size(1), ykB = YB.size(1);
@@ -368,6 +380,7 @@ torch::Tensor red_gemm_split(torch::Tensor XA, torch::Tensor XB,
     p.out = out.data_ptr<float>();
     p.R = R; p.K = (int)K; p.N = (int)N;
     p.det = det ? 1 : 0;
+    p.y_fp8 = y8 ? 1 : 0;
     p.x_vec = 1;
     p.y_vec = 1;
     red_gemm_launch(p, is_f32(XA), stream());
@@ -671,13 +684,16 @@ void check_fp8(const torch::Tensor& t, const char* name) {
 // a multiple of 16 (flagship and large-N configs satisfy this; the Python
 // layer falls back to bf16 otherwise).
 torch::Tensor bdgcn_mode1_fp8_train(torch::Tensor X8, torch::Tensor GT8,
-                                    bool id_skip) {
+                                    bool id_skip, bool no_fill) {
     check_fp8(X8, "X8");
     check_fp8(GT8, "GT8");
     const bool dyn = GT8.dim() == 4;
     const long B = X8.size(0), No = X8.size(1), Nd = X8.size(2), C = X8.size(3);
     const long Se = dyn ? GT8.size(1) : GT8.size(0);
-    const long S = id_skip ? Se + 1 : Se;
+    // no_fill: identity-slot-free U8 (see bdgcn_mode1) — consumers read the
+    // identity block straight from X8 via the split-row fp8 kernels
+    const bool red = id_skip && no_fill;
+    const long S = (id_skip && !red) ? Se + 1 : Se;
     TORCH_CHECK(GT8.size(-1) == No && GT8.size(-2) == No, "shape mismatch");
     TORCH_CHECK(!dyn || GT8.size(0) == B, "dynamic GT8 batch mismatch");
     TORCH_CHECK(B * Se <= 65535, "too many instances");
@@ -688,7 +704,8 @@ torch::Tensor bdgcn_mode1_fp8_train(torch::Tensor X8, torch::Tensor GT8,
     AxisGemmParams p{};
     p.AT = GT8.data_ptr();
     p.X = X8.data_ptr();
-    p.OUT = id_skip ? (void*)((char*)U8.data_ptr() + C) : U8.data_ptr();
+    p.OUT = (id_skip && !red) ? (void*)((char*)U8.data_ptr() + C)
+                              : U8.data_ptr();
     p.M = (int)No; p.K = (int)No; p.L = (int)(Nd * C);
     p.a_div = (int)Se; p.a_bs1 = dyn ? Se * No * No : 0; p.a_bs2 = No * No;
     p.x_div = (int)Se; p.x_bs1 = No * Nd * C; p.x_bs2 = 0;
@@ -700,7 +717,7 @@ torch::Tensor bdgcn_mode1_fp8_train(torch::Tensor X8, torch::Tensor GT8,
     p.a_vec = (No % 16 == 0);
     p.x_vec = 1;
     axis_gemm_fp8_launch(p, (int)(B * Se), 0, stream());
-    if (id_skip)
+    if (id_skip && !red)
         fill_slot0(U8, X8, S);
     return U8;
 }
@@ -766,7 +783,8 @@ std::vector<torch::Tensor> bdgcn_mode2_fp8_train(torch::Tensor V8,
 // scale. Same contraction as bdgcn_mode2_bwd (bf16 twin of this path).
 torch::Tensor bdgcn_mode2_bwd_fp8(torch::Tensor dY8, torch::Tensor A28,
                                   long S, torch::Tensor inv_scale,
-                                  torch::Tensor dY_bf16, bool id_skip) {
+                                  torch::Tensor dY_bf16, bool id_skip,
+                                  bool no_fill) {
     check_fp8(dY8, "dY8");
     check_fp8(A28, "A28");
     TORCH_CHECK(inv_scale.is_cuda() && inv_scale.scalar_type() == torch::kFloat,
@@ -775,10 +793,14 @@ torch::Tensor bdgcn_mode2_bwd_fp8(torch::Tensor dY8, torch::Tensor A28,
     const long B = dY8.size(0), Nm = dY8.size(1), H = dY8.size(3);
     const long N = A28.size(-1);
     const long Se = id_skip ? S - 1 : S;
+    // no_fill: reduced dV, no materialized identity-gradient slot — the
+    // split-row consumers read the exact bf16 dY directly (same numerics)
+    const bool red = id_skip && no_fill;
+    const long So = red ? Se : S;
     TORCH_CHECK(A28.size(-2) == N * Se && dY8.size(2) == N, "A28 shape");
     TORCH_CHECK(B <= 65535, "too many instances");
     TORCH_CHECK((Nm * H) % 256 == 0 && H % 16 == 0, "fp8 bwd shape gate");
-    auto dV = torch::empty({B, Nm, N, S, H},
+    auto dV = torch::empty({B, Nm, N, So, H},
                            dY8.options().dtype(torch::kBFloat16));
     AxisGemmParams p{};
     p.AT = A28.data_ptr();
@@ -788,18 +810,18 @@ torch::Tensor bdgcn_mode2_bwd_fp8(torch::Tensor dY8, torch::Tensor A28,
     p.M = (int)(N * Se); p.K = (int)N; p.L = (int)(Nm * H);
     p.a_div = 1; p.a_bs1 = dyn ? N * Se * N : 0; p.a_bs2 = 0;
     p.x_div = 1; p.x_bs1 = Nm * N * H; p.x_bs2 = 0;
-    p.o_div = 1; p.o_bs1 = Nm * N * S * H; p.o_bs2 = 0;
+    p.o_div = 1; p.o_bs1 = Nm * N * So * H; p.o_bs2 = 0;
     p.kdiv = 1; p.k_lo = H;
     p.qdiv = (int)H; p.q_hi = N * H;
     p.o_row = H;
-    if (id_skip) {
+    if (id_skip && !red) {
         p.o_mdiv = (int)Se; p.o_m_hi = S * H; p.o_m_lo = H; p.o_m_base = H;
     }
-    p.ogdiv = (int)H; p.og_hi = N * S * H;
+    p.ogdiv = (int)H; p.og_hi = N * So * H;
     p.a_vec = (N % 16 == 0);
     p.x_vec = 1;
     axis_gemm_fp8_launch(p, (int)B, 2, stream());
-    if (id_skip)  // identity-support gradient rows: exact bf16 dY, unquantized
+    if (id_skip && !red)  // identity-support gradient rows: exact bf16 dY
         fill_slot0(dV, dY_bf16, S);
     return dV;
 }
@@ -859,6 +881,36 @@ torch::Tensor row_gemm_fp8(torch::Tensor X8, torch::Tensor W8) {
     auto OUT = torch::empty({R, N}, X8.options());
     RowGemmParams p{};
     p.X = X8.data_ptr();
+    p.W = W8.data_ptr();
+    p.OUT = OUT.data_ptr();
+    p.bias = nullptr;
+    p.R = R; p.K = (int)K; p.N = (int)N;
+    p.o_row = N; p.o_off = 0;
+    p.relu = 0;
+    p.x_vec = 1;
+    row_gemm_fp8_launch(p, stream());
+    return OUT;
+}
+
+// Split-row fp8 projection: logical rows [XA8 | XB8] — the identity block
+// (X8) streams from its source, no fp8 slot-0 fill (see row_gemm_split).
+torch::Tensor row_gemm_fp8_split(torch::Tensor XA8, torch::Tensor XB8,
+                                 torch::Tensor W8) {
+    check_fp8(XA8, "XA8");
+    check_fp8(XB8, "XB8");
+    check_fp8(W8, "W8");
+    const long R = XA8.size(0), kA = XA8.size(1), kB = XB8.size(1);
+    const long K = kA + kB, N = W8.size(1);
+    TORCH_CHECK(XB8.size(0) == R, "row mismatch");
+    TORCH_CHECK(W8.size(0) == K, "W8 shape mismatch");
+    TORCH_CHECK(N <= 128, "row_gemm_fp8: N must be <= 128");
+    TORCH_CHECK(kA % 16 == 0 && kB % 16 == 0 && K <= 2048,
+                "row_gemm_fp8_split: part widths must be 16-aligned");
+    auto OUT = torch::empty({R, N}, XA8.options());
+    RowGemmParams p{};
+    p.X = XB8.data_ptr();
+    p.X2 = XA8.data_ptr();
+    p.k0 = (int)kA;
     p.W = W8.data_ptr();
     p.OUT = OUT.data_ptr();
     p.bias = nullptr;
@@ -936,6 +988,44 @@ torch::Tensor row_gemm_fp8_out(torch::Tensor X, torch::Tensor W,
     p.o_row = N; p.o_off = 0;
     p.relu = 0;
     p.x_vec = (K % chunk_elems(X) == 0);
+    row_gemm_launch(p, 0, stream());
+    return OUT8;
+}
+
+// Split-row variant of row_gemm_fp8_out: logical bf16 rows [XA | XB]
+// (dY | reduced dV) -> scaled fp8 dU8, identity-gradient block read from dY.
+torch::Tensor row_gemm_fp8_out_split(torch::Tensor XA, torch::Tensor XB,
+                                     torch::Tensor W, torch::Tensor q_scale,
+                                     torch::Tensor amax_out) {
+    check_in(XA, "XA");
+    check_in(XB, "XB");
+    check_in(W, "W");
+    TORCH_CHECK(XA.scalar_type() == torch::kBFloat16 &&
+                    XB.scalar_type() == torch::kBFloat16, "bf16 only");
+    TORCH_CHECK(q_scale.is_cuda() && q_scale.scalar_type() == torch::kFloat);
+    TORCH_CHECK(amax_out.is_cuda() && amax_out.scalar_type() == torch::kFloat);
+    const long R = XA.size(0), kA = XA.size(1), kB = XB.size(1);
+    const long K = kA + kB, N = W.size(1);
+    TORCH_CHECK(XB.size(0) == R, "row mismatch");
+    TORCH_CHECK(W.size(0) == K, "W shape mismatch");
+    TORCH_CHECK(N <= 128 && K <= 2048, "row_gemm shape gate");
+    const int ch = chunk_elems(XA);
+    TORCH_CHECK(kA % ch == 0 && kB % ch == 0,
+                "row_gemm_fp8_out_split: part widths must be chunk-aligned");
+    auto OUT8 = torch::empty({R, N}, XA.options().dtype(torch::kFloat8_e4m3fn));
+    RowGemmParams p{};
+    p.X = XB.data_ptr();
+    p.X2 = XA.data_ptr();
+    p.k0 = (int)kA;
+    p.W = W.data_ptr();
+    p.OUT = nullptr;
+    p.OUT8 = OUT8.data_ptr();
+    p.q_scale = q_scale.data_ptr<float>();
+    p.amax_out = amax_out.data_ptr<float>();
+    p.R = R; p.K = (int)K; p.N = (int)N;
+    p.o_row = N; p.o_off = 0;
+    p.relu = 0;
+    p.x_vec = 1;
     row_gemm_launch(p, 0, stream());
     return OUT8;
 }
@@ -1172,20 +1262,29 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("bdgcn_mode2_fp8", &bdgcn_mode2_fp8, "fp8 e4m3 mode-2 probe (measurement only)");
     m.def("bdgcn_mode1_fp8", &bdgcn_mode1_fp8, "fp8 e4m3 mode-1 probe (measurement only)");
     m.def("bdgcn_mode1_fp8_train", &bdgcn_mode1_fp8_train,
-          "fp8-forward mode-1 with bf16 backward twin");
+          "fp8-forward mode-1 with bf16 backward twin",
+          py::arg("X8"), py::arg("GT8"), py::arg("id_skip") = false,
+          py::arg("no_fill") = false);
     m.def("bdgcn_mode2_fp8_train", &bdgcn_mode2_fp8_train,
           "fp8-forward mode-2 + bias + act, bf16 out + fp8 twin");
     m.def("row_gemm_fp8", &row_gemm_fp8, "fp8 projection GEMM");
+    m.def("row_gemm_fp8_split", &row_gemm_fp8_split,
+          "fp8 projection GEMM over split rows [XA8 | XB8]");
     m.def("rwd_supports", &rwd_supports, "fused random-walk-diffusion support build (K8)");
     m.def("dual_rwd_supports", &dual_rwd_supports, "fused dual-RWD support build");
     m.def("cheby_supports", &cheby_supports, "fused Chebyshev support build");
     m.def("localpool_supports", &localpool_supports, "fused localpool support build");
     m.def("bdgcn_mode2_bwd_fp8", &bdgcn_mode2_bwd_fp8,
-          "scaled fp8 gradient contraction dV");
+          "scaled fp8 gradient contraction dV",
+          py::arg("dY8"), py::arg("A28"), py::arg("S"), py::arg("inv_scale"),
+          py::arg("dY_bf16"), py::arg("id_skip") = false,
+          py::arg("no_fill") = false);
     m.def("bdgcn_mode1_bwd_fp8", &bdgcn_mode1_bwd_fp8,
           "scaled fp8 gradient contraction dX");
     m.def("relu_bwd_colsum_fp8", &relu_bwd_colsum_fp8,
           "ReLU bwd + colsum + fused scaled fp8 dY8");
+    m.def("row_gemm_fp8_out_split", &row_gemm_fp8_out_split,
+          "split-row row GEMM emitting scaled fp8 only");
     m.def("row_gemm_fp8_out", &row_gemm_fp8_out,
           "row GEMM emitting scaled fp8 only, amax tracked");
     m.def("fp8_scale_update", &fp8_scale_update,

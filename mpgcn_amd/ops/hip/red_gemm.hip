@@ -90,10 +90,19 @@ __launch_bounds__(256) __global__ void red_gemm_kernel(RedGemmParams p) {
             if (row < p.R) {
                 if constexpr (YF8) {
                     const unsigned char* Y8p = (const unsigned char*)p.Y;
+                    long y8b = row * (long)p.N + n0;
+                    if (p.y2) {  // split fp8 rows [y2 (y_k0) | Y (N - y_k0)]
+                        if (n0 < p.y_k0) {
+                            Y8p = (const unsigned char*)p.y2;
+                            y8b = row * (long)p.y_k0 + n0;
+                        } else {
+                            y8b = row * (long)(p.N - p.y_k0) + (n0 - p.y_k0);
+                        }
+                    }
                     if (p.y_vec && n0 + CH <= p.N) {
                         // CH(=8) fp8 bytes -> 8 bf16 via packed converts
                         unsigned long long raw;
-                        __builtin_memcpy(&raw, &Y8p[row * p.N + n0], 8);
+                        __builtin_memcpy(&raw, &Y8p[y8b], 8);
                         const int lo = (int)raw, hi = (int)(raw >> 32);
                         const f32x2 f0 = __builtin_amdgcn_cvt_pk_f32_fp8(lo, false);
                         const f32x2 f1 = __builtin_amdgcn_cvt_pk_f32_fp8(lo, true);
@@ -106,7 +115,7 @@ __launch_bounds__(256) __global__ void red_gemm_kernel(RedGemmParams p) {
                     } else {
                         for (int i = 0; i < CH; ++i)
                             if (n0 + i < p.N) {
-                                const int b = Y8p[row * p.N + n0 + i];
+                                const int b = Y8p[y8b + i];
                                 tmp[i] = (T)__builtin_amdgcn_cvt_pk_f32_fp8(b, false)[0];
                             }
                     }
