@@ -52,9 +52,6 @@ struct RowGemmParams {
     long o_row, o_off;  // OUT element index = m * o_row + o_off + n
     int relu;
     int x_vec;
-    int o_vec;  // OUT rows/offsets chunk-aligned: stage the MFMA accumulator
-                // fragments through per-wave LDS and store b128 row-major
-                // (the raw fragment layout stores 2 B per lane otherwise)
 };
 
 struct LstmStepParams {

@@ -19,9 +19,6 @@ __launch_bounds__(256) __global__ void row_gemm_kernel(RowGemmParams p, int nf16
     const int KP = p.K + MT::LDS_PAD;
     extern __shared__ __attribute__((aligned(16))) char smem[];
     T* ldsWT = (T*)smem;  // [nf16*16][K + PAD]
-    // per-wave output staging tile (vectorized-store epilogue): [16][NP]
-    const int NP = nf16 * 16 + MT::LDS_PAD;
-    T* ldsO = ldsWT + (long)nf16 * 16 * KP;
 
     const T* __restrict__ X = (const T*)p.X;
     const T* __restrict__ X2 = (const T*)p.X2;  // identity block (k < k0)
@@ -82,54 +79,20 @@ __launch_bounds__(256) __global__ void row_gemm_kernel(RowGemmParams p, int nf16
             }
         }
 
-        if (p.o_vec && O && r0 + 16 <= p.R) {
-            // stage the fragment-layout values into this wave's LDS tile,
-            // then store b128 row-major — the raw MFMA layout scatters 2 B
-            // per lane (quarter-wave 32 B runs), which wastes write width
-            T* myO = ldsO + w * 16 * NP;
-            for (int nf = 0; nf < nf16; ++nf) {
+        for (int nf = 0; nf < nf16; ++nf) {
 #pragma unroll
-                for (int r = 0; r < 4; ++r) {
-                    const int mrow = kgrp * 4 + r;
-                    const int n = nf * 16 + lrow;
+            for (int r = 0; r < 4; ++r) {
+                const long m = r0 + kgrp * 4 + r;
+                const int n = nf * 16 + lrow;
+                if (m < p.R && n < p.N) {
                     float v = acc[nf][r];
-                    if (n < p.N) {  // tail fragment rows stage garbage into
-                                    // LDS padding; never stored to global
-                        if (p.bias) v += p.bias[n];
-                        if (p.relu) v = fmaxf(v, 0.f);
-                        if (O8) {
-                            amax = fmaxf(amax, fabsf(v));
-                            O8[(r0 + mrow) * p.o_row + p.o_off + n] =
-                                from_f32<unsigned char>(v * qs);
-                        }
-                    }
-                    myO[mrow * NP + n] = from_f32<T>(v);
-                }
-            }
-            // wave-local visibility: only this wave's ds writes matter
-            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-            const int cpr = p.N / CH;  // chunks per row (o_vec: N % CH == 0)
-            for (int idx = lane; idx < 16 * cpr; idx += WAVE) {
-                const int mrow = idx / cpr, ck = idx % cpr;
-                *(Chunk16*)&O[(r0 + mrow) * p.o_row + p.o_off + ck * CH] =
-                    *(const Chunk16*)&myO[mrow * NP + ck * CH];
-            }
-        } else {
-            for (int nf = 0; nf < nf16; ++nf) {
-#pragma unroll
-                for (int r = 0; r < 4; ++r) {
-                    const long m = r0 + kgrp * 4 + r;
-                    const int n = nf * 16 + lrow;
-                    if (m < p.R && n < p.N) {
-                        float v = acc[nf][r];
-                        if (p.bias) v += p.bias[n];
-                        if (p.relu) v = fmaxf(v, 0.f);
-                        if (O) O[m * p.o_row + p.o_off + n] = from_f32<T>(v);
-                        if (O8) {
-                            amax = fmaxf(amax, fabsf(v));
-                            O8[m * p.o_row + p.o_off + n] =
-                                from_f32<unsigned char>(v * qs);
-                        }
+                    if (p.bias) v += p.bias[n];
+                    if (p.relu) v = fmaxf(v, 0.f);
+                    if (O) O[m * p.o_row + p.o_off + n] = from_f32<T>(v);
+                    if (O8) {
+                        amax = fmaxf(amax, fabsf(v));
+                        O8[m * p.o_row + p.o_off + n] =
+                            from_f32<unsigned char>(v * qs);
                     }
                 }
             }
@@ -149,12 +112,7 @@ extern "C" void row_gemm_launch(RowGemmParams p, int is_f32, hipStream_t stream)
     const int nf16 = (p.N + 15) / 16;
     const int elem = is_f32 ? 4 : 2;
     const int pad = is_f32 ? 4 : 8;
-    const int ch = 16 / elem;
-    p.o_vec = (p.OUT && p.N % ch == 0 && p.o_row % ch == 0 &&
-               p.o_off % ch == 0);
-    const size_t smem = (size_t)nf16 * 16 * (p.K + pad) * elem +
-                        (size_t)4 * 16 * (nf16 * 16 + pad) * elem +
-                        64;  // +64 B slack: B-frag k-tail reads may overrun the last LDS row
+    const size_t smem = (size_t)nf16 * 16 * (p.K + pad) * elem + 64;  // +64 B slack: B-frag k-tail reads may overrun the last LDS row
     long tiles = (p.R + 63) / 64;
     if (tiles > 16384) tiles = 16384;
     dim3 grid((unsigned)tiles), block(256);
@@ -170,10 +128,7 @@ extern "C" void row_gemm_launch(RowGemmParams p, int is_f32, hipStream_t stream)
 // (profiles/SUMMARY.md: 2.2 TB/s bf16), so byte halving is the lever.
 extern "C" void row_gemm_fp8_launch(RowGemmParams p, hipStream_t stream) {
     const int nf16 = (p.N + 15) / 16;
-    p.o_vec = (p.OUT && p.N % 16 == 0 && p.o_row % 16 == 0 &&
-               p.o_off % 16 == 0);
-    const size_t smem = (size_t)nf16 * 16 * (p.K + 16) * 1 +
-                        (size_t)4 * 16 * (nf16 * 16 + 16) * 1 + 64;
+    const size_t smem = (size_t)nf16 * 16 * (p.K + 16) * 1 + 64;
     long tiles = (p.R + 63) / 64;
     if (tiles > 16384) tiles = 16384;
     dim3 grid((unsigned)tiles), block(256);
