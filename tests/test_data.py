@@ -107,3 +107,58 @@ def test_rank_sharding_partitions_train_set():
     per = xf.shape[0] // 2
     assert torch.equal(x0, xf[:per])
     assert torch.equal(x1, xf[per:2 * per])
+
+
+# ---- property-based sweep of the windowing/split/key invariants ----------
+from hypothesis import given, settings, strategies as st  # noqa: E402
+
+
+@settings(max_examples=25, deadline=None)
+@given(obs=st.integers(2, 10), pred=st.integers(1, 4),
+       T=st.integers(40, 90), N=st.integers(3, 10),
+       batch=st.integers(1, 8), seed=st.integers(0, 1000))
+def test_windowing_property_sweep(obs, pred, T, N, batch, seed):
+    """For ANY obs/pred/T/N/batch: every mode's concatenated batches equal the
+    reference's materialized window copies at the right split offsets, and
+    mode lengths partition the sample count with the reference's arithmetic
+    (Data_Container_OD.py:83-163)."""
+    p = _params(N=N, T=T, batch=batch, obs=obs, pred=pred)
+    p["seed"] = seed
+    data = DataInput(p).load_data()
+    gen = DataGenerator(obs, pred, p["split_ratio"])
+    loaders = gen.get_data_loader(data, p)
+    x_ref, y_ref = _reference_windows(data["OD"], obs, pred)
+    n = x_ref.shape[0]
+    ml = gen.split2len(n)
+    assert ml["train"] + ml["validate"] + ml["test"] == n
+    assert ml["train"] == n - int(1.6 / 10 * n) - int(2 / 10 * n)
+    start = 0
+    for mode in ("train", "validate", "test"):
+        xs = [x for x, *_ in loaders[mode]]
+        got = torch.cat(xs) if xs else torch.empty(0)
+        assert got.shape[0] == ml[mode]
+        if ml[mode]:
+            assert torch.equal(got, x_ref[start:start + ml[mode]])
+            ys = torch.cat([y for _, y, *_ in loaders[mode]])
+            assert torch.equal(ys, y_ref[start:start + ml[mode]])
+        start += ml[mode]
+
+
+@settings(max_examples=20, deadline=None)
+@given(obs=st.integers(2, 9), T=st.integers(40, 80), batch=st.integers(1, 5),
+       mode=st.sampled_from(["train", "validate", "test"]))
+def test_day_of_week_key_property(obs, T, batch, mode):
+    """key = (obs + global_index) % 7 for EVERY sample of EVERY mode."""
+    p = _params(N=6, T=T, batch=batch, obs=obs, pred=1)
+    data = DataInput(p).load_data()
+    gen = DataGenerator(obs, 1, p["split_ratio"])
+    loaders = gen.get_data_loader(data, p)
+    O_dyn = data["O_dyn_G"]
+    n = data["OD"].shape[0] - obs - 1
+    ml = gen.split2len(n)
+    g = {"train": 0, "validate": ml["train"],
+         "test": ml["train"] + ml["validate"]}[mode]
+    for x, y, O_g, D_g in loaders[mode]:
+        for i in range(O_g.shape[0]):
+            assert torch.equal(O_g[i], O_dyn[:, :, (obs + g + i) % 7])
+        g += O_g.shape[0]
