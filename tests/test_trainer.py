@@ -172,3 +172,47 @@ def test_fused_adam_trainer_option(tmp_path, capsys):
     final_ref = float(ref_losses[-1][1])
     assert abs(final_fused - final_ref) < 0.25 * abs(final_ref) + 1e-3, (
         final_fused, final_ref)
+
+
+def test_main_cli_three_perspectives_attention(tmp_path):
+    """CLI path for the extended surface: 3 perspectives + attention fusion
+    trains end-to-end and writes the reference-schema checkpoint whose keys
+    include the third branch and the attention-fusion parameters."""
+    import subprocess
+    import sys
+    from pathlib import Path
+
+    repo = Path(__file__).resolve().parent.parent
+    out = tmp_path / "cli3_out"
+    r = subprocess.run(
+        [sys.executable, str(repo / "Main.py"), "-GPU", "cpu",
+         "-synthetic-nodes", "12", "-synthetic-days", "60",
+         "-M", "3", "-fusion", "attention", "-epoch", "1", "-out", str(out)],
+        capture_output=True, text=True, timeout=240, cwd=str(repo))
+    assert r.returncode == 0, r.stderr[-2000:]
+    ckpt = torch.load(out / "MPGCN_od.pkl", map_location="cpu",
+                      weights_only=True)
+    keys = set(ckpt["state_dict"])
+    assert any(k.startswith("branch_models.2.") for k in keys)
+    assert any("fusion" in k or "attn" in k for k in keys), sorted(keys)[-5:]
+
+
+def test_rollout_equals_manual_feedback(tmp_path):
+    """trainer._rollout's autoregressive feedback must equal a hand-rolled
+    loop: each horizon's input is the previous window shifted by one with the
+    model's own prediction appended (Model_Trainer.py:160-163 semantics)."""
+    params, trainer, loaders = _setup(tmp_path, num_epochs=1, pred_len=3)
+    trainer.model.eval()
+    x, y, O_g, D_g = next(iter(loaders["test"]))
+    dyn = (trainer.preprocess_dynamic_graph(O_g),
+           trainer.preprocess_dynamic_graph(D_g))
+    with torch.no_grad():
+        got = trainer._rollout(x, dyn, region=False)
+        cur = x
+        steps = []
+        for _ in range(3):
+            s = trainer.model(x_seq=cur, G_list=trainer._graph_list(dyn))
+            cur = torch.cat([cur[:, 1:], s], dim=1)
+            steps.append(s)
+        want = torch.cat(steps, dim=1)
+    assert torch.equal(got, want)
