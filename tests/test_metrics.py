@@ -57,3 +57,35 @@ def test_evaluate_torch_matches_numpy():
     assert abs(mae - tmae) < 1e-6
     assert abs(mape - tmape) < 1e-6
     assert abs(m.PCC(yp, yt) - tpcc) < 1e-5
+
+
+# ---- property sweep: MetricAccumulator == whole-array numpy on any split ----
+from hypothesis import given, settings, strategies as st  # noqa: E402
+
+
+@settings(max_examples=30, deadline=None)
+@given(n=st.integers(2, 400), chunks=st.integers(1, 6),
+       scale=st.sampled_from([0.1, 1.0, 50.0]), seed=st.integers(0, 999))
+def test_metric_accumulator_chunking_invariance(n, chunks, scale, seed):
+    """Streaming sufficient statistics must be invariant to HOW the data is
+    chunked and match the reference numpy definitions on the concatenation —
+    this is exactly what distributed eval relies on (per-rank partial batches
+    all-reduced, trainer.test)."""
+    from mpgcn_amd.train.metrics import MetricAccumulator
+
+    rng = np.random.default_rng(seed)
+    p = (rng.normal(size=(n,)) * scale).astype(np.float64)
+    t = (rng.normal(size=(n,)) * scale + 2.0).astype(np.float64)
+    acc = MetricAccumulator()
+    bounds = sorted(rng.integers(0, n, size=max(chunks - 1, 0)).tolist())
+    pieces = np.split(np.arange(n), bounds)
+    for idx in pieces:
+        if len(idx):
+            acc.update(torch.from_numpy(p[idx]), torch.from_numpy(t[idx]))
+    mse, rmse, mae, mape, pcc = acc.finalize()
+    assert np.isclose(mse, M.MSE(p, t), rtol=1e-10)
+    assert np.isclose(rmse, M.RMSE(p, t), rtol=1e-10)
+    assert np.isclose(mae, M.MAE(p, t), rtol=1e-10)
+    assert np.isclose(mape, M.MAPE(p, t), rtol=1e-10)
+    if np.std(p) > 1e-12 and np.std(t) > 1e-12:
+        assert np.isclose(pcc, M.PCC(p, t), rtol=1e-8)
